@@ -1,0 +1,98 @@
+"""Launcher — process lifecycle.
+
+Parity target: reference ``veles/launcher.py`` (standalone / master /
+slave modes over Twisted+ZeroMQ).
+
+MI355X redesign: the master/slave topology is replaced by one process
+per GPU launched via ``torch.distributed`` (torchrun or
+``--gpus N`` self-spawn); the Launcher owns device selection, RCCL/gloo
+process-group bring-up (through ``veles.znicz_amd.parallel``), workflow
+initialize/run, and snapshot resume.  ``is_master``/``is_slave`` are
+kept: rank 0 is "master" (it writes snapshots and logs), every rank is a
+worker.
+"""
+
+import os
+
+from veles.backends import Device, NumpyDevice
+from veles.logger import Logger
+
+
+class Launcher(Logger):
+    def __init__(self, **kwargs):
+        super(Launcher, self).__init__()
+        self.workflow = None
+        self.device = None
+        self.dist = None
+        self.backend = kwargs.get("backend", None)
+        self.snapshot = kwargs.get("snapshot", None)
+        self.stealth = kwargs.get("stealth", False)
+        self._stopped = False
+
+    # -- topology ----------------------------------------------------------
+    @property
+    def rank(self):
+        return self.dist.rank if self.dist else 0
+
+    @property
+    def world(self):
+        return self.dist.world if self.dist else 1
+
+    @property
+    def is_master(self):
+        return self.rank == 0
+
+    @property
+    def is_slave(self):
+        return self.rank != 0
+
+    @property
+    def is_standalone(self):
+        return self.world == 1
+
+    # -- unit-ref protocol (workflows register with their launcher) --------
+    def add_ref(self, unit):
+        self.workflow = unit
+
+    def del_ref(self, unit):
+        pass
+
+    def on_workflow_finished(self):
+        self.stop()
+
+    # -- lifecycle ----------------------------------------------------------
+    def make_device(self):
+        if self.backend == "numpy":
+            return NumpyDevice()
+        if self.device is None:
+            self.device = Device()
+        return self.device
+
+    def init_distributed(self):
+        """Bring up the DP context if launched under torchrun."""
+        if "WORLD_SIZE" in os.environ and \
+                int(os.environ["WORLD_SIZE"]) > 1:
+            from veles.znicz_amd.parallel import DistributedContext
+            self.dist = DistributedContext.create()
+        return self.dist
+
+    def boot(self, **kwargs):
+        device = self.make_device()
+        self.init_distributed()
+        wf = self.workflow
+        if self.dist is not None:
+            wf.dist = self.dist
+            if wf.loader is not None:
+                wf.loader.shard(self.dist.rank, self.dist.world)
+        wf.initialize(device=device, **kwargs)
+        wf.run()
+        return wf
+
+    def stop(self):
+        if self._stopped:
+            return
+        self._stopped = True
+        if self.workflow is not None and self.workflow.is_running:
+            self.workflow.stop()
+        if self.dist is not None:
+            self.dist.finalize()

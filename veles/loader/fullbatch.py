@@ -1,0 +1,131 @@
+"""In-memory (full-batch) loaders.
+
+Parity target: reference ``veles/loader/fullbatch.py``
+(``FullBatchLoader``, ``IFullBatchLoader``, ``FullBatchLoaderMSE``,
+``on_device`` option).
+
+MI355X redesign: with 288 GB HBM3E the whole dataset comfortably lives
+device-resident (``on_device=True``); the minibatch fill is then a
+device-side ``index_select`` gather (no H2D traffic in the hot loop).
+On the numpy path it is a host ``take``.
+"""
+
+import numpy
+
+from veles.loader.base import Loader
+from veles.memory import Array
+
+
+class IFullBatchLoader(object):
+    """Marker (parity)."""
+
+
+class FullBatchLoader(Loader):
+    """Dataset held in one Array; minibatch = gather by indices."""
+
+    def __init__(self, workflow, **kwargs):
+        super(FullBatchLoader, self).__init__(workflow, **kwargs)
+        self.original_data = Array()
+        self.original_labels = Array()
+        self.on_device = kwargs.get("on_device", True)
+        self.data_dtype = kwargs.get("data_dtype", numpy.float32)
+        self.device_dtype = kwargs.get("device_dtype", None)
+
+    def load_data(self):
+        """Subclasses fill original_data/original_labels + class_lengths."""
+        raise NotImplementedError
+
+    def create_minibatch_data(self):
+        shape = (self.max_minibatch_size,) + \
+            tuple(self.original_data.shape[1:])
+        self.minibatch_data.mem = numpy.zeros(shape, dtype=self.data_dtype)
+        if self.device_dtype:
+            self.minibatch_data.device_dtype = self.device_dtype
+
+    def initialize(self, device=None, **kwargs):
+        super(FullBatchLoader, self).initialize(device=device, **kwargs)
+        if self.on_device and device is not None and device.is_gpu:
+            if self.device_dtype:
+                self.original_data.device_dtype = self.device_dtype
+            self.original_data.initialize(device)
+            self.original_labels.initialize(device)
+
+    def fill_minibatch(self):
+        size = self.minibatch_size
+        idx = self.minibatch_indices.mem[:size]
+        if self.on_device and self.original_data.on_gpu:
+            import torch
+            dev_idx = torch.from_numpy(
+                numpy.ascontiguousarray(idx.astype(numpy.int64))).to(
+                self.original_data.devmem.device, non_blocking=True)
+            # device-side gather straight into the minibatch buffer
+            torch.index_select(
+                self.original_data.devmem, 0, dev_idx,
+                out=self.minibatch_data.devmem[:size])
+            self.minibatch_data.touch_device()
+            if self.original_labels.devmem is not None:
+                torch.index_select(
+                    self.original_labels.devmem, 0, dev_idx,
+                    out=self.minibatch_labels.devmem[:size])
+                self.minibatch_labels.touch_device()
+            # host mirrors for Decision/debug paths
+            self.minibatch_labels.mem[:size] = \
+                self.original_labels.mem[idx]
+        else:
+            self.minibatch_data.map_invalidate()
+            self.minibatch_data.mem[:size] = self.original_data.mem[idx]
+            if self.original_labels:
+                self.minibatch_labels.map_invalidate()
+                self.minibatch_labels.mem[:size] = \
+                    self.original_labels.mem[idx]
+
+    def _pad_minibatch(self, size):
+        if not (self.on_device and self.original_data.on_gpu):
+            super(FullBatchLoader, self)._pad_minibatch(size)
+        else:
+            self.minibatch_data.devmem[size:] = 0
+            self.minibatch_labels.mem[size:] = -1
+
+
+class FullBatchLoaderMSE(FullBatchLoader):
+    """Adds regression targets (for MSE evaluators / autoencoders)."""
+
+    def __init__(self, workflow, **kwargs):
+        super(FullBatchLoaderMSE, self).__init__(workflow, **kwargs)
+        self.original_targets = Array()
+        self.minibatch_targets = Array()
+        self.targets_shape = None
+
+    def create_minibatch_data(self):
+        super(FullBatchLoaderMSE, self).create_minibatch_data()
+        tshape = (self.max_minibatch_size,) + \
+            tuple(self.original_targets.shape[1:])
+        self.minibatch_targets.mem = numpy.zeros(
+            tshape, dtype=self.data_dtype)
+        if self.device_dtype:
+            self.minibatch_targets.device_dtype = self.device_dtype
+
+    def initialize(self, device=None, **kwargs):
+        super(FullBatchLoaderMSE, self).initialize(device=device, **kwargs)
+        if self.on_device and device is not None and device.is_gpu:
+            if self.device_dtype:
+                self.original_targets.device_dtype = self.device_dtype
+            self.original_targets.initialize(device)
+            self.minibatch_targets.initialize(device)
+
+    def fill_minibatch(self):
+        super(FullBatchLoaderMSE, self).fill_minibatch()
+        size = self.minibatch_size
+        idx = self.minibatch_indices.mem[:size]
+        if self.on_device and self.original_targets.on_gpu:
+            import torch
+            dev_idx = torch.from_numpy(
+                numpy.ascontiguousarray(idx.astype(numpy.int64))).to(
+                self.original_targets.devmem.device, non_blocking=True)
+            torch.index_select(
+                self.original_targets.devmem, 0, dev_idx,
+                out=self.minibatch_targets.devmem[:size])
+            self.minibatch_targets.touch_device()
+        else:
+            self.minibatch_targets.map_invalidate()
+            self.minibatch_targets.mem[:size] = self.original_targets.mem[idx]

@@ -1,0 +1,212 @@
+"""Python surface of the hand-written CDNA4 kernel extension ``_mzk``.
+
+This is the MI355X-native replacement for the reference's runtime-compiled
+OpenCL/CUDA kernel sets (znicz/ocl/*.cl + znicz/cuda/*.cu) and the
+opencl4py/cuda4py binding layer: every kernel is precompiled into one
+in-tree HIP extension (hipcc --offload-arch=gfx950) and called through
+thin wrappers here.  On a GPU box the extension is REQUIRED — ops fail
+loudly instead of falling back to eager PyTorch.
+
+Build: ``python veles/znicz_amd/ops/setup.py build_ext --inplace`` (or
+``__graft_entry__.build()`` at the repo root).
+"""
+
+import os
+
+_ext = None
+_ext_err = None
+
+
+def _load():
+    global _ext, _ext_err
+    if _ext is not None or _ext_err is not None:
+        return _ext
+    try:
+        import torch  # noqa: F401  (must be imported before the ext .so)
+        from veles.znicz_amd.ops import _mzk  # built in-tree
+        _ext = _mzk
+    except ImportError as e:
+        _ext_err = e
+    return _ext
+
+
+def ext():
+    return _load()
+
+
+def available():
+    return _load() is not None
+
+
+def require():
+    if _load() is None:
+        raise RuntimeError(
+            "the _mzk HIP extension is not built (%s); on a GPU box this "
+            "is fatal — run `python veles/znicz_amd/ops/setup.py "
+            "build_ext --inplace` (hipcc --offload-arch=gfx950)"
+            % _ext_err)
+    return _ext
+
+
+# ---------------------------------------------------------------------------
+# tensor helpers
+# ---------------------------------------------------------------------------
+
+def as2d_bf16(arr):
+    """View an Array's device tensor as 2-D [batch, features], bf16."""
+    import torch
+    t = arr.devmem
+    assert t is not None, "Array has no device tensor (%r)" % (arr,)
+    assert t.dtype == torch.bfloat16, \
+        "expected bf16 device tensor, got %s" % t.dtype
+    return t.view(t.shape[0], -1)
+
+
+def as2d(arr):
+    t = arr.devmem
+    return t.view(t.shape[0], -1)
+
+
+def attach_working_copies(weights_arr, transposed=False):
+    """Give a weights Array its bf16 working copy and bf16ᵀ copy.
+
+    weights_arr.devmem is the fp32 master; .bf16 mirrors it; .bf16_t is
+    the transposed bf16 image used by whichever of fwd/dgrad needs the
+    opposite layout.  The fused SGD kernel refreshes both.
+    """
+    import torch
+    w = weights_arr.devmem
+    if getattr(weights_arr, "bf16", None) is None:
+        weights_arr.bf16 = w.to(torch.bfloat16)
+        weights_arr.bf16_t = w.t().contiguous().to(torch.bfloat16)
+    return weights_arr
+
+
+def refresh_working_copies(weights_arr):
+    """Host-side refresh (used after map_write edits); the SGD kernel
+    does this on-device in the normal path."""
+    w = weights_arr.devmem
+    if getattr(weights_arr, "bf16", None) is not None:
+        import torch
+        weights_arr.bf16.copy_(w.to(torch.bfloat16))
+        weights_arr.bf16_t.copy_(w.t().to(torch.bfloat16))
+
+
+def labels_on_device(labels_arr, device):
+    """int32 labels tensor on the compute device (cached on the Array)."""
+    import torch
+    t = getattr(labels_arr, "_dev_i32", None)
+    if t is None or t.shape[0] != labels_arr.shape[0]:
+        t = torch.empty(labels_arr.shape[0], dtype=torch.int32,
+                        device=device.torch_device)
+        labels_arr._dev_i32 = t
+    if labels_arr.devmem is not None and \
+            labels_arr.devmem.dtype == torch.int32:
+        return labels_arr.devmem
+    import numpy
+    t.copy_(torch.from_numpy(numpy.ascontiguousarray(
+        labels_arr.mem.astype(numpy.int32))), non_blocking=True)
+    return t
+
+
+# ---------------------------------------------------------------------------
+# kernel wrappers (1:1 with bindings.cpp)
+# ---------------------------------------------------------------------------
+
+def gemm_nt(a, b, out, bias=None, activation=0):
+    """out[M,N] = act(a[M,K] · b[N,K]ᵀ + bias).  a,b bf16; out bf16|f32."""
+    require().gemm_nt(a, b, out, bias, int(activation))
+
+
+def gemm_tn(a, b, out, beta=0.0):
+    """out[M,N] (fp32) = a[K,M]ᵀ · b[K,N] (+beta*out).  a,b bf16."""
+    require().gemm_tn(a, b, out, float(beta))
+
+
+def act_bwd_inplace(err, output, activation):
+    """err ∘= act'(output) — both bf16, same shape."""
+    require().act_bwd(err, output, int(activation))
+
+
+def act_fwd_inplace(x, activation):
+    require().act_fwd(x, int(activation))
+
+
+def col_sum(a, out):
+    """out[N] (fp32) = Σ_rows a[M,N] (bf16)."""
+    require().col_sum(a, out)
+
+
+def sgd_step(param_arr, grad, vel, lr, decay, l1_vs_l2, moment, scale,
+             transposed_copy=False):
+    """Fused SGD: master fp32 update + bf16 (+bf16ᵀ) refresh.
+
+    grad is the raw batch-summed fp32 gradient; ``scale`` (=1/batch or
+    1/(batch*world)) normalizes it inside the kernel.
+    """
+    master = param_arr.devmem
+    bf16 = getattr(param_arr, "bf16", None)
+    bf16_t = getattr(param_arr, "bf16_t", None)
+    require().sgd_step(master, grad, vel, bf16, bf16_t,
+                       float(lr), float(decay), float(l1_vs_l2),
+                       float(moment), float(scale))
+
+
+def softmax_fwd(x, max_idx):
+    """Row-wise in-place softmax on fp32 x[M,N]; writes argmax to
+    max_idx[M] (int32)."""
+    require().softmax_fwd(x, max_idx)
+
+
+def eval_softmax(probs, labels, err_out, n_err, batch):
+    require().eval_softmax(probs, labels, err_out, n_err, int(batch))
+
+
+def eval_mse(y, target, err_out, mse, metrics, batch, root):
+    require().eval_mse(y, target, err_out, mse, metrics, int(batch),
+                       int(root))
+
+
+# conv / pooling / LRN / dropout -------------------------------------------
+
+def conv_fwd(x, w, out, bias, conv_desc, activation=0):
+    require().conv_fwd(x, w, out, bias, conv_desc, int(activation))
+
+
+def conv_dgrad(err_out, w_dgrad, err_in, conv_desc):
+    require().conv_dgrad(err_out, w_dgrad, err_in, conv_desc)
+
+
+def conv_wgrad(err_out, x, grad_w, conv_desc):
+    require().conv_wgrad(err_out, x, grad_w, conv_desc)
+
+
+def pool_fwd(x, out, offs, desc, mode):
+    require().pool_fwd(x, out, offs, desc, int(mode))
+
+
+def pool_bwd(err_out, err_in, offs, desc, mode):
+    require().pool_bwd(err_out, err_in, offs, desc, int(mode))
+
+
+def lrn_fwd(x, out, denom, alpha, beta, k, n):
+    require().lrn_fwd(x, out, denom, float(alpha), float(beta), float(k),
+                      int(n))
+
+
+def lrn_bwd(err_out, x, out, denom, err_in, alpha, beta, k, n):
+    require().lrn_bwd(err_out, x, out, denom, err_in, float(alpha),
+                      float(beta), float(k), int(n))
+
+
+def dropout_fwd(x, out, mask, ratio, seed, counter):
+    require().dropout_fwd(x, out, mask, float(ratio), int(seed),
+                          int(counter))
+
+
+def dropout_bwd(err_out, mask, err_in):
+    require().dropout_bwd(err_out, mask, err_in)
+
+
+def transpose2d(a, out):
+    require().transpose2d(a, out)
