@@ -1,0 +1,365 @@
+"""HIP kernel numerics vs plain numpy/fp32 references (MI355X only).
+
+Every kernel in ops/_mzk is cross-checked against a numpy fp32 reference
+on asymmetric random data (transpose-detecting per the CDNA4 guide).
+bf16-aware tolerances: fp32-accumulated GEMM vs fp32 reference on bf16
+inputs ⇒ relative error ~1e-2.
+"""
+
+import numpy
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+torch = pytest.importorskip("torch")
+if not torch.cuda.is_available():
+    pytest.skip("no GPU", allow_module_level=True)
+
+from veles.znicz_amd import ops  # noqa: E402
+
+RNG = numpy.random.RandomState(42)
+
+
+def dev_bf16(a):
+    return torch.from_numpy(numpy.ascontiguousarray(a)).to(
+        "cuda", torch.bfloat16)
+
+
+def to_np(t):
+    return t.detach().float().cpu().numpy()
+
+
+def bf_round(a):
+    """Round fp32 numpy through bf16 (what the kernels actually see)."""
+    return to_np(dev_bf16(a))
+
+
+def assert_close(got, want, rtol=2e-2, atol=None):
+    want = numpy.asarray(want, numpy.float32)
+    if atol is None:
+        atol = 1e-2 * max(1.0, float(numpy.abs(want).max()))
+    numpy.testing.assert_allclose(got, want, rtol=rtol, atol=atol)
+
+
+# ---------------------------------------------------------------------------
+# GEMM cores
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("M,N,K", [(64, 64, 64), (128, 128, 128),
+                                   (256, 1000, 4096), (100, 70, 72),
+                                   (130, 257, 192)])
+def test_gemm_nt(M, N, K):
+    a = RNG.uniform(-1, 1, (M, K)).astype(numpy.float32)
+    b = RNG.uniform(-1, 1, (N, K)).astype(numpy.float32)
+    da, db = dev_bf16(a), dev_bf16(b)
+    out = torch.empty((M, N), dtype=torch.bfloat16, device="cuda")
+    ops.gemm_nt(da, db, out)
+    want = bf_round(a).dot(bf_round(b).T)
+    assert_close(to_np(out), want, atol=0.05 * max(1, K / 500))
+
+
+def test_gemm_nt_bias_act_f32out():
+    M, N, K = 96, 200, 128
+    a = RNG.uniform(-1, 1, (M, K)).astype(numpy.float32)
+    b = RNG.uniform(-1, 1, (N, K)).astype(numpy.float32)
+    bias = RNG.uniform(-1, 1, N).astype(numpy.float32)
+    out = torch.empty((M, N), dtype=torch.float32, device="cuda")
+    ops.gemm_nt(dev_bf16(a), dev_bf16(b), out,
+                bias=torch.from_numpy(bias).cuda(), activation=0)
+    want = bf_round(a).dot(bf_round(b).T) + bias
+    assert_close(to_np(out), want)
+    # strict relu epilogue
+    out2 = torch.empty((M, N), dtype=torch.bfloat16, device="cuda")
+    ops.gemm_nt(dev_bf16(a), dev_bf16(b), out2,
+                bias=torch.from_numpy(bias).cuda(), activation=3)
+    assert_close(to_np(out2), numpy.maximum(want, 0))
+
+
+@pytest.mark.parametrize("M,N,K", [(64, 64, 64), (256, 128, 512),
+                                   (96, 200, 1000), (384, 2304, 2048)])
+def test_gemm_tn(M, N, K):
+    a = RNG.uniform(-1, 1, (K, M)).astype(numpy.float32)
+    b = RNG.uniform(-1, 1, (K, N)).astype(numpy.float32)
+    out = torch.zeros((M, N), dtype=torch.float32, device="cuda")
+    ops.gemm_tn(dev_bf16(a), dev_bf16(b), out)
+    want = bf_round(a).T.dot(bf_round(b))
+    assert_close(to_np(out), want, atol=0.05 * max(1, K / 500))
+    # beta accumulate
+    ops.gemm_tn(dev_bf16(a), dev_bf16(b), out, beta=1.0)
+    assert_close(to_np(out), 2 * want, atol=0.1 * max(1, K / 500))
+
+
+def test_transpose2d():
+    a = RNG.uniform(-1, 1, (130, 70)).astype(numpy.float32)
+    da = dev_bf16(a)
+    out = torch.empty((70, 130), dtype=torch.bfloat16, device="cuda")
+    ops.transpose2d(da, out)
+    numpy.testing.assert_array_equal(to_np(out), bf_round(a).T)
+
+
+# ---------------------------------------------------------------------------
+# elementwise
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("act", [1, 2, 3, 4])
+def test_act_fwd_bwd(act):
+    from veles.znicz_amd import activation_defs as adef
+    x = RNG.uniform(-2, 2, 1003).astype(numpy.float32)
+    dx = dev_bf16(x)
+    ops.act_fwd_inplace(dx, act)
+    y = adef.apply_forward(act, bf_round(x))
+    assert_close(to_np(dx), y, atol=0.03)
+    err = RNG.uniform(-1, 1, 1003).astype(numpy.float32)
+    derr = dev_bf16(err)
+    ops.act_bwd_inplace(derr, dx, act)
+    want = bf_round(err) * adef.derivative_from_output(act, to_np(dx))
+    assert_close(to_np(derr), want, atol=0.05)
+
+
+def test_sgd_step_momentum_decay():
+    n = 5000
+    w = RNG.uniform(-1, 1, n).astype(numpy.float32)
+    g = RNG.uniform(-1, 1, n).astype(numpy.float32)
+    v = RNG.uniform(-0.1, 0.1, n).astype(numpy.float32)
+    dw = torch.from_numpy(w.copy()).cuda()
+    dg = torch.from_numpy(g).cuda()
+    dv = torch.from_numpy(v.copy()).cuda()
+    wb = dw.to(torch.bfloat16)
+
+    class FakeArr:
+        devmem = dw
+        bf16 = wb
+    lr, decay, l1, mom, scale = 0.1, 0.01, 0.3, 0.9, 1.0 / 32
+    ops.sgd_step(FakeArr, dg, dv, lr, decay, l1, mom, scale)
+    l1l2 = l1 * numpy.sign(w) + (1 - l1) * w
+    gg = g * scale + decay * l1l2
+    vv = mom * v + lr * gg
+    ww = w - vv
+    numpy.testing.assert_allclose(to_np(dw), ww, rtol=1e-5, atol=1e-6)
+    numpy.testing.assert_allclose(to_np(dv), vv, rtol=1e-5, atol=1e-6)
+    assert_close(to_np(wb), ww, atol=0.02)
+
+
+def test_col_sum():
+    M, N = 300, 200
+    a = RNG.uniform(-1, 1, (M, N)).astype(numpy.float32)
+    out = torch.empty(N, dtype=torch.float32, device="cuda")
+    ops.col_sum(dev_bf16(a), out)
+    assert_close(to_np(out), bf_round(a).sum(axis=0), atol=0.2)
+
+
+def test_dropout():
+    n = 100000
+    x = numpy.ones(n, numpy.float32)
+    dx = dev_bf16(x)
+    dy = torch.empty_like(dx)
+    mask = torch.empty(n, dtype=torch.uint8, device="cuda")
+    ops.dropout_fwd(dx, dy, mask, 0.5, 1234, 1)
+    m = mask.cpu().numpy()
+    keep = m.mean()
+    assert 0.48 < keep < 0.52
+    y = to_np(dy)
+    assert numpy.allclose(y[m == 0], 0)
+    assert_close(y[m == 1], 2.0 * numpy.ones((m == 1).sum()), atol=0.02)
+    # bwd: err *= mask * 2
+    err = dev_bf16(numpy.ones(n, numpy.float32))
+    ops.dropout_bwd(err, mask, 0.5)
+    e = to_np(err)
+    assert numpy.allclose(e[m == 0], 0)
+    assert_close(e[m == 1], 2.0 * numpy.ones((m == 1).sum()), atol=0.02)
+    # deterministic given (seed, counter)
+    dy2 = torch.empty_like(dx)
+    mask2 = torch.empty_like(mask)
+    ops.dropout_fwd(dx, dy2, mask2, 0.5, 1234, 1)
+    assert torch.equal(mask, mask2)
+
+
+# ---------------------------------------------------------------------------
+# softmax + evaluators
+# ---------------------------------------------------------------------------
+
+def test_softmax_and_eval():
+    M, N, batch = 64, 10, 50
+    logits = RNG.uniform(-3, 3, (M, N)).astype(numpy.float32)
+    x = torch.from_numpy(logits.copy()).cuda()
+    mi = torch.empty(M, dtype=torch.int32, device="cuda")
+    ops.softmax_fwd(x, mi)
+    e = numpy.exp(logits - logits.max(axis=1, keepdims=True))
+    probs = e / e.sum(axis=1, keepdims=True)
+    numpy.testing.assert_allclose(to_np(x), probs, rtol=1e-4, atol=1e-5)
+    numpy.testing.assert_array_equal(mi.cpu().numpy(),
+                                     logits.argmax(axis=1))
+    labels = RNG.randint(0, N, M).astype(numpy.int32)
+    dl = torch.from_numpy(labels).cuda()
+    err = torch.empty((M, N), dtype=torch.bfloat16, device="cuda")
+    n_err = torch.zeros(2, dtype=torch.int32, device="cuda")
+    ops.eval_softmax(x, dl, err, n_err, batch)
+    want_err = probs.copy()
+    want_err[numpy.arange(M), labels] -= 1
+    want_err[batch:] = 0
+    assert_close(to_np(err), want_err, atol=0.02)
+    want_nerr = int((probs[:batch].argmax(axis=1) !=
+                     labels[:batch]).sum())
+    assert int(n_err.cpu()[0]) == want_nerr
+
+
+def test_eval_mse():
+    M, N, batch = 40, 30, 33
+    y = RNG.uniform(-1, 1, (M, N)).astype(numpy.float32)
+    t = RNG.uniform(-1, 1, (M, N)).astype(numpy.float32)
+    err = torch.empty((M, N), dtype=torch.bfloat16, device="cuda")
+    mse = torch.empty(M, dtype=torch.float32, device="cuda")
+    metrics = torch.zeros(3, dtype=torch.float32, device="cuda")
+    metrics[2] = 1e30
+    ops.eval_mse(dev_bf16(y), dev_bf16(t), err, mse, metrics, batch, 1)
+    d = bf_round(y) - bf_round(t)
+    assert_close(to_np(err)[:batch], d[:batch], atol=0.02)
+    want_mse = numpy.sqrt((d[:batch] ** 2).mean(axis=1))
+    numpy.testing.assert_allclose(mse.cpu().numpy()[:batch], want_mse,
+                                  rtol=1e-2, atol=1e-3)
+
+
+# ---------------------------------------------------------------------------
+# conv / pooling / LRN via the unit numpy paths
+# ---------------------------------------------------------------------------
+
+def unit_pair_check(make_unit, data, atol_fwd=0.05, extra_check=None):
+    """Run a forward unit on numpy and on HIP, compare outputs."""
+    from veles.backends import NumpyDevice, ROCmDevice
+    from veles.dummy import DummyWorkflow
+    from veles.memory import Array
+
+    outs = {}
+    for dev_kind in ("numpy", "rocm"):
+        wf = DummyWorkflow()
+        unit = make_unit(wf)
+        unit.input = Array(data.copy())
+        unit.input.device_dtype = "bf16"
+        dev = NumpyDevice() if dev_kind == "numpy" else ROCmDevice(0)
+        unit.input.initialize(dev)
+        r = unit.initialize(device=dev)
+        assert r is not False
+        unit.run()
+        unit.output.map_read()
+        outs[dev_kind] = unit.output.mem.copy()
+    assert_close(outs["rocm"], outs["numpy"], atol=atol_fwd)
+    return outs
+
+
+def test_conv_unit_fwd():
+    from veles.znicz_amd.conv import ConvStrictRELU
+    from veles import prng
+    data = RNG.uniform(-1, 1, (4, 13, 13, 8)).astype(numpy.float32)
+
+    def make(wf):
+        prng.get().seed(77)
+        return ConvStrictRELU(wf, n_kernels=16, kx=3, ky=3,
+                              padding=(1, 1, 1, 1), sliding=(1, 1))
+    unit_pair_check(make, data, atol_fwd=0.08)
+
+
+def test_conv_unit_fwd_stride():
+    from veles.znicz_amd.conv import Conv
+    from veles import prng
+    data = RNG.uniform(-1, 1, (2, 21, 21, 8)).astype(numpy.float32)
+
+    def make(wf):
+        prng.get().seed(78)
+        return Conv(wf, n_kernels=24, kx=5, ky=5, sliding=(4, 4))
+    unit_pair_check(make, data, atol_fwd=0.08)
+
+
+def test_pooling_max_avg():
+    from veles.znicz_amd.pooling import MaxPooling, AvgPooling
+    data = RNG.uniform(-1, 1, (3, 9, 9, 16)).astype(numpy.float32)
+    unit_pair_check(lambda wf: MaxPooling(wf, kx=3, ky=3,
+                                          sliding=(2, 2)), data,
+                    atol_fwd=0.02)
+    unit_pair_check(lambda wf: AvgPooling(wf, kx=2, ky=2), data,
+                    atol_fwd=0.02)
+
+
+def test_lrn_fwd():
+    from veles.znicz_amd.normalization import LRNormalizerForward
+    data = RNG.uniform(-1, 1, (2, 5, 5, 32)).astype(numpy.float32)
+    unit_pair_check(lambda wf: LRNormalizerForward(
+        wf, alpha=0.001, beta=0.75, n=5, k=1.0), data, atol_fwd=0.02)
+
+
+# ---------------------------------------------------------------------------
+# full backward numerics: GPU vs numpy on one training step
+# ---------------------------------------------------------------------------
+
+def _train_steps(device, steps=3):
+    """Run MNIST-shape MLP a few steps; return final weights."""
+    from veles import prng
+    from veles.config import root
+    from veles.dummy import DummyLauncher
+    prng.get().seed(1234)
+    prng.get("loader").seed(5678)
+    root.mnist.decision.max_epochs = 1
+    root.mnist.loader.update({"n_train": 120, "n_valid": 0,
+                              "minibatch_size": 60})
+    from veles.znicz_amd.models.mnist import MnistWorkflow
+    wf = MnistWorkflow(DummyLauncher())
+    wf.initialize(device=device)
+    for _ in range(steps):
+        wf.loader.run()
+        for f in wf.forwards:
+            f.run()
+        wf.evaluator.run()
+        wf.decision.run()
+        for g in reversed(wf.gds):
+            g.run()
+    for f in wf.forwards:
+        f.weights.map_read()
+    return [f.weights.mem.copy() for f in wf.forwards], wf
+
+
+def test_mlp_step_matches_numpy():
+    from veles.backends import NumpyDevice, ROCmDevice
+    w_np, _ = _train_steps(NumpyDevice())
+    w_gpu, _ = _train_steps(ROCmDevice(0))
+    for a, b in zip(w_np, w_gpu):
+        assert_close(b, a, atol=0.03)
+
+
+def test_mnist_trains_on_gpu():
+    from veles import prng
+    from veles.backends import ROCmDevice
+    from veles.config import root
+    from veles.dummy import DummyLauncher
+    from veles.loader.base import TRAIN
+    prng.get().seed(1234)
+    prng.get("loader").seed(5678)
+    root.mnist.decision.max_epochs = 4
+    root.mnist.loader.update({"n_train": 600, "n_valid": 100,
+                              "minibatch_size": 60})
+    from veles.znicz_amd.models.mnist import MnistWorkflow
+    wf = MnistWorkflow(DummyLauncher())
+    wf.initialize(device=ROCmDevice(0))
+    wf.run()
+    assert wf.is_finished
+    assert wf.decision.epoch_n_err_pt[TRAIN] < 10.0
+
+
+def test_cifar_cnn_trains_on_gpu():
+    """BASELINE config 2 model learns on the HIP conv/pool path."""
+    from veles import prng
+    from veles.backends import ROCmDevice
+    from veles.config import root
+    from veles.dummy import DummyLauncher
+    from veles.loader.base import TRAIN
+    prng.get().seed(4321)
+    prng.get("loader").seed(8765)
+    root.cifar.update({"loader": {"minibatch_size": 128, "n_train": 512,
+                                  "n_valid": 128},
+                       "decision": {"max_epochs": 3}})
+    from veles.znicz_amd.models.cifar import CifarWorkflow
+    wf = CifarWorkflow(DummyLauncher())
+    wf.initialize(device=ROCmDevice(0))
+    wf.run()
+    assert wf.is_finished
+    # prototype data with 10 classes: must be well below chance (90%)
+    assert wf.decision.epoch_n_err_pt[TRAIN] < 60.0, \
+        wf.decision.epoch_n_err_pt

@@ -117,8 +117,7 @@ class GradientDescent(GradientDescentBase):
         ops.sgd_step(self.weights, self.gradient_weights.devmem,
                      self.accumulated_gradient_weights.devmem,
                      self.learning_rate, self.weights_decay, self.l1_vs_l2,
-                     self.gradient_moment, 1.0 / batch,
-                     transposed_copy=self.weights_transposed)
+                     self.gradient_moment, 1.0 / batch)
         self.weights.touch_device()
         if self.include_bias and self.bias is not None and self.bias:
             ops.sgd_step(self.bias, self.gradient_bias.devmem,
@@ -127,6 +126,12 @@ class GradientDescent(GradientDescentBase):
                          self.l1_vs_l2_bias, self.gradient_moment_bias,
                          1.0 / batch)
             self.bias.touch_device()
+        self.post_update()
+
+    def post_update(self):
+        """Refresh the bf16ᵀ image consumed by fwd/dgrad."""
+        from veles.znicz_amd import ops
+        ops.refresh_fc_copies(self.weights)
 
 
 class GDTanh(GradientDescent):

@@ -27,7 +27,13 @@ class SyntheticClassLoader(FullBatchLoader):
 
     def load_data(self):
         rng = numpy.random.RandomState(self.seed)
-        protos = rng.uniform(-1, 1, (self.n_classes,) + self.sample_shape) \
+        sample_elems = int(numpy.prod(self.sample_shape))
+        # cap prototype memory: big models (AlexNet 1000-way synthetic)
+        # get a reduced prototype pool — the bench measures throughput,
+        # not accuracy; small models get one true prototype per class.
+        n_proto = self.n_classes if self.n_classes * sample_elems <= \
+            (1 << 26) else min(self.n_classes, 16)
+        protos = rng.uniform(-1, 1, (n_proto,) + self.sample_shape) \
             .astype(numpy.float32)
         total = self.n_test + self.n_valid + self.n_train
         data = numpy.zeros((total,) + self.sample_shape, numpy.float32)
@@ -35,7 +41,7 @@ class SyntheticClassLoader(FullBatchLoader):
         for i in range(total):
             c = i % self.n_classes
             labels[i] = c
-            data[i] = protos[c] + rng.normal(
+            data[i] = protos[c % n_proto] + rng.normal(
                 0, self.noise, self.sample_shape).astype(numpy.float32)
         self.original_data.mem = data
         self.original_labels.mem = labels

@@ -170,6 +170,10 @@ class GradientDescentBase(NNLayerBase):
             self.init_array(self.accumulated_gradient_bias,
                             device_dtype="f32")
 
+    def post_update(self):
+        """Refresh derived weight images after an SGD apply (overridden
+        by units that keep transposed/permuted bf16 copies)."""
+
     # -- numpy reference update --------------------------------------------
     def numpy_apply_gradient(self, w_arr, g, acc_arr, lr, decay, l1_vs_l2,
                              moment):

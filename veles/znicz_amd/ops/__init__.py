@@ -137,19 +137,47 @@ def col_sum(a, out):
     require().col_sum(a, out)
 
 
-def sgd_step(param_arr, grad, vel, lr, decay, l1_vs_l2, moment, scale,
-             transposed_copy=False):
-    """Fused SGD: master fp32 update + bf16 (+bf16ᵀ) refresh.
+def sgd_step(param_arr, grad, vel, lr, decay, l1_vs_l2, moment, scale):
+    """Fused SGD: master fp32 update + bf16 working-copy refresh.
 
     grad is the raw batch-summed fp32 gradient; ``scale`` (=1/batch or
-    1/(batch*world)) normalizes it inside the kernel.
+    1/(batch*world)) normalizes it inside the kernel.  Extra layout
+    copies (bf16ᵀ, conv dgrad image) are refreshed by the owning GD
+    unit's ``post_update()``.
     """
     master = param_arr.devmem
     bf16 = getattr(param_arr, "bf16", None)
-    bf16_t = getattr(param_arr, "bf16_t", None)
-    require().sgd_step(master, grad, vel, bf16, bf16_t,
+    require().sgd_step(master.view(-1), grad.view(-1), vel.view(-1),
+                       None if bf16 is None else bf16.view(-1), None,
                        float(lr), float(decay), float(l1_vs_l2),
                        float(moment), float(scale))
+
+
+def attach_conv_copies(weights_arr):
+    """Conv weights [K,KY,KX,C]: bf16 GEMM image [K, KY·KX·C] and the
+    dgrad image [C,KY,KX,K] (= W permuted, no flip — the dgrad loader
+    indexes (kh,kw) directly)."""
+    import torch
+    w = weights_arr.devmem
+    if getattr(weights_arr, "bf16", None) is None:
+        weights_arr.bf16 = w.reshape(w.shape[0], -1).to(torch.bfloat16) \
+            .contiguous()
+        weights_arr.bf16_dgrad = w.permute(3, 1, 2, 0).contiguous().to(
+            torch.bfloat16)
+    return weights_arr
+
+
+def refresh_conv_copies(weights_arr):
+    import torch
+    w = weights_arr.devmem          # fp32 master, already updated
+    k = w.shape[0]
+    weights_arr.bf16_dgrad.copy_(
+        weights_arr.bf16.view(w.shape).permute(3, 1, 2, 0))
+
+
+def refresh_fc_copies(weights_arr):
+    """bf16ᵀ image from the (just-refreshed) bf16 working copy."""
+    require().transpose2d(weights_arr.bf16, weights_arr.bf16_t)
 
 
 def softmax_fwd(x, max_idx):
@@ -204,8 +232,9 @@ def dropout_fwd(x, out, mask, ratio, seed, counter):
                           int(counter))
 
 
-def dropout_bwd(err_out, mask, err_in):
-    require().dropout_bwd(err_out, mask, err_in)
+def dropout_bwd(err, mask, ratio):
+    """err ∘= mask * 1/(1-ratio), in place."""
+    require().dropout_bwd(err, mask, float(ratio))
 
 
 def transpose2d(a, out):

@@ -1,0 +1,279 @@
+// _mzk — Python bindings for the hand-written CDNA4 kernel set.
+// The MI355X-native replacement for the reference's opencl4py/cuda4py
+// binding layer: torch tensors in, kernel launches on the CURRENT torch
+// stream (so the DP engine's comm-stream overlap works unchanged).
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "hip/api.h"
+
+// from gemm.hip
+void gemm_nt_dense(const unsigned short*, const unsigned short*, void*,
+                   bool, const float*, const unsigned short*, int, int,
+                   int, int, hipStream_t);
+void gemm_tn_dense(const unsigned short*, const unsigned short*, float*,
+                   int, int, int, float, hipStream_t);
+void conv_fwd_gemm(const unsigned short*, const unsigned short*, void*,
+                   bool, const float*, const unsigned short*,
+                   const ConvDesc&, int, hipStream_t);
+void conv_dgrad_gemm(const unsigned short*, const unsigned short*,
+                     unsigned short*, const unsigned short*,
+                     const ConvDesc&, hipStream_t);
+void conv_wgrad_gemm(const unsigned short*, const unsigned short*, float*,
+                     const ConvDesc&, float, hipStream_t);
+// from elementwise.hip
+void act_fwd_launch(unsigned short*, long, int, hipStream_t);
+void act_bwd_launch(unsigned short*, const unsigned short*, long, int,
+                    hipStream_t);
+void sgd_launch(float*, const float*, float*, unsigned short*, long, float,
+                float, float, float, float, hipStream_t);
+void col_sum_launch(const unsigned short*, float*, int, int, hipStream_t);
+void transpose2d_launch(const unsigned short*, unsigned short*, int, int,
+                        hipStream_t);
+void dropout_fwd_launch(const unsigned short*, unsigned short*,
+                        unsigned char*, long, float, unsigned long long,
+                        hipStream_t);
+void dropout_bwd_launch(unsigned short*, const unsigned char*, long, float,
+                        hipStream_t);
+// from softmax.hip
+void softmax_launch(float*, int*, int, int, hipStream_t);
+void eval_softmax_launch(const float*, const int*, unsigned short*, int*,
+                         int, int, int, hipStream_t);
+void eval_mse_launch(const unsigned short*, const unsigned short*,
+                     unsigned short*, float*, float*, int, int, int, int,
+                     hipStream_t);
+// from pool_lrn.hip
+void pool_fwd_launch(const unsigned short*, unsigned short*, int*,
+                     const PoolDesc&, int, hipStream_t);
+void pool_bwd_launch(const unsigned short*, unsigned short*, const int*,
+                     const PoolDesc&, int, hipStream_t);
+void lrn_fwd_launch(const unsigned short*, unsigned short*, float*, long,
+                    int, float, float, float, int, hipStream_t);
+void lrn_bwd_launch(const unsigned short*, const unsigned short*,
+                    const float*, unsigned short*, long, int, float, float,
+                    int, hipStream_t);
+
+namespace {
+
+hipStream_t cur_stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+const unsigned short* ubf(const torch::Tensor& t) {
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16, "expected bf16");
+  TORCH_CHECK(t.is_contiguous(), "expected contiguous");
+  return reinterpret_cast<const unsigned short*>(t.data_ptr());
+}
+
+unsigned short* ubf_mut(torch::Tensor& t) {
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16, "expected bf16");
+  TORCH_CHECK(t.is_contiguous(), "expected contiguous");
+  return reinterpret_cast<unsigned short*>(t.data_ptr());
+}
+
+// persistent 1 KiB zero page for padded/out-of-range DMA reads
+const unsigned short* zero_page(const torch::Device& dev) {
+  static torch::Tensor zp;
+  if (!zp.defined() || zp.device() != dev)
+    zp = torch::zeros({512}, torch::TensorOptions()
+                                 .dtype(torch::kBFloat16)
+                                 .device(dev));
+  return ubf(zp);
+}
+
+ConvDesc make_conv_desc(const std::vector<int64_t>& v) {
+  TORCH_CHECK(v.size() == 13, "conv desc needs 13 ints");
+  ConvDesc d{(int)v[0], (int)v[1], (int)v[2], (int)v[3], (int)v[4],
+             (int)v[5], (int)v[6], (int)v[7], (int)v[8], (int)v[9],
+             (int)v[10], (int)v[11], (int)v[12]};
+  return d;
+}
+
+void gemm_nt(torch::Tensor a, torch::Tensor b, torch::Tensor out,
+             c10::optional<torch::Tensor> bias, int64_t act) {
+  const int M = a.size(0), K = a.size(1), N = b.size(0);
+  TORCH_CHECK(b.size(1) == K, "gemm_nt: K mismatch");
+  TORCH_CHECK(out.size(0) == M && out.size(1) == N, "gemm_nt: out shape");
+  TORCH_CHECK(K % 8 == 0, "gemm_nt: K must be a multiple of 8 (pad)");
+  const float* bp = nullptr;
+  if (bias.has_value()) {
+    TORCH_CHECK(bias->scalar_type() == torch::kFloat32, "bias fp32");
+    bp = bias->data_ptr<float>();
+  }
+  bool f32out = out.scalar_type() == torch::kFloat32;
+  void* op = f32out ? (void*)out.data_ptr<float>()
+                    : (void*)ubf_mut(out);
+  gemm_nt_dense(ubf(a), ubf(b), op, f32out, bp, zero_page(a.device()), M,
+                N, K, (int)act, cur_stream());
+}
+
+void gemm_tn(torch::Tensor a, torch::Tensor b, torch::Tensor out,
+             double beta) {
+  // out[M,N] = aᵀ·b with a[K,M], b[K,N]
+  const int K = a.size(0), M = a.size(1), N = b.size(1);
+  TORCH_CHECK(b.size(0) == K, "gemm_tn: K mismatch");
+  TORCH_CHECK(out.scalar_type() == torch::kFloat32, "gemm_tn: fp32 out");
+  TORCH_CHECK(out.numel() == (int64_t)M * N, "gemm_tn: out shape");
+  gemm_tn_dense(ubf(a), ubf(b), out.data_ptr<float>(), M, N, K,
+                (float)beta, cur_stream());
+}
+
+void conv_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor out,
+              c10::optional<torch::Tensor> bias,
+              std::vector<int64_t> desc, int64_t act) {
+  ConvDesc d = make_conv_desc(desc);
+  const float* bp = nullptr;
+  if (bias.has_value()) bp = bias->data_ptr<float>();
+  bool f32out = out.scalar_type() == torch::kFloat32;
+  void* op = f32out ? (void*)out.data_ptr<float>()
+                    : (void*)ubf_mut(out);
+  conv_fwd_gemm(ubf(x), ubf(w), op, f32out, bp, zero_page(x.device()), d,
+                (int)act, cur_stream());
+}
+
+void conv_dgrad(torch::Tensor eo, torch::Tensor wdg, torch::Tensor ei,
+                std::vector<int64_t> desc) {
+  ConvDesc d = make_conv_desc(desc);
+  conv_dgrad_gemm(ubf(eo), ubf(wdg), ubf_mut(ei), zero_page(eo.device()),
+                  d, cur_stream());
+}
+
+void conv_wgrad(torch::Tensor eo, torch::Tensor x, torch::Tensor gw,
+                std::vector<int64_t> desc, double beta) {
+  ConvDesc d = make_conv_desc(desc);
+  conv_wgrad_gemm(ubf(eo), ubf(x), gw.data_ptr<float>(), d, (float)beta,
+                  cur_stream());
+}
+
+void act_fwd(torch::Tensor x, int64_t act) {
+  act_fwd_launch(ubf_mut(x), x.numel(), (int)act, cur_stream());
+}
+
+void act_bwd(torch::Tensor err, torch::Tensor y, int64_t act) {
+  TORCH_CHECK(err.numel() == y.numel(), "act_bwd size");
+  act_bwd_launch(ubf_mut(err), ubf(y), err.numel(), (int)act,
+                 cur_stream());
+}
+
+void sgd_step(torch::Tensor w, torch::Tensor g, torch::Tensor vel,
+              c10::optional<torch::Tensor> wb,
+              c10::optional<torch::Tensor> wbt, double lr, double decay,
+              double l1, double moment, double scale) {
+  unsigned short* wbp = nullptr;
+  if (wb.has_value()) wbp = ubf_mut(*wb);
+  sgd_launch(w.data_ptr<float>(), g.data_ptr<float>(),
+             vel.data_ptr<float>(), wbp, w.numel(), (float)lr,
+             (float)decay, (float)l1, (float)moment, (float)scale,
+             cur_stream());
+  if (wbt.has_value()) {
+    TORCH_CHECK(wb.has_value() && w.dim() == 2, "wbt needs 2-D wb");
+    transpose2d_launch(ubf(*wb), ubf_mut(*wbt), w.size(0), w.size(1),
+                       cur_stream());
+  }
+}
+
+void col_sum(torch::Tensor a, torch::Tensor out) {
+  col_sum_launch(ubf(a), out.data_ptr<float>(), a.size(0), a.size(1),
+                 cur_stream());
+}
+
+void transpose2d(torch::Tensor a, torch::Tensor out) {
+  transpose2d_launch(ubf(a), ubf_mut(out), a.size(0), a.size(1),
+                     cur_stream());
+}
+
+void softmax_fwd(torch::Tensor x, torch::Tensor max_idx) {
+  softmax_launch(x.data_ptr<float>(), max_idx.data_ptr<int>(), x.size(0),
+                 x.size(1), cur_stream());
+}
+
+void eval_softmax(torch::Tensor probs, torch::Tensor labels,
+                  torch::Tensor err, torch::Tensor n_err, int64_t batch) {
+  eval_softmax_launch(probs.data_ptr<float>(), labels.data_ptr<int>(),
+                      ubf_mut(err), n_err.data_ptr<int>(), probs.size(0),
+                      probs.size(1), (int)batch, cur_stream());
+}
+
+void eval_mse(torch::Tensor y, torch::Tensor t, torch::Tensor err,
+              torch::Tensor mse, torch::Tensor metrics, int64_t batch,
+              int64_t root) {
+  eval_mse_launch(ubf(y), ubf(t), ubf_mut(err), mse.data_ptr<float>(),
+                  metrics.data_ptr<float>(), y.size(0), y.size(1),
+                  (int)batch, (int)root, cur_stream());
+}
+
+void pool_fwd(torch::Tensor x, torch::Tensor y, torch::Tensor offs,
+              std::vector<int64_t> v, int64_t mode) {
+  TORCH_CHECK(v.size() == 11, "pool desc needs 11 ints");
+  PoolDesc d{(int)v[0], (int)v[1], (int)v[2], (int)v[3], (int)v[4],
+             (int)v[5], (int)v[6], (int)v[7], (int)v[8], (int)v[9],
+             (unsigned long long)v[10]};
+  pool_fwd_launch(ubf(x), ubf_mut(y), offs.data_ptr<int>(), d, (int)mode,
+                  cur_stream());
+}
+
+void pool_bwd(torch::Tensor ey, torch::Tensor ex, torch::Tensor offs,
+              std::vector<int64_t> v, int64_t mode) {
+  TORCH_CHECK(v.size() == 11, "pool desc needs 11 ints");
+  PoolDesc d{(int)v[0], (int)v[1], (int)v[2], (int)v[3], (int)v[4],
+             (int)v[5], (int)v[6], (int)v[7], (int)v[8], (int)v[9],
+             (unsigned long long)v[10]};
+  pool_bwd_launch(ubf(ey), ubf_mut(ex), offs.data_ptr<int>(), d, (int)mode,
+                  cur_stream());
+}
+
+void lrn_fwd(torch::Tensor x, torch::Tensor y, torch::Tensor denom,
+             double alpha, double beta, double k, int64_t win) {
+  const int C = x.size(-1);
+  lrn_fwd_launch(ubf(x), ubf_mut(y), denom.data_ptr<float>(),
+                 x.numel() / C, C, (float)alpha, (float)beta, (float)k,
+                 (int)win, cur_stream());
+}
+
+void lrn_bwd(torch::Tensor ey, torch::Tensor x, torch::Tensor y,
+             torch::Tensor denom, torch::Tensor ex, double alpha,
+             double beta, double k, int64_t win) {
+  const int C = x.size(-1);
+  (void)y; (void)k;
+  lrn_bwd_launch(ubf(ey), ubf(x), denom.data_ptr<float>(), ubf_mut(ex),
+                 x.numel() / C, C, (float)alpha, (float)beta, (int)win,
+                 cur_stream());
+}
+
+void dropout_fwd(torch::Tensor x, torch::Tensor y, torch::Tensor mask,
+                 double ratio, int64_t seed, int64_t counter) {
+  unsigned long long s =
+      (unsigned long long)seed * 0x100000001B3ull + (unsigned long long)counter;
+  dropout_fwd_launch(ubf(x), ubf_mut(y), mask.data_ptr<unsigned char>(),
+                     x.numel(), (float)ratio, s, cur_stream());
+}
+
+void dropout_bwd(torch::Tensor err, torch::Tensor mask, double ratio) {
+  dropout_bwd_launch(ubf_mut(err), mask.data_ptr<unsigned char>(),
+                     err.numel(), (float)ratio, cur_stream());
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "veles.znicz_amd hand-written CDNA4 (gfx950) kernels";
+  m.def("gemm_nt", &gemm_nt, "C=act(A·Bᵀ+bias), bf16 MFMA");
+  m.def("gemm_tn", &gemm_tn, "C=AᵀB (K-major wgrad core)");
+  m.def("conv_fwd", &conv_fwd, "implicit-im2col conv forward");
+  m.def("conv_dgrad", &conv_dgrad, "conv backward-data (gather)");
+  m.def("conv_wgrad", &conv_wgrad, "conv backward-weights (K-major)");
+  m.def("act_fwd", &act_fwd);
+  m.def("act_bwd", &act_bwd);
+  m.def("sgd_step", &sgd_step);
+  m.def("col_sum", &col_sum);
+  m.def("transpose2d", &transpose2d);
+  m.def("softmax_fwd", &softmax_fwd);
+  m.def("eval_softmax", &eval_softmax);
+  m.def("eval_mse", &eval_mse);
+  m.def("pool_fwd", &pool_fwd);
+  m.def("pool_bwd", &pool_bwd);
+  m.def("lrn_fwd", &lrn_fwd);
+  m.def("lrn_bwd", &lrn_bwd);
+  m.def("dropout_fwd", &dropout_fwd);
+  m.def("dropout_bwd", &dropout_bwd);
+}

@@ -1,0 +1,185 @@
+// Elementwise CDNA4 kernels: activations, dropout (counter-based PRNG),
+// fused SGD apply (fp32 master + bf16 refresh), column reduce, transpose.
+// Replaces reference znicz/ocl/{err_output.cl, dropout.cl,
+// gradient_descent.cl weights/bias_update, random.cl} — redesigned:
+// vectorized bf16x8 loads (wave64 coalescing), no PRNG state buffers,
+// SGD applied once per (all-reduced) gradient bucket.
+#include "common.h"
+
+// -- activation forward/backward (in-place on bf16 tensors) -----------------
+
+template <int A>
+__global__ void act_fwd_k(unsigned short* x, long n) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long p = i * 8; p < n; p += stride * 8) {
+    unsigned short v[8];
+    const long todo = min((long)8, n - p);
+    if (todo == 8) {
+      *(short8*)v = *(const short8*)(x + p);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) v[j] = f2bf(act_fwd<A>(bf2f(v[j])));
+      *(short8*)(x + p) = *(const short8*)v;
+    } else {
+      for (int j = 0; j < todo; ++j)
+        x[p + j] = f2bf(act_fwd<A>(bf2f(x[p + j])));
+    }
+  }
+}
+
+// err ∘= act'(y)
+template <int A>
+__global__ void act_bwd_k(unsigned short* err, const unsigned short* y,
+                          long n) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long p = i * 8; p < n; p += stride * 8) {
+    const long todo = min((long)8, n - p);
+    if (todo == 8) {
+      unsigned short e[8], v[8];
+      *(short8*)e = *(const short8*)(err + p);
+      *(short8*)v = *(const short8*)(y + p);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        e[j] = f2bf(bf2f(e[j]) * act_dy<A>(bf2f(v[j])));
+      *(short8*)(err + p) = *(const short8*)e;
+    } else {
+      for (int j = 0; j < todo; ++j)
+        err[p + j] = f2bf(bf2f(err[p + j]) * act_dy<A>(bf2f(y[p + j])));
+    }
+  }
+}
+
+void act_fwd_launch(unsigned short* x, long n, int act, hipStream_t s) {
+  int blocks = (int)min((n + 8 * 256 - 1) / (8 * 256), (long)2048);
+#define GO(A) hipLaunchKernelGGL(act_fwd_k<A>, dim3(blocks), dim3(256), 0, \
+                                 s, x, n)
+  ACT_SWITCH(act, GO);
+#undef GO
+  HIP_CHECK_LAST();
+}
+
+void act_bwd_launch(unsigned short* err, const unsigned short* y, long n,
+                    int act, hipStream_t s) {
+  int blocks = (int)min((n + 8 * 256 - 1) / (8 * 256), (long)2048);
+#define GO(A) hipLaunchKernelGGL(act_bwd_k<A>, dim3(blocks), dim3(256), 0, \
+                                 s, err, y, n)
+  ACT_SWITCH(act, GO);
+#undef GO
+  HIP_CHECK_LAST();
+}
+
+// -- fused SGD: master fp32 update + bf16 working-copy refresh --------------
+//  g = grad*scale + decay*(l1*sign(w) + (1-l1)*w)
+//  v = moment*v + lr*g ;  w -= v ;  bf16[i] = w
+__global__ void sgd_k(float* __restrict__ w, const float* __restrict__ g,
+                      float* __restrict__ vel,
+                      unsigned short* __restrict__ wb, long n, float lr,
+                      float decay, float l1, float moment, float scale) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long p = i; p < n; p += stride) {
+    float wi = w[p];
+    float l1l2 = l1 * (wi > 0.f ? 1.f : (wi < 0.f ? -1.f : 0.f)) +
+                 (1.f - l1) * wi;
+    float gi = g[p] * scale + decay * l1l2;
+    float v = moment * vel[p] + lr * gi;
+    if (moment != 0.f) vel[p] = v;
+    wi -= v;
+    w[p] = wi;
+    if (wb) wb[p] = f2bf(wi);
+  }
+}
+
+void sgd_launch(float* w, const float* g, float* vel, unsigned short* wb,
+                long n, float lr, float decay, float l1, float moment,
+                float scale, hipStream_t s) {
+  int blocks = (int)min((n + 255) / 256, (long)2048);
+  hipLaunchKernelGGL(sgd_k, dim3(blocks), dim3(256), 0, s, w, g, vel, wb,
+                     n, lr, decay, l1, moment, scale);
+  HIP_CHECK_LAST();
+}
+
+// -- column sum (bias gradient): out[N] = Σ_m A[M][N], bf16 -> fp32 --------
+__global__ void col_sum_k(const unsigned short* __restrict__ a,
+                          float* __restrict__ out, int M, int N) {
+  int col = blockIdx.x * blockDim.x + threadIdx.x;
+  if (col >= N) return;
+  float s = 0.f;
+  for (long m = 0; m < M; ++m) s += bf2f(a[m * (long)N + col]);
+  out[col] = s;
+}
+
+void col_sum_launch(const unsigned short* a, float* out, int M, int N,
+                    hipStream_t s) {
+  hipLaunchKernelGGL(col_sum_k, dim3((N + 255) / 256), dim3(256), 0, s, a,
+                     out, M, N);
+  HIP_CHECK_LAST();
+}
+
+// -- bf16 2-D transpose (LDS 32x32 tiles, +1 pad) ---------------------------
+__global__ void transpose2d_k(const unsigned short* __restrict__ in,
+                              unsigned short* __restrict__ out, int R,
+                              int C) {
+  __shared__ unsigned short tile[32][33];
+  int c0 = blockIdx.x * 32, r0 = blockIdx.y * 32;
+  int tx = threadIdx.x & 31, ty = threadIdx.x >> 5;   // 256 thr: 8 rows
+  for (int i = 0; i < 32; i += 8) {
+    int r = r0 + ty + i, c = c0 + tx;
+    tile[ty + i][tx] = (r < R && c < C) ? in[(long)r * C + c] : 0;
+  }
+  __syncthreads();
+  for (int i = 0; i < 32; i += 8) {
+    int r = c0 + ty + i, c = r0 + tx;   // transposed coords
+    if (r < C && c < R) out[(long)r * R + c] = tile[tx][ty + i];
+  }
+}
+
+void transpose2d_launch(const unsigned short* in, unsigned short* out,
+                        int R, int C, hipStream_t s) {
+  dim3 grid((C + 31) / 32, (R + 31) / 32);
+  hipLaunchKernelGGL(transpose2d_k, grid, dim3(256), 0, s, in, out, R, C);
+  HIP_CHECK_LAST();
+}
+
+// -- dropout ---------------------------------------------------------------
+__global__ void dropout_fwd_k(const unsigned short* __restrict__ x,
+                              unsigned short* __restrict__ y,
+                              unsigned char* __restrict__ mask, long n,
+                              float ratio, float scale,
+                              unsigned long long seed) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long p = i; p < n; p += stride) {
+    unsigned char keep = uniform01(seed, (unsigned long long)p) >= ratio;
+    mask[p] = keep;
+    y[p] = keep ? f2bf(bf2f(x[p]) * scale) : (unsigned short)0;
+  }
+}
+
+__global__ void dropout_bwd_k(unsigned short* __restrict__ err,
+                              const unsigned char* __restrict__ mask,
+                              long n, float scale) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long p = i; p < n; p += stride)
+    err[p] = mask[p] ? f2bf(bf2f(err[p]) * scale) : (unsigned short)0;
+}
+
+void dropout_fwd_launch(const unsigned short* x, unsigned short* y,
+                        unsigned char* mask, long n, float ratio,
+                        unsigned long long seed, hipStream_t s) {
+  int blocks = (int)min((n + 255) / 256, (long)4096);
+  float scale = 1.f / (1.f - ratio);
+  hipLaunchKernelGGL(dropout_fwd_k, dim3(blocks), dim3(256), 0, s, x, y,
+                     mask, n, ratio, scale, seed);
+  HIP_CHECK_LAST();
+}
+
+void dropout_bwd_launch(unsigned short* err, const unsigned char* mask,
+                        long n, float ratio, hipStream_t s) {
+  int blocks = (int)min((n + 255) / 256, (long)4096);
+  hipLaunchKernelGGL(dropout_bwd_k, dim3(blocks), dim3(256), 0, s, err,
+                     mask, n, 1.f / (1.f - ratio));
+  HIP_CHECK_LAST();
+}

@@ -1,0 +1,407 @@
+#include "hip/hip_runtime.h"
+// MFMA bf16 GEMM cores for gfx950 — the compute heart of veles.znicz_amd.
+//
+// Replaces the reference's tiled OpenCL/CUDA GEMM template
+// (znicz/ocl/matrix_multiplication.cl + all2all.cl/conv.cl includes) with
+// two CDNA4-native cores:
+//
+//   core NT:  C[M,N] = act(A[M,K] · B[N,K]ᵀ + bias)
+//     both operands K-innermost (the natural layout of FC fwd, FC dgrad
+//     with a persistent transposed-weight copy, and implicit-im2col conv
+//     fwd/dgrad).  128×128 macro-tile, BK=64, 256 threads (4 waves in
+//     2×2), v_mfma_f32_16x16x32_bf16, double-buffered LDS filled by
+//     global_load_lds_dwordx4 (async DMA, 16 B/lane), im2col unrolling
+//     done IN the per-lane DMA source address (never materialized in
+//     HBM; padding reads a zero page).
+//
+//   core TN:  C[M,N](fp32) = beta·C + Σ_k A[k][m]·B[k][n]
+//     both operands K-major (batch-major) — the wgrad shape for FC and
+//     conv (k = N·OH·OW).  Staged by coalesced element loads packed to
+//     short8 + swizzled ds_write_b128 (XOR (row&7)<<4 — bank-conflict
+//     free), fragments read back with the matching XOR.
+//
+// Wave64 / LDS-tiled throughout; nothing here is a CUDA translation.
+#include <stdexcept>
+#include "common.h"
+
+#define BM 128
+#define BN 128
+#define BK 64
+#define NT_LDS_A (BM * BK * 2)          // 16 KiB
+#define NT_LDS_BUF (NT_LDS_A * 2)       // A + B per buffer
+#define NT_LDS_TOTAL (NT_LDS_BUF * 2)   // double-buffered = 64 KiB
+
+// ---------------------------------------------------------------------------
+// operand loaders (per-lane global source address generators)
+// ---------------------------------------------------------------------------
+
+// dense row-major [rows][ld], K-innermost; k past Klim reads the zero page
+struct DenseNT {
+  const unsigned short* base;
+  const unsigned short* zero;
+  long ld;
+  int Klim;
+  __device__ const void* addr(int row, int k8) const {
+    int k = k8 * 8;
+    return k < Klim ? (const void*)(base + (long)row * ld + k)
+                    : (const void*)zero;
+  }
+};
+
+// implicit im2col over NHWC input: row=(n,oh,ow), kdim=(kh,kw,c) c-inner.
+// Requires C % 8 == 0 so each 16-B chunk stays inside one pixel's C-run.
+struct Im2colNT {
+  const unsigned short* x;
+  const unsigned short* zero;
+  ConvDesc d;
+  __device__ const void* addr(int row, int k8) const {
+    int k = k8 * 8;
+    int c = k % d.C, t = k / d.C;
+    int kw = t % d.KW, kh = t / d.KW;
+    int ow = row % d.OW;
+    t = row / d.OW;
+    int oh = t % d.OH, n = t / d.OH;
+    int ih = oh * d.SY - d.PT + kh;
+    int iw = ow * d.SX - d.PL + kw;
+    if (kh >= d.KH || (unsigned)ih >= (unsigned)d.H ||
+        (unsigned)iw >= (unsigned)d.W)
+      return zero;
+    return x + (((long)n * d.H + ih) * d.W + iw) * d.C + c;
+  }
+};
+
+// conv dgrad gather over NHWC err_output: row=(n,ih,iw) of the FORWARD
+// input, kdim=(kh,kw,co) co-inner; strided positions read the zero page.
+struct DgradNT {
+  const unsigned short* eo;
+  const unsigned short* zero;
+  ConvDesc d;
+  __device__ const void* addr(int row, int k8) const {
+    int k = k8 * 8;
+    int co = k % d.K, t = k / d.K;
+    int kw = t % d.KW, kh = t / d.KW;
+    int iw = row % d.W;
+    t = row / d.W;
+    int ih = t % d.H, n = t / d.H;
+    int ohn = ih + d.PT - kh;
+    int own = iw + d.PL - kw;
+    if (kh >= d.KH || ohn < 0 || own < 0 || ohn % d.SY || own % d.SX)
+      return zero;
+    int oh = ohn / d.SY, ow = own / d.SX;
+    if (oh >= d.OH || ow >= d.OW) return zero;
+    return eo + (((long)n * d.OH + oh) * d.OW + ow) * d.K + co;
+  }
+};
+
+// ---------------------------------------------------------------------------
+// core NT
+// ---------------------------------------------------------------------------
+
+template <typename LA, typename LB, typename OUT_T, int ACT>
+__global__ __launch_bounds__(256, 2) void gemm_nt_k(
+    LA la, LB lb, OUT_T* __restrict__ C, const float* __restrict__ bias,
+    int M, int N, int K, int tiles_n) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int tile = blockIdx.x;
+  const int tm = tile / tiles_n, tn = tile % tiles_n;
+  const int m0 = tm * BM, n0 = tn * BN;
+  const int wr = wave >> 1, wc = wave & 1;   // 2x2 waves of 64x64
+
+  const int nk = (K + BK - 1) / BK;
+
+  auto stage = [&](int buf, int kt) {
+    char* As = smem + buf * NT_LDS_BUF;
+    char* Bs = As + NT_LDS_A;
+    const int k8base = kt * (BK / 8);
+    const int rl = lane >> 3;          // row within 8-row chunk
+    const int k8 = lane & 7;           // 16-B chunk within the 64-k row
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int chunk = wave * 4 + i;  // 16 chunks x 8 rows = 128 rows
+      const int row = chunk * 8 + rl;
+      const void* sa = la.addr(min(m0 + row, M - 1), k8base + k8);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)sa,
+          (__attribute__((address_space(3))) void*)(As + chunk * 1024),
+          16, 0, 0);
+      const void* sb = lb.addr(min(n0 + row, N - 1), k8base + k8);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)sb,
+          (__attribute__((address_space(3))) void*)(Bs + chunk * 1024),
+          16, 0, 0);
+    }
+  };
+
+  f32x4 acc[4][4] = {};
+
+  stage(0, 0);
+  __syncthreads();
+  int buf = 0;
+  for (int t = 0; t < nk; ++t) {
+    if (t + 1 < nk) stage(buf ^ 1, t + 1);
+    const char* As = smem + buf * NT_LDS_BUF;
+    const char* Bs = As + NT_LDS_A;
+    const int fr = lane & 15;          // fragment row/col
+    const int fo = (lane >> 4) * 16;   // k-octet byte offset
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      short8 af[4], bf[4];
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+        af[i] = *(const short8*)(As + (wr * 64 + i * 16 + fr) * 128 +
+                                 ks * 64 + fo);
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+        bf[i] = *(const short8*)(Bs + (wc * 64 + i * 16 + fr) * 128 +
+                                 ks * 64 + fo);
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+    }
+    __syncthreads();
+    buf ^= 1;
+  }
+
+  // epilogue: bias + activation + store (row = (l>>4)*4+r, col = l&15)
+  const int colw = n0 + wc * 64 + (lane & 15);
+#pragma unroll
+  for (int ni = 0; ni < 4; ++ni) {
+    const int col = colw + ni * 16;
+    if (col >= N) continue;
+    const float bv = bias ? bias[col] : 0.f;
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi) {
+      const int rbase = m0 + wr * 64 + mi * 16 + ((lane >> 4) << 2);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = rbase + r;
+        if (row >= M) continue;
+        float v = act_fwd<ACT>(acc[mi][ni][r] + bv);
+        if constexpr (sizeof(OUT_T) == 2)
+          C[(long)row * N + col] = (OUT_T)f2bf(v);
+        else
+          C[(long)row * N + col] = (OUT_T)v;
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// core TN (K-major operands; fp32 out += )
+// ---------------------------------------------------------------------------
+
+// dense K-major: element (k, m) at base[k*ld + m]; OOB -> 0
+struct DenseTN {
+  const unsigned short* base;
+  long ld;
+  int Klim, Mlim;
+  __device__ unsigned short load(int k, int m) const {
+    bool ok = (k < Klim) & (m < Mlim);
+    long off = ok ? (long)k * ld + m : 0;
+    unsigned short v = base[off];
+    return ok ? v : (unsigned short)0;
+  }
+};
+
+// im2col K-major: (k = n*OH*OW + oh*OW + ow, column n = (kh,kw,c))
+struct Im2colTN {
+  const unsigned short* x;
+  ConvDesc d;
+  int Klim, Mlim;
+  __device__ unsigned short load(int k, int m) const {
+    bool ok = (k < Klim) & (m < Mlim);
+    int c = m % d.C, t = m / d.C;
+    int kw = t % d.KW, kh = t / d.KW;
+    int ow = k % d.OW;
+    t = k / d.OW;
+    int oh = t % d.OH, n = t / d.OH;
+    int ih = oh * d.SY - d.PT + kh;
+    int iw = ow * d.SX - d.PL + kw;
+    ok &= (kh < d.KH) & ((unsigned)ih < (unsigned)d.H) &
+          ((unsigned)iw < (unsigned)d.W);
+    long off = ok ? (((long)n * d.H + ih) * d.W + iw) * d.C + c : 0;
+    unsigned short v = x[off];
+    return ok ? v : (unsigned short)0;
+  }
+};
+
+template <typename LA, typename LB>
+__global__ __launch_bounds__(256, 2) void gemm_tn_k(
+    LA la, LB lb, float* __restrict__ C, int M, int N, int K, float beta,
+    int tiles_n) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int tile = blockIdx.x;
+  const int tm = tile / tiles_n, tn = tile % tiles_n;
+  const int m0 = tm * BM, n0 = tn * BN;
+  const int wr = wave >> 1, wc = wave & 1;
+  const int nk = (K + BK - 1) / BK;
+
+  // stage one operand's [BK][128] K-major slab as a swizzled [128][BK]
+  // image: thread covers (col, k-octet) chunks; 8 coalesced u16 loads
+  // pack to short8; one swizzled ds_write_b128.
+  auto stage_op = [&](char* dst, const auto& ld_, int kt, int lim0,
+                      int is_b) {
+    const int k0 = kt * BK;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int chunk = i * 256 + tid;        // 1024 chunks
+      const int col = chunk & 127;            // m (or n) within tile
+      const int ko = chunk >> 7;              // k-octet 0..7
+      unsigned short v[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        v[j] = ld_.load(k0 + ko * 8 + j, lim0 + col);
+      const int slot = ko ^ (col & 7);        // bank-conflict-free
+      *(short8*)(dst + col * 128 + slot * 16) = *(short8*)v;
+    }
+    (void)is_b;
+  };
+
+  f32x4 acc[4][4] = {};
+  char* As0 = smem;
+  char* Bs0 = smem + NT_LDS_A;
+
+  for (int t = 0; t < nk; ++t) {
+    const int buf = t & 1;
+    char* As = As0 + buf * NT_LDS_BUF;
+    char* Bs = Bs0 + buf * NT_LDS_BUF;
+    stage_op(As, la, t, m0, 0);
+    stage_op(Bs, lb, t, n0, 1);
+    __syncthreads();
+    const int fr = lane & 15;
+    const int fo = lane >> 4;                 // k-octet 0..3 (+4 for ks=1)
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      short8 af[4], bf[4];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int row = wr * 64 + i * 16 + fr;
+        const int slot = (ks * 4 + fo) ^ (row & 7);
+        af[i] = *(const short8*)(As + row * 128 + slot * 16);
+      }
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int row = wc * 64 + i * 16 + fr;
+        const int slot = (ks * 4 + fo) ^ (row & 7);
+        bf[i] = *(const short8*)(Bs + row * 128 + slot * 16);
+      }
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  const int colw = n0 + wc * 64 + (lane & 15);
+#pragma unroll
+  for (int ni = 0; ni < 4; ++ni) {
+    const int col = colw + ni * 16;
+    if (col >= N) continue;
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi) {
+      const int rbase = m0 + wr * 64 + mi * 16 + ((lane >> 4) << 2);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = rbase + r;
+        if (row >= M) continue;
+        float* p = C + (long)row * N + col;
+        *p = (beta == 0.f ? 0.f : beta * *p) + acc[mi][ni][r];
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// launch helpers (called from bindings.cpp)
+// ---------------------------------------------------------------------------
+
+static inline int cdiv(int a, int b) { return (a + b - 1) / b; }
+
+template <typename LA, typename LB, typename OUT_T>
+void launch_gemm_nt(const LA& la, const LB& lb, OUT_T* C, const float* bias,
+                    int M, int N, int K, int act, hipStream_t stream) {
+  const int tn = cdiv(N, BN);
+  dim3 grid(cdiv(M, BM) * tn), block(256);
+#define GO(A)                                                                \
+  hipLaunchKernelGGL((gemm_nt_k<LA, LB, OUT_T, A>), grid, block,             \
+                     NT_LDS_TOTAL, stream, la, lb, C, bias, M, N, K, tn)
+  ACT_SWITCH(act, GO);
+#undef GO
+  HIP_CHECK_LAST();
+}
+
+template <typename LA, typename LB>
+void launch_gemm_tn(const LA& la, const LB& lb, float* C, int M, int N,
+                    int K, float beta, hipStream_t stream) {
+  const int tn = cdiv(N, BN);
+  dim3 grid(cdiv(M, BM) * tn), block(256);
+  hipLaunchKernelGGL((gemm_tn_k<LA, LB>), grid, block, NT_LDS_TOTAL, stream,
+                     la, lb, C, M, N, K, beta, tn);
+  HIP_CHECK_LAST();
+}
+
+// explicit entry points used by bindings.cpp --------------------------------
+
+void gemm_nt_dense(const unsigned short* A, const unsigned short* B,
+                   void* C, bool c_is_f32, const float* bias,
+                   const unsigned short* zero, int M, int N, int K,
+                   int act, hipStream_t s) {
+  DenseNT la{A, zero, K, K}, lb{B, zero, K, K};
+  if (c_is_f32)
+    launch_gemm_nt(la, lb, (float*)C, bias, M, N, K, act, s);
+  else
+    launch_gemm_nt(la, lb, (unsigned short*)C, bias, M, N, K, act, s);
+}
+
+void gemm_tn_dense(const unsigned short* A, const unsigned short* B,
+                   float* C, int M, int N, int K, float beta,
+                   hipStream_t s) {
+  DenseTN la{A, M, K, M}, lb{B, N, K, N};
+  launch_gemm_tn(la, lb, C, M, N, K, beta, s);
+}
+
+void conv_fwd_gemm(const unsigned short* x, const unsigned short* w,
+                   void* out, bool out_f32, const float* bias,
+                   const unsigned short* zero, const ConvDesc& d, int act,
+                   hipStream_t s) {
+  const int M = d.N * d.OH * d.OW;
+  const int K = d.KH * d.KW * d.C;
+  Im2colNT la{x, zero, d};
+  DenseNT lb{w, zero, K, K};
+  if (out_f32)
+    launch_gemm_nt(la, lb, (float*)out, bias, M, d.K, K, act, s);
+  else
+    launch_gemm_nt(la, lb, (unsigned short*)out, bias, M, d.K, K, act, s);
+}
+
+void conv_dgrad_gemm(const unsigned short* eo, const unsigned short* wdg,
+                     unsigned short* ei, const unsigned short* zero,
+                     const ConvDesc& d, hipStream_t s) {
+  const int M = d.N * d.H * d.W;       // input pixels
+  const int N = d.C;                   // input channels
+  const int K = d.KH * d.KW * d.K;     // co-inner gather dim
+  DgradNT la{eo, zero, d};
+  DenseNT lb{wdg, zero, K, K};         // wdg: [C][KH][KW][K]
+  launch_gemm_nt(la, lb, ei, (const float*)nullptr, M, N, K, ACT_LINEAR, s);
+}
+
+void conv_wgrad_gemm(const unsigned short* eo, const unsigned short* x,
+                     float* gw, const ConvDesc& d, float beta,
+                     hipStream_t s) {
+  const int Kdim = d.N * d.OH * d.OW;
+  const int M = d.K;                   // n_kernels
+  const int N = d.KH * d.KW * d.C;
+  DenseTN la{eo, d.K, Kdim, d.K};
+  Im2colTN lb{x, d, Kdim, N};
+  launch_gemm_tn(la, lb, gw, M, N, Kdim, beta, s);
+}

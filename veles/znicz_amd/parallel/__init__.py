@@ -159,8 +159,7 @@ class GradientReducer(object):
                 ops.sgd_step(u.weights, u.gradient_weights.devmem,
                              u.accumulated_gradient_weights.devmem,
                              u.learning_rate, u.weights_decay, u.l1_vs_l2,
-                             u.gradient_moment, scale,
-                             transposed_copy=u.weights_transposed)
+                             u.gradient_moment, scale)
                 u.weights.touch_device()
                 if u.include_bias and u.bias is not None and u.bias:
                     ops.sgd_step(u.bias, u.gradient_bias.devmem,
@@ -169,6 +168,7 @@ class GradientReducer(object):
                                  u.weights_decay_bias, u.l1_vs_l2_bias,
                                  u.gradient_moment_bias, scale)
                     u.bias.touch_device()
+                u.post_update()
             self._tail_event.record(self.comm_stream)
             self._have_tail = True
         self._pending_apply = []

@@ -22,6 +22,13 @@ from veles.znicz_amd.nn_units import (
 # import unit modules so their MAPPINGs register
 from veles.znicz_amd import all2all  # noqa: F401
 from veles.znicz_amd import gd  # noqa: F401
+from veles.znicz_amd import conv  # noqa: F401
+from veles.znicz_amd import gd_conv  # noqa: F401
+from veles.znicz_amd import pooling  # noqa: F401
+from veles.znicz_amd import gd_pooling  # noqa: F401
+from veles.znicz_amd import normalization  # noqa: F401
+from veles.znicz_amd import dropout  # noqa: F401
+from veles.znicz_amd import activation  # noqa: F401
 
 
 class CommSync(Unit):
@@ -184,9 +191,16 @@ class StandardWorkflow(StandardWorkflowBase):
             unit = self.backward_unit_class(tpe)(self, **gd_kw)
             unit.name = "gd%d_%s" % (i, tpe)
             unit.link_from(prev)
-            unit.link_attrs(fwd, "input", "output", "weights", "bias")
-            unit.include_bias = fwd.include_bias
-            unit.weights_transposed = fwd.weights_transposed
+            unit.forward_unit = fwd
+            unit.link_attrs(fwd, "input", "output")
+            if getattr(fwd, "has_weights", True):
+                unit.link_attrs(fwd, "weights", "bias")
+                unit.include_bias = fwd.include_bias
+                unit.weights_transposed = fwd.weights_transposed
+            else:
+                unit.include_bias = False
+            for extra in getattr(unit, "LINK_FROM_FORWARD", ()):
+                unit.link_attrs(fwd, extra)
             unit.link_attrs(self.loader, ("batch_size", "minibatch_size"))
             if i == len(self.layers) - 1:
                 unit.link_attrs(self.evaluator, "err_output")
