@@ -29,7 +29,9 @@ class FullBatchLoader(Loader):
         self.original_labels = Array()
         self.on_device = kwargs.get("on_device", True)
         self.data_dtype = kwargs.get("data_dtype", numpy.float32)
-        self.device_dtype = kwargs.get("device_dtype", None)
+        # device working dtype of activations: bf16 (host mirror stays
+        # fp32, so snapshots are fp32) unless overridden
+        self.device_dtype = kwargs.get("device_dtype", "bf16")
 
     def load_data(self):
         """Subclasses fill original_data/original_labels + class_lengths."""

@@ -205,8 +205,8 @@ def conv_dgrad(err_out, w_dgrad, err_in, conv_desc):
     require().conv_dgrad(err_out, w_dgrad, err_in, conv_desc)
 
 
-def conv_wgrad(err_out, x, grad_w, conv_desc):
-    require().conv_wgrad(err_out, x, grad_w, conv_desc)
+def conv_wgrad(err_out, x, grad_w, conv_desc, beta=0.0):
+    require().conv_wgrad(err_out, x, grad_w, conv_desc, float(beta))
 
 
 def pool_fwd(x, out, offs, desc, mode):

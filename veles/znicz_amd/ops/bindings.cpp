@@ -94,7 +94,6 @@ void gemm_nt(torch::Tensor a, torch::Tensor b, torch::Tensor out,
   const int M = a.size(0), K = a.size(1), N = b.size(0);
   TORCH_CHECK(b.size(1) == K, "gemm_nt: K mismatch");
   TORCH_CHECK(out.size(0) == M && out.size(1) == N, "gemm_nt: out shape");
-  TORCH_CHECK(K % 8 == 0, "gemm_nt: K must be a multiple of 8 (pad)");
   const float* bp = nullptr;
   if (bias.has_value()) {
     TORCH_CHECK(bias->scalar_type() == torch::kFloat32, "bias fp32");
@@ -230,9 +229,10 @@ void lrn_fwd(torch::Tensor x, torch::Tensor y, torch::Tensor denom,
                  (int)win, cur_stream());
 }
 
-void lrn_bwd(torch::Tensor ey, torch::Tensor x, torch::Tensor y,
-             torch::Tensor denom, torch::Tensor ex, double alpha,
-             double beta, double k, int64_t win) {
+void lrn_bwd(torch::Tensor ey, torch::Tensor x,
+             c10::optional<torch::Tensor> y, torch::Tensor denom,
+             torch::Tensor ex, double alpha, double beta, double k,
+             int64_t win) {
   const int C = x.size(-1);
   (void)y; (void)k;
   lrn_bwd_launch(ubf(ey), ubf(x), denom.data_ptr<float>(), ubf_mut(ex),

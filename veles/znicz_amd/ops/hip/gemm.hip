@@ -45,6 +45,10 @@ struct DenseNT {
     return k < Klim ? (const void*)(base + (long)row * ld + k)
                     : (const void*)zero;
   }
+  // single-element access for the K%8 tail chunk
+  __device__ unsigned short elem(int row, int k) const {
+    return k < Klim ? base[(long)row * ld + k] : (unsigned short)0;
+  }
 };
 
 // implicit im2col over NHWC input: row=(n,oh,ow), kdim=(kh,kw,c) c-inner.
@@ -66,6 +70,20 @@ struct Im2colNT {
         (unsigned)iw >= (unsigned)d.W)
       return zero;
     return x + (((long)n * d.H + ih) * d.W + iw) * d.C + c;
+  }
+  __device__ unsigned short elem(int row, int k) const {
+    // only hit when K%8 != 0 (conv K-dims are C-padded so never here)
+    int c = k % d.C, t = k / d.C;
+    int kw = t % d.KW, kh = t / d.KW;
+    int ow = row % d.OW;
+    t = row / d.OW;
+    int oh = t % d.OH, n = t / d.OH;
+    int ih = oh * d.SY - d.PT + kh;
+    int iw = ow * d.SX - d.PL + kw;
+    if (kh >= d.KH || (unsigned)ih >= (unsigned)d.H ||
+        (unsigned)iw >= (unsigned)d.W)
+      return 0;
+    return x[(((long)n * d.H + ih) * d.W + iw) * d.C + c];
   }
 };
 
@@ -89,6 +107,20 @@ struct DgradNT {
     int oh = ohn / d.SY, ow = own / d.SX;
     if (oh >= d.OH || ow >= d.OW) return zero;
     return eo + (((long)n * d.OH + oh) * d.OW + ow) * d.K + co;
+  }
+  __device__ unsigned short elem(int row, int k) const {
+    int co = k % d.K, t = k / d.K;
+    int kw = t % d.KW, kh = t / d.KW;
+    int iw = row % d.W;
+    t = row / d.W;
+    int ih = t % d.H, n = t / d.H;
+    int ohn = ih + d.PT - kh;
+    int own = iw + d.PL - kw;
+    if (kh >= d.KH || ohn < 0 || own < 0 || ohn % d.SY || own % d.SX)
+      return 0;
+    int oh = ohn / d.SY, ow = own / d.SX;
+    if (oh >= d.OH || ow >= d.OW) return 0;
+    return eo[(((long)n * d.OH + oh) * d.OW + ow) * d.K + co];
   }
 };
 
@@ -116,20 +148,35 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_k(
     const int k8base = kt * (BK / 8);
     const int rl = lane >> 3;          // row within 8-row chunk
     const int k8 = lane & 7;           // 16-B chunk within the 64-k row
+    const int kc0 = (k8base + k8) * 8;
+    const bool straddle = (kc0 < K) & (kc0 + 8 > K);  // K%8 tail chunk
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
       const int chunk = wave * 4 + i;  // 16 chunks x 8 rows = 128 rows
       const int row = chunk * 8 + rl;
-      const void* sa = la.addr(min(m0 + row, M - 1), k8base + k8);
-      __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) void*)sa,
-          (__attribute__((address_space(3))) void*)(As + chunk * 1024),
-          16, 0, 0);
-      const void* sb = lb.addr(min(n0 + row, N - 1), k8base + k8);
-      __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) void*)sb,
-          (__attribute__((address_space(3))) void*)(Bs + chunk * 1024),
-          16, 0, 0);
+      if (!straddle) {
+        const void* sa = la.addr(min(m0 + row, M - 1), k8base + k8);
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) void*)sa,
+            (__attribute__((address_space(3))) void*)(As + chunk * 1024),
+            16, 0, 0);
+        const void* sb = lb.addr(min(n0 + row, N - 1), k8base + k8);
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) void*)sb,
+            (__attribute__((address_space(3))) void*)(Bs + chunk * 1024),
+            16, 0, 0);
+      } else {
+        // rare slow path: per-element guarded loads + LDS write at the
+        // same slot the DMA would have filled (lane*16 within chunk)
+        unsigned short va[8], vb[8];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          va[j] = la.elem(min(m0 + row, M - 1), kc0 + j);
+          vb[j] = lb.elem(min(n0 + row, N - 1), kc0 + j);
+        }
+        *(short8*)(As + chunk * 1024 + lane * 16) = *(short8*)va;
+        *(short8*)(Bs + chunk * 1024 + lane * 16) = *(short8*)vb;
+      }
     }
   };
 
