@@ -50,8 +50,8 @@ void pool_bwd_launch(const unsigned short*, unsigned short*, const int*,
 void lrn_fwd_launch(const unsigned short*, unsigned short*, float*, long,
                     int, float, float, float, int, hipStream_t);
 void lrn_bwd_launch(const unsigned short*, const unsigned short*,
-                    const float*, unsigned short*, long, int, float, float,
-                    int, hipStream_t);
+                    const float*, float*, unsigned short*, long, int,
+                    float, float, int, hipStream_t);
 
 namespace {
 
@@ -235,9 +235,10 @@ void lrn_bwd(torch::Tensor ey, torch::Tensor x,
              int64_t win) {
   const int C = x.size(-1);
   (void)y; (void)k;
-  lrn_bwd_launch(ubf(ey), ubf(x), denom.data_ptr<float>(), ubf_mut(ex),
-                 x.numel() / C, C, (float)alpha, (float)beta, (int)win,
-                 cur_stream());
+  auto ws = torch::empty_like(denom);  // fp32 t_j workspace (pass 1)
+  lrn_bwd_launch(ubf(ey), ubf(x), denom.data_ptr<float>(),
+                 ws.data_ptr<float>(), ubf_mut(ex), x.numel() / C, C,
+                 (float)alpha, (float)beta, (int)win, cur_stream());
 }
 
 void dropout_fwd(torch::Tensor x, torch::Tensor y, torch::Tensor mask,

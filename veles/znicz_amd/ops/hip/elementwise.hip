@@ -101,19 +101,33 @@ void sgd_launch(float* w, const float* g, float* vel, unsigned short* wb,
 }
 
 // -- column sum (bias gradient): out[N] = Σ_m A[M][N], bf16 -> fp32 --------
+// Two-stage: grid.y row-slabs each accumulate locally (coalesced 2-B
+// lane reads), then ONE fp32 atomic per (col, slab) — keeps the whole
+// 256-CU chip busy on the tall-skinny conv bias shapes (M ≈ 10⁵..10⁶).
 __global__ void col_sum_k(const unsigned short* __restrict__ a,
-                          float* __restrict__ out, int M, int N) {
+                          float* __restrict__ out, int M, int N,
+                          int rows_per_slab) {
   int col = blockIdx.x * blockDim.x + threadIdx.x;
   if (col >= N) return;
+  long m0 = (long)blockIdx.y * rows_per_slab;
+  long m1 = min((long)M, m0 + rows_per_slab);
   float s = 0.f;
-  for (long m = 0; m < M; ++m) s += bf2f(a[m * (long)N + col]);
-  out[col] = s;
+  const unsigned short* p = a + m0 * N + col;
+  for (long m = m0; m < m1; ++m, p += N) s += bf2f(*p);
+  if (gridDim.y == 1)
+    out[col] = s;
+  else
+    unsafeAtomicAdd(&out[col], s);  // hw global_atomic_add_f32 (gfx950)
 }
 
 void col_sum_launch(const unsigned short* a, float* out, int M, int N,
                     hipStream_t s) {
-  hipLaunchKernelGGL(col_sum_k, dim3((N + 255) / 256), dim3(256), 0, s, a,
-                     out, M, N);
+  int bx = (N + 255) / 256;
+  int slabs = (int)min((long)(M + 63) / 64, (long)max(1, 1024 / bx));
+  if (slabs > 1) hipMemsetAsync(out, 0, (size_t)N * 4, s);
+  int rps = (M + slabs - 1) / slabs;
+  hipLaunchKernelGGL(col_sum_k, dim3(bx, slabs), dim3(256), 0, s, a,
+                     out, M, N, rps);
   HIP_CHECK_LAST();
 }
 

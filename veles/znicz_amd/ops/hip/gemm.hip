@@ -276,6 +276,11 @@ struct Im2colTN {
   }
 };
 
+// Split-K: gridDim.y slices the K loop; each slice accumulates its
+// partial tile with ONE fp32 atomic per element (hw global fadd) — the
+// wgrad shapes (M=n_kernels ≤ 384, N=KH·KW·C ≤ a few k, K=N·OH·OW ≈ 10⁵..10⁶)
+// have too few MN tiles to fill 256 CUs without it.  beta is applied by
+// the launcher (memset / prescale) when split.
 template <typename LA, typename LB>
 __global__ __launch_bounds__(256, 2) void gemm_tn_k(
     LA la, LB lb, float* __restrict__ C, int M, int N, int K, float beta,
@@ -288,6 +293,9 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_k(
   const int m0 = tm * BM, n0 = tn * BN;
   const int wr = wave >> 1, wc = wave & 1;
   const int nk = (K + BK - 1) / BK;
+  const int sp = gridDim.y, sid = blockIdx.y;
+  const int t_lo = (int)(((long)nk * sid) / sp);
+  const int t_hi = (int)(((long)nk * (sid + 1)) / sp);
 
   // stage one operand's [BK][128] K-major slab as a swizzled [128][BK]
   // image: thread covers (col, k-octet) chunks; 8 coalesced u16 loads
@@ -314,7 +322,7 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_k(
   char* As0 = smem;
   char* Bs0 = smem + NT_LDS_A;
 
-  for (int t = 0; t < nk; ++t) {
+  for (int t = t_lo; t < t_hi; ++t) {
     const int buf = t & 1;
     char* As = As0 + buf * NT_LDS_BUF;
     char* Bs = Bs0 + buf * NT_LDS_BUF;
@@ -361,10 +369,19 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_k(
         const int row = rbase + r;
         if (row >= M) continue;
         float* p = C + (long)row * N + col;
-        *p = (beta == 0.f ? 0.f : beta * *p) + acc[mi][ni][r];
+        if (sp > 1)
+          unsafeAtomicAdd(p, acc[mi][ni][r]);
+        else
+          *p = (beta == 0.f ? 0.f : beta * *p) + acc[mi][ni][r];
       }
     }
   }
+}
+
+__global__ void scale_f32_k(float* __restrict__ p, long n, float beta) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long q = i; q < n; q += stride) p[q] *= beta;
 }
 
 // ---------------------------------------------------------------------------
@@ -390,7 +407,20 @@ template <typename LA, typename LB>
 void launch_gemm_tn(const LA& la, const LB& lb, float* C, int M, int N,
                     int K, float beta, hipStream_t stream) {
   const int tn = cdiv(N, BN);
-  dim3 grid(cdiv(M, BM) * tn), block(256);
+  const int tiles = cdiv(M, BM) * tn;
+  const int nk = cdiv(K, BK);
+  const int splits = max(1, min(nk, 1024 / tiles));
+  if (splits > 1) {
+    if (beta == 0.f) {
+      hipMemsetAsync(C, 0, (size_t)M * N * sizeof(float), stream);
+    } else if (beta != 1.f) {
+      long n = (long)M * N;
+      int b = (int)min((n + 255) / 256, (long)2048);
+      hipLaunchKernelGGL(scale_f32_k, dim3(b), dim3(256), 0, stream, C, n,
+                         beta);
+    }
+  }
+  dim3 grid(tiles, splits), block(256);
   hipLaunchKernelGGL((gemm_tn_k<LA, LB>), grid, block, NT_LDS_TOTAL, stream,
                      la, lb, C, M, N, K, beta, tn);
   HIP_CHECK_LAST();

@@ -152,70 +152,85 @@ void pool_bwd_launch(const unsigned short* ey, unsigned short* ex,
 
 // -- cross-channel LRN ------------------------------------------------------
 //   y = x * (k + alpha * Σ_{j∈window} x_j²)^(-beta) ; denom saved fp32
+// Thread per ELEMENT (c innermost ⇒ lane-coalesced; window neighbours
+// are L1 hits), one fast powf each — not per (c, j) pair.
 __global__ void lrn_fwd_k(const unsigned short* __restrict__ x,
                           unsigned short* __restrict__ y,
-                          float* __restrict__ denom, long rows, int C,
+                          float* __restrict__ denom, long total, int C,
                           float alpha, float beta, float kk, int win) {
-  long row = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
   const int half = win / 2;
-  if (row >= rows) return;
-  const unsigned short* xr = x + row * C;
-  unsigned short* yr = y + row * C;
-  float* dr = denom + row * C;
-  for (int c = 0; c < C; ++c) {
-    float s = 0.f;
+  for (long p = i; p < total; p += stride) {
+    int c = (int)(p % C);
+    long r0 = p - c;
     int j0 = max(0, c - half), j1 = min(C - 1, c + half);
+    float s = 0.f;
     for (int j = j0; j <= j1; ++j) {
-      float v = bf2f(xr[j]);
+      float v = bf2f(x[r0 + j]);
       s += v * v;
     }
     float d = kk + alpha * s;
-    dr[c] = d;
-    yr[c] = f2bf(bf2f(xr[c]) * __powf(d, -beta));
+    denom[p] = d;
+    y[p] = f2bf(bf2f(x[p]) * __powf(d, -beta));
   }
 }
 
-//   dx_c = dy_c·D_c^{-β} − 2αβ·x_c · Σ_{j: c∈W_j} dy_j·x_j·D_j^{-β-1}
+// backward in two fully-parallel passes (ws = fp32 workspace, numel total):
+//   pass 1:  t_j = dy_j·x_j·D_j^{-β-1}
+__global__ void lrn_bwd_pre_k(const unsigned short* __restrict__ ey,
+                              const unsigned short* __restrict__ x,
+                              const float* __restrict__ denom,
+                              float* __restrict__ t, long total,
+                              float beta) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long p = i; p < total; p += stride)
+    t[p] = bf2f(ey[p]) * bf2f(x[p]) * __powf(denom[p], -beta - 1.f);
+}
+
+//   pass 2:  dx_c = dy_c·D_c^{-β} − 2αβ·x_c · Σ_{|j−c|≤half} t_j
 __global__ void lrn_bwd_k(const unsigned short* __restrict__ ey,
                           const unsigned short* __restrict__ x,
                           const float* __restrict__ denom,
-                          unsigned short* __restrict__ ex, long rows,
+                          const float* __restrict__ t,
+                          unsigned short* __restrict__ ex, long total,
                           int C, float alpha, float beta, int win) {
-  long row = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
   const int half = win / 2;
-  if (row >= rows) return;
-  const unsigned short* er = ey + row * C;
-  const unsigned short* xr = x + row * C;
-  const float* dr = denom + row * C;
-  unsigned short* or_ = ex + row * C;
-  for (int c = 0; c < C; ++c) {
+  for (long p = i; p < total; p += stride) {
+    int c = (int)(p % C);
+    long r0 = p - c;
     int j0 = max(0, c - half), j1 = min(C - 1, c + half);
     float acc = 0.f;
-    for (int j = j0; j <= j1; ++j) {
-      float dj = dr[j];
-      acc += bf2f(er[j]) * bf2f(xr[j]) * __powf(dj, -beta - 1.f);
-    }
-    float v = bf2f(er[c]) * __powf(dr[c], -beta) -
-              2.f * alpha * beta * bf2f(xr[c]) * acc;
-    or_[c] = f2bf(v);
+    for (int j = j0; j <= j1; ++j) acc += t[r0 + j];
+    float v = bf2f(ey[p]) * __powf(denom[p], -beta) -
+              2.f * alpha * beta * bf2f(x[p]) * acc;
+    ex[p] = f2bf(v);
   }
 }
 
 void lrn_fwd_launch(const unsigned short* x, unsigned short* y,
                     float* denom, long rows, int C, float alpha,
                     float beta, float k, int win, hipStream_t s) {
-  dim3 grid((unsigned)((rows + 255) / 256)), block(256);
-  hipLaunchKernelGGL(lrn_fwd_k, grid, block, 0, s, x, y, denom, rows, C,
-                     alpha, beta, k, win);
+  long total = rows * C;
+  int blocks = (int)min((total + 255) / 256, (long)8192);
+  hipLaunchKernelGGL(lrn_fwd_k, dim3(blocks), dim3(256), 0, s, x, y,
+                     denom, total, C, alpha, beta, k, win);
   HIP_CHECK_LAST();
 }
 
 void lrn_bwd_launch(const unsigned short* ey, const unsigned short* x,
-                    const float* denom, unsigned short* ex, long rows,
-                    int C, float alpha, float beta, int win,
+                    const float* denom, float* ws, unsigned short* ex,
+                    long rows, int C, float alpha, float beta, int win,
                     hipStream_t s) {
-  dim3 grid((unsigned)((rows + 255) / 256)), block(256);
-  hipLaunchKernelGGL(lrn_bwd_k, grid, block, 0, s, ey, x, denom, ex, rows,
-                     C, alpha, beta, win);
+  long total = rows * C;
+  int blocks = (int)min((total + 255) / 256, (long)8192);
+  hipLaunchKernelGGL(lrn_bwd_pre_k, dim3(blocks), dim3(256), 0, s, ey, x,
+                     denom, ws, total, beta);
+  HIP_CHECK_LAST();
+  hipLaunchKernelGGL(lrn_bwd_k, dim3(blocks), dim3(256), 0, s, ey, x,
+                     denom, ws, ex, total, C, alpha, beta, win);
   HIP_CHECK_LAST();
 }
