@@ -84,6 +84,14 @@ __device__ __forceinline__ float uniform01(unsigned long long seed,
   return (float)(z >> 40) * (1.0f / 16777216.0f);  // 24-bit mantissa
 }
 
+// host-side: log2 of the pow2 lane->channel tile (≤256) for NHWC kernels:
+// lanes cover channels with shifts/masks only — no per-element div/mod.
+static inline int chan_lg(int C) {
+  int lg = 0;
+  while ((1 << lg) < C && lg < 8) ++lg;
+  return lg;
+}
+
 #define HIP_CHECK_LAST()                                                     \
   do {                                                                       \
     hipError_t e_ = hipGetLastError();                                       \

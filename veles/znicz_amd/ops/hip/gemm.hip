@@ -241,38 +241,55 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_k(
 // core TN (K-major operands; fp32 out += )
 // ---------------------------------------------------------------------------
 
-// dense K-major: element (k, m) at base[k*ld + m]; OOB -> 0
+// dense K-major: element (k, m) at base[k*ld + m]; OOB -> 0.
+// loadrow8 fills 8 consecutive k at fixed m — address walks by ld, the
+// only per-element work is the bound check (no div/mod in the loop).
 struct DenseTN {
   const unsigned short* base;
   long ld;
   int Klim, Mlim;
-  __device__ unsigned short load(int k, int m) const {
-    bool ok = (k < Klim) & (m < Mlim);
-    long off = ok ? (long)k * ld + m : 0;
-    unsigned short v = base[off];
-    return ok ? v : (unsigned short)0;
+  __device__ void loadrow8(int k0, int m, unsigned short* v) const {
+    const bool mok = m < Mlim;
+    long off = (long)k0 * ld + m;
+#pragma unroll
+    for (int j = 0; j < 8; ++j, off += ld) {
+      bool ok = mok & (k0 + j < Klim);
+      unsigned short val = base[ok ? off : 0];
+      v[j] = ok ? val : (unsigned short)0;
+    }
   }
 };
 
-// im2col K-major: (k = n*OH*OW + oh*OW + ow, column n = (kh,kw,c))
+// im2col K-major: (k = n*OH*OW + oh*OW + ow, column m = (kh,kw,c)).
+// One div/mod pair decodes k0, then (ow, oh, n) walk incrementally —
+// the naive per-(k,m) decode was the single hottest ALU path of the
+// whole AlexNet step (wgrad staging).
 struct Im2colTN {
   const unsigned short* x;
   ConvDesc d;
   int Klim, Mlim;
-  __device__ unsigned short load(int k, int m) const {
-    bool ok = (k < Klim) & (m < Mlim);
+  __device__ void loadrow8(int k0, int m, unsigned short* v) const {
     int c = m % d.C, t = m / d.C;
     int kw = t % d.KW, kh = t / d.KW;
-    int ow = k % d.OW;
-    t = k / d.OW;
+    const bool mok = (m < Mlim) & (kh < d.KH);
+    int ow = k0 % d.OW;
+    t = k0 / d.OW;
     int oh = t % d.OH, n = t / d.OH;
-    int ih = oh * d.SY - d.PT + kh;
-    int iw = ow * d.SX - d.PL + kw;
-    ok &= (kh < d.KH) & ((unsigned)ih < (unsigned)d.H) &
-          ((unsigned)iw < (unsigned)d.W);
-    long off = ok ? (((long)n * d.H + ih) * d.W + iw) * d.C + c : 0;
-    unsigned short v = x[off];
-    return ok ? v : (unsigned short)0;
+    const int iwk = kw - d.PL, ihk = kh - d.PT;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int ih = oh * d.SY + ihk;
+      int iw = ow * d.SX + iwk;
+      bool ok = mok & (k0 + j < Klim) & ((unsigned)ih < (unsigned)d.H) &
+                ((unsigned)iw < (unsigned)d.W);
+      long off = ok ? (((long)n * d.H + ih) * d.W + iw) * d.C + c : 0;
+      unsigned short val = x[off];
+      v[j] = ok ? val : (unsigned short)0;
+      if (++ow == d.OW) {
+        ow = 0;
+        if (++oh == d.OH) { oh = 0; ++n; }
+      }
+    }
   }
 };
 
@@ -298,8 +315,8 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_k(
   const int t_hi = (int)(((long)nk * (sid + 1)) / sp);
 
   // stage one operand's [BK][128] K-major slab as a swizzled [128][BK]
-  // image: thread covers (col, k-octet) chunks; 8 coalesced u16 loads
-  // pack to short8; one swizzled ds_write_b128.
+  // image: thread covers (col, k-octet) chunks; loadrow8 walks 8 k's
+  // incrementally (coalesced across lanes); one swizzled ds_write_b128.
   auto stage_op = [&](char* dst, const auto& ld_, int kt, int lim0,
                       int is_b) {
     const int k0 = kt * BK;
@@ -309,9 +326,7 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_k(
       const int col = chunk & 127;            // m (or n) within tile
       const int ko = chunk >> 7;              // k-octet 0..7
       unsigned short v[8];
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        v[j] = ld_.load(k0 + ko * 8 + j, lim0 + col);
+      ld_.loadrow8(k0 + ko * 8, lim0 + col, v);
       const int slot = ko ^ (col & 7);        // bank-conflict-free
       *(short8*)(dst + col * 128 + slot * 16) = *(short8*)v;
     }

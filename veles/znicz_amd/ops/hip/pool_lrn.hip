@@ -4,21 +4,42 @@
 // c-innermost layout so consecutive lanes stream consecutive channels.
 #include "common.h"
 
-// -- forward: thread per output element ------------------------------------
+// -- forward: lanes cover channels via pow2 mask (no div/mod per elem),
+// rows (n, oh, ow) decoded once per row -------------------------------------
+template <int MODE>
+__device__ __forceinline__ void pool_fwd_one(
+    const unsigned short* __restrict__ x, unsigned short* __restrict__ y,
+    int* __restrict__ offs, const PoolDesc& d, long idx, int n, int ih0,
+    int iw0, int c);
+
 template <int MODE>
 __global__ void pool_fwd_k(const unsigned short* __restrict__ x,
                            unsigned short* __restrict__ y,
-                           int* __restrict__ offs, PoolDesc d) {
-  long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  const long total = (long)d.N * d.OH * d.OW * d.C;
-  if (idx >= total) return;
-  int c = idx % d.C;
-  long t = idx / d.C;
-  int ow = t % d.OW;
-  t /= d.OW;
-  int oh = t % d.OH;
-  int n = t / d.OH;
-  const int ih0 = oh * d.SY, iw0 = ow * d.SX;
+                           int* __restrict__ offs, PoolDesc d, int lg) {
+  const int C2 = 1 << lg;
+  const int c0 = threadIdx.x & (C2 - 1);
+  const int rin = threadIdx.x >> lg;
+  const int rpb = blockDim.x >> lg;
+  const unsigned rows = (unsigned)d.N * d.OH * d.OW;
+  for (unsigned row = blockIdx.x * rpb + rin; row < rows;
+       row += gridDim.x * rpb) {
+    unsigned t = row / d.OW;
+    const int ow = row - t * d.OW;
+    const int n = t / d.OH;
+    const int oh = t - n * d.OH;
+    const int ih0 = oh * d.SY, iw0 = ow * d.SX;
+    for (int c = c0; c < d.C; c += C2) {
+      const long idx = (long)row * d.C + c;
+      pool_fwd_one<MODE>(x, y, offs, d, idx, n, ih0, iw0, c);
+    }
+  }
+}
+
+template <int MODE>
+__device__ __forceinline__ void pool_fwd_one(
+    const unsigned short* __restrict__ x, unsigned short* __restrict__ y,
+    int* __restrict__ offs, const PoolDesc& d, long idx, int n, int ih0,
+    int iw0, int c) {
   float best = 0.f, besta = -1.f, sum = 0.f;
   int bestoff = -1, count = 0;
   float stoch_target = 0.f, stoch_acc = 0.f;
@@ -87,44 +108,54 @@ __global__ void pool_fwd_k(const unsigned short* __restrict__ x,
   }
 }
 
-// -- backward: thread per INPUT element (gather, no atomics) ----------------
+// -- backward: gather over input elements (no atomics); same pow2
+// lane->channel mapping, row (n, ih, iw) decoded once per row ---------------
 template <int MODE>
 __global__ void pool_bwd_k(const unsigned short* __restrict__ ey,
                            unsigned short* __restrict__ ex,
-                           const int* __restrict__ offs, PoolDesc d) {
-  long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  const long total = (long)d.N * d.H * d.W * d.C;
-  if (idx >= total) return;
-  int c = idx % d.C;
-  long t = idx / d.C;
-  int iw = t % d.W;
-  t /= d.W;
-  int ih = t % d.H;
-  int n = t / d.H;
-  float g = 0.f;
-  // output cells whose window covers (ih, iw)
-  int oy0 = (ih - d.KY + 1 <= 0) ? 0 : (ih - d.KY) / d.SY + 1;
-  int oy1 = min(d.OH - 1, ih / d.SY);
-  int ox0 = (iw - d.KX + 1 <= 0) ? 0 : (iw - d.KX) / d.SX + 1;
-  int ox1 = min(d.OW - 1, iw / d.SX);
-  for (int oy = oy0; oy <= oy1; ++oy)
-    for (int ox = ox0; ox <= ox1; ++ox) {
-      long o = (((long)n * d.OH + oy) * d.OW + ox) * d.C + c;
-      if (MODE == POOL_AVG) {
-        g += bf2f(ey[o]) / (float)max(offs[o], 1);
-      } else {
-        if (offs[o] == (int)idx) g += bf2f(ey[o]);
-      }
+                           const int* __restrict__ offs, PoolDesc d,
+                           int lg) {
+  const int C2 = 1 << lg;
+  const int c0 = threadIdx.x & (C2 - 1);
+  const int rin = threadIdx.x >> lg;
+  const int rpb = blockDim.x >> lg;
+  const unsigned rows = (unsigned)d.N * d.H * d.W;
+  for (unsigned row = blockIdx.x * rpb + rin; row < rows;
+       row += gridDim.x * rpb) {
+    unsigned t = row / d.W;
+    const int iw = row - t * d.W;
+    const int n = t / d.H;
+    const int ih = t - n * d.H;
+    // output cells whose window covers (ih, iw)
+    const int oy0 = (ih - d.KY + 1 <= 0) ? 0 : (ih - d.KY) / d.SY + 1;
+    const int oy1 = min(d.OH - 1, ih / d.SY);
+    const int ox0 = (iw - d.KX + 1 <= 0) ? 0 : (iw - d.KX) / d.SX + 1;
+    const int ox1 = min(d.OW - 1, iw / d.SX);
+    for (int c = c0; c < d.C; c += C2) {
+      const long idx = (long)row * d.C + c;
+      float g = 0.f;
+      for (int oy = oy0; oy <= oy1; ++oy)
+        for (int ox = ox0; ox <= ox1; ++ox) {
+          long o = (((long)n * d.OH + oy) * d.OW + ox) * d.C + c;
+          if (MODE == POOL_AVG) {
+            g += bf2f(ey[o]) / (float)max(offs[o], 1);
+          } else {
+            if (offs[o] == (int)idx) g += bf2f(ey[o]);
+          }
+        }
+      ex[idx] = f2bf(g);
     }
-  ex[idx] = f2bf(g);
+  }
 }
 
 void pool_fwd_launch(const unsigned short* x, unsigned short* y, int* offs,
                      const PoolDesc& d, int mode, hipStream_t s) {
-  long total = (long)d.N * d.OH * d.OW * d.C;
-  dim3 grid((unsigned)((total + 255) / 256)), block(256);
+  const int lg = chan_lg(d.C);
+  const int rpb = 256 >> lg;
+  long rows = (long)d.N * d.OH * d.OW;
+  dim3 grid((unsigned)min((rows + rpb - 1) / rpb, (long)32768)), block(256);
 #define GO(M_)                                                              \
-  hipLaunchKernelGGL(pool_fwd_k<M_>, grid, block, 0, s, x, y, offs, d)
+  hipLaunchKernelGGL(pool_fwd_k<M_>, grid, block, 0, s, x, y, offs, d, lg)
   switch (mode) {
     case POOL_AVG: GO(POOL_AVG); break;
     case POOL_MAXABS: GO(POOL_MAXABS); break;
@@ -138,10 +169,12 @@ void pool_fwd_launch(const unsigned short* x, unsigned short* y, int* offs,
 void pool_bwd_launch(const unsigned short* ey, unsigned short* ex,
                      const int* offs, const PoolDesc& d, int mode,
                      hipStream_t s) {
-  long total = (long)d.N * d.H * d.W * d.C;
-  dim3 grid((unsigned)((total + 255) / 256)), block(256);
+  const int lg = chan_lg(d.C);
+  const int rpb = 256 >> lg;
+  long rows = (long)d.N * d.H * d.W;
+  dim3 grid((unsigned)min((rows + rpb - 1) / rpb, (long)32768)), block(256);
 #define GO(M_)                                                              \
-  hipLaunchKernelGGL(pool_bwd_k<M_>, grid, block, 0, s, ey, ex, offs, d)
+  hipLaunchKernelGGL(pool_bwd_k<M_>, grid, block, 0, s, ey, ex, offs, d, lg)
   switch (mode) {
     case POOL_AVG: GO(POOL_AVG); break;
     default: GO(POOL_MAX); break;  // max/maxabs/stochastic share offsets
@@ -152,32 +185,39 @@ void pool_bwd_launch(const unsigned short* ey, unsigned short* ex,
 
 // -- cross-channel LRN ------------------------------------------------------
 //   y = x * (k + alpha * Σ_{j∈window} x_j²)^(-beta) ; denom saved fp32
-// Thread per ELEMENT (c innermost ⇒ lane-coalesced; window neighbours
-// are L1 hits), one fast powf each — not per (c, j) pair.
+// pow2 lane->channel mapping (coalesced, div-free); one fast powf per
+// element — not per (c, j) pair.
 __global__ void lrn_fwd_k(const unsigned short* __restrict__ x,
                           unsigned short* __restrict__ y,
-                          float* __restrict__ denom, long total, int C,
-                          float alpha, float beta, float kk, int win) {
-  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  long stride = (long)gridDim.x * blockDim.x;
+                          float* __restrict__ denom, unsigned rows, int C,
+                          int lg, float alpha, float beta, float kk,
+                          int win) {
+  const int C2 = 1 << lg;
+  const int c0 = threadIdx.x & (C2 - 1);
+  const int rin = threadIdx.x >> lg;
+  const int rpb = blockDim.x >> lg;
   const int half = win / 2;
-  for (long p = i; p < total; p += stride) {
-    int c = (int)(p % C);
-    long r0 = p - c;
-    int j0 = max(0, c - half), j1 = min(C - 1, c + half);
-    float s = 0.f;
-    for (int j = j0; j <= j1; ++j) {
-      float v = bf2f(x[r0 + j]);
-      s += v * v;
+  for (unsigned row = blockIdx.x * rpb + rin; row < rows;
+       row += gridDim.x * rpb) {
+    const unsigned short* xr = x + (long)row * C;
+    float* dr = denom + (long)row * C;
+    unsigned short* yr = y + (long)row * C;
+    for (int c = c0; c < C; c += C2) {
+      int j0 = max(0, c - half), j1 = min(C - 1, c + half);
+      float s = 0.f;
+      for (int j = j0; j <= j1; ++j) {
+        float v = bf2f(xr[j]);
+        s += v * v;
+      }
+      float d = kk + alpha * s;
+      dr[c] = d;
+      yr[c] = f2bf(bf2f(xr[c]) * __powf(d, -beta));
     }
-    float d = kk + alpha * s;
-    denom[p] = d;
-    y[p] = f2bf(bf2f(x[p]) * __powf(d, -beta));
   }
 }
 
 // backward in two fully-parallel passes (ws = fp32 workspace, numel total):
-//   pass 1:  t_j = dy_j·x_j·D_j^{-β-1}
+//   pass 1:  t_j = dy_j·x_j·D_j^{-β-1}   (vectorized 8-wide)
 __global__ void lrn_bwd_pre_k(const unsigned short* __restrict__ ey,
                               const unsigned short* __restrict__ x,
                               const float* __restrict__ denom,
@@ -185,8 +225,23 @@ __global__ void lrn_bwd_pre_k(const unsigned short* __restrict__ ey,
                               float beta) {
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   long stride = (long)gridDim.x * blockDim.x;
-  for (long p = i; p < total; p += stride)
-    t[p] = bf2f(ey[p]) * bf2f(x[p]) * __powf(denom[p], -beta - 1.f);
+  for (long p = i * 8; p < total; p += stride * 8) {
+    if (p + 8 <= total) {
+      unsigned short e8[8], x8[8];
+      *(short8*)e8 = *(const short8*)(ey + p);
+      *(short8*)x8 = *(const short8*)(x + p);
+      float o[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        o[j] = bf2f(e8[j]) * bf2f(x8[j]) *
+               __powf(denom[p + j], -beta - 1.f);
+      *(f32x4*)(t + p) = *(f32x4*)o;
+      *(f32x4*)(t + p + 4) = *(f32x4*)(o + 4);
+    } else {
+      for (long q = p; q < total; ++q)
+        t[q] = bf2f(ey[q]) * bf2f(x[q]) * __powf(denom[q], -beta - 1.f);
+    }
+  }
 }
 
 //   pass 2:  dx_c = dy_c·D_c^{-β} − 2αβ·x_c · Σ_{|j−c|≤half} t_j
@@ -194,30 +249,40 @@ __global__ void lrn_bwd_k(const unsigned short* __restrict__ ey,
                           const unsigned short* __restrict__ x,
                           const float* __restrict__ denom,
                           const float* __restrict__ t,
-                          unsigned short* __restrict__ ex, long total,
-                          int C, float alpha, float beta, int win) {
-  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  long stride = (long)gridDim.x * blockDim.x;
+                          unsigned short* __restrict__ ex, unsigned rows,
+                          int C, int lg, float alpha, float beta,
+                          int win) {
+  const int C2 = 1 << lg;
+  const int c0 = threadIdx.x & (C2 - 1);
+  const int rin = threadIdx.x >> lg;
+  const int rpb = blockDim.x >> lg;
   const int half = win / 2;
-  for (long p = i; p < total; p += stride) {
-    int c = (int)(p % C);
-    long r0 = p - c;
-    int j0 = max(0, c - half), j1 = min(C - 1, c + half);
-    float acc = 0.f;
-    for (int j = j0; j <= j1; ++j) acc += t[r0 + j];
-    float v = bf2f(ey[p]) * __powf(denom[p], -beta) -
-              2.f * alpha * beta * bf2f(x[p]) * acc;
-    ex[p] = f2bf(v);
+  for (unsigned row = blockIdx.x * rpb + rin; row < rows;
+       row += gridDim.x * rpb) {
+    const unsigned short* er = ey + (long)row * C;
+    const unsigned short* xr = x + (long)row * C;
+    const float* dr = denom + (long)row * C;
+    const float* tr = t + (long)row * C;
+    unsigned short* or_ = ex + (long)row * C;
+    for (int c = c0; c < C; c += C2) {
+      int j0 = max(0, c - half), j1 = min(C - 1, c + half);
+      float acc = 0.f;
+      for (int j = j0; j <= j1; ++j) acc += tr[j];
+      float v = bf2f(er[c]) * __powf(dr[c], -beta) -
+                2.f * alpha * beta * bf2f(xr[c]) * acc;
+      or_[c] = f2bf(v);
+    }
   }
 }
 
 void lrn_fwd_launch(const unsigned short* x, unsigned short* y,
                     float* denom, long rows, int C, float alpha,
                     float beta, float k, int win, hipStream_t s) {
-  long total = rows * C;
-  int blocks = (int)min((total + 255) / 256, (long)8192);
-  hipLaunchKernelGGL(lrn_fwd_k, dim3(blocks), dim3(256), 0, s, x, y,
-                     denom, total, C, alpha, beta, k, win);
+  const int lg = chan_lg(C);
+  const int rpb = 256 >> lg;
+  dim3 grid((unsigned)min((rows + rpb - 1) / rpb, (long)32768)), block(256);
+  hipLaunchKernelGGL(lrn_fwd_k, grid, block, 0, s, x, y, denom,
+                     (unsigned)rows, C, lg, alpha, beta, k, win);
   HIP_CHECK_LAST();
 }
 
@@ -226,11 +291,14 @@ void lrn_bwd_launch(const unsigned short* ey, const unsigned short* x,
                     long rows, int C, float alpha, float beta, int win,
                     hipStream_t s) {
   long total = rows * C;
-  int blocks = (int)min((total + 255) / 256, (long)8192);
+  int blocks = (int)min((total + 8 * 256 - 1) / (8 * 256), (long)4096);
   hipLaunchKernelGGL(lrn_bwd_pre_k, dim3(blocks), dim3(256), 0, s, ey, x,
                      denom, ws, total, beta);
   HIP_CHECK_LAST();
-  hipLaunchKernelGGL(lrn_bwd_k, dim3(blocks), dim3(256), 0, s, ey, x,
-                     denom, ws, ex, total, C, alpha, beta, win);
+  const int lg = chan_lg(C);
+  const int rpb = 256 >> lg;
+  dim3 grid((unsigned)min((rows + rpb - 1) / rpb, (long)32768));
+  hipLaunchKernelGGL(lrn_bwd_k, grid, dim3(256), 0, s, ey, x, denom, ws,
+                     ex, (unsigned)rows, C, lg, alpha, beta, win);
   HIP_CHECK_LAST();
 }
