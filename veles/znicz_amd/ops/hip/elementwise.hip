@@ -101,12 +101,60 @@ void sgd_launch(float* w, const float* g, float* vel, unsigned short* wb,
 }
 
 // -- column sum (bias gradient): out[N] = Σ_m A[M][N], bf16 -> fp32 --------
-// Two-stage: grid.y row-slabs each accumulate locally (coalesced 2-B
-// lane reads), then ONE fp32 atomic per (col, slab) — keeps the whole
-// 256-CU chip busy on the tall-skinny conv bias shapes (M ≈ 10⁵..10⁶).
+// Thread owns 8 consecutive cols (one b128 load per row — wave reads
+// 1 KiB contiguous); row-slab blocks reduce through LDS, then one fp32
+// atomic per (octet, slab).  Fills 256 CUs on tall-skinny conv bias
+// shapes (M ≈ 10⁵..10⁶, N ≤ a few hundred).
 __global__ void col_sum_k(const unsigned short* __restrict__ a,
                           float* __restrict__ out, int M, int N,
-                          int rows_per_slab) {
+                          int rows_per_slab, int lgo) {
+  const int O2 = 1 << lgo;
+  const int oct = threadIdx.x & (O2 - 1);
+  const int rsub = threadIdx.x >> lgo;
+  const int rstep = blockDim.x >> lgo;
+  const int cb = oct * 8;
+  float s[8] = {};
+  const long m0 = (long)blockIdx.x * rows_per_slab;
+  const long m1 = min((long)M, m0 + rows_per_slab);
+  if (cb < N) {
+    if (cb + 8 <= N && (N & 7) == 0) {
+      const unsigned short* p = a + (m0 + rsub) * (long)N + cb;
+      for (long m = m0 + rsub; m < m1; m += rstep,
+                p += (long)N * rstep) {
+        unsigned short v[8];
+        *(short8*)v = *(const short8*)p;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) s[j] += bf2f(v[j]);
+      }
+    } else {
+      for (long m = m0 + rsub; m < m1; m += rstep)
+        for (int j = 0; j < 8 && cb + j < N; ++j)
+          s[j] += bf2f(a[m * (long)N + cb + j]);
+    }
+  }
+  __shared__ float red[256][8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) red[threadIdx.x][j] = s[j];
+  __syncthreads();
+  if (rsub == 0 && cb < N) {
+#pragma unroll 1
+    for (int r = 1; r < rstep; ++r)
+#pragma unroll
+      for (int j = 0; j < 8; ++j) s[j] += red[oct + r * O2][j];
+    for (int j = 0; j < 8 && cb + j < N; ++j) {
+      if (gridDim.x == 1)
+        out[cb + j] = s[j];
+      else
+        unsafeAtomicAdd(&out[cb + j], s[j]);  // hw fadd (gfx950)
+    }
+  }
+}
+
+// wide-N fallback (N ≥ 2048 exceeds one block's octet span; those
+// matrices are short FC-bias shapes where column parallelism suffices)
+__global__ void col_sum_wide_k(const unsigned short* __restrict__ a,
+                               float* __restrict__ out, int M, int N,
+                               int rows_per_slab) {
   int col = blockIdx.x * blockDim.x + threadIdx.x;
   if (col >= N) return;
   long m0 = (long)blockIdx.y * rows_per_slab;
@@ -117,17 +165,29 @@ __global__ void col_sum_k(const unsigned short* __restrict__ a,
   if (gridDim.y == 1)
     out[col] = s;
   else
-    unsafeAtomicAdd(&out[col], s);  // hw global_atomic_add_f32 (gfx950)
+    unsafeAtomicAdd(&out[col], s);
 }
 
 void col_sum_launch(const unsigned short* a, float* out, int M, int N,
                     hipStream_t s) {
-  int bx = (N + 255) / 256;
-  int slabs = (int)min((long)(M + 63) / 64, (long)max(1, 1024 / bx));
-  if (slabs > 1) hipMemsetAsync(out, 0, (size_t)N * 4, s);
+  if (N >= 2048) {
+    int bx = (N + 255) / 256;
+    int slabs = (int)min((long)(M + 63) / 64, (long)max(1, 1024 / bx));
+    if (slabs > 1) (void)hipMemsetAsync(out, 0, (size_t)N * 4, s);
+    int rps = (M + slabs - 1) / slabs;
+    hipLaunchKernelGGL(col_sum_wide_k, dim3(bx, slabs), dim3(256), 0, s,
+                       a, out, M, N, rps);
+    HIP_CHECK_LAST();
+    return;
+  }
+  const int noct = (N + 7) >> 3;
+  int lgo = 0;
+  while ((1 << lgo) < noct && lgo < 8) ++lgo;
+  int slabs = (int)min((long)(M + 31) / 32, (long)1024);
+  if (slabs > 1) (void)hipMemsetAsync(out, 0, (size_t)N * 4, s);
   int rps = (M + slabs - 1) / slabs;
-  hipLaunchKernelGGL(col_sum_k, dim3(bx, slabs), dim3(256), 0, s, a,
-                     out, M, N, rps);
+  hipLaunchKernelGGL(col_sum_k, dim3(slabs), dim3(256), 0, s, a,
+                     out, M, N, rps, lgo);
   HIP_CHECK_LAST();
 }
 

@@ -185,33 +185,68 @@ void pool_bwd_launch(const unsigned short* ey, unsigned short* ex,
 
 // -- cross-channel LRN ------------------------------------------------------
 //   y = x * (k + alpha * Σ_{j∈window} x_j²)^(-beta) ; denom saved fp32
-// pow2 lane->channel mapping (coalesced, div-free); one fast powf per
-// element — not per (c, j) pair.
+// Octet-vectorized: each thread owns 8 consecutive channels — 3 short8
+// halo loads, window sums in registers, vector stores.  Requires
+// half ≤ 8 (AlexNet win=5 ⇒ half=2).
+template <int WIN>
 __global__ void lrn_fwd_k(const unsigned short* __restrict__ x,
                           unsigned short* __restrict__ y,
                           float* __restrict__ denom, unsigned rows, int C,
-                          int lg, float alpha, float beta, float kk,
-                          int win) {
-  const int C2 = 1 << lg;
-  const int c0 = threadIdx.x & (C2 - 1);
-  const int rin = threadIdx.x >> lg;
-  const int rpb = blockDim.x >> lg;
-  const int half = win / 2;
+                          int lgo, float alpha, float beta, float kk) {
+  const int O2 = 1 << lgo;               // pow2 octet slots per row
+  const int o0 = threadIdx.x & (O2 - 1);
+  const int rin = threadIdx.x >> lgo;
+  const int rpb = blockDim.x >> lgo;
+  constexpr int half = WIN / 2;          // compile-time: v[] stays in regs
+  const int noct = (C + 7) >> 3;
   for (unsigned row = blockIdx.x * rpb + rin; row < rows;
        row += gridDim.x * rpb) {
     const unsigned short* xr = x + (long)row * C;
     float* dr = denom + (long)row * C;
     unsigned short* yr = y + (long)row * C;
-    for (int c = c0; c < C; c += C2) {
-      int j0 = max(0, c - half), j1 = min(C - 1, c + half);
-      float s = 0.f;
-      for (int j = j0; j <= j1; ++j) {
-        float v = bf2f(xr[j]);
-        s += v * v;
+    for (int oc = o0; oc < noct; oc += O2) {
+      const int cb = oc * 8;
+      float v[24];
+#pragma unroll
+      for (int h = 0; h < 3; ++h) {
+        const int c8 = cb + (h - 1) * 8;
+        if (c8 >= 0 && c8 + 8 <= C) {
+          unsigned short t8[8];
+          *(short8*)t8 = *(const short8*)(xr + c8);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) v[h * 8 + j] = bf2f(t8[j]);
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            int c = c8 + j;
+            v[h * 8 + j] = (c >= 0 && c < C) ? bf2f(xr[c]) : 0.f;
+          }
+        }
       }
-      float d = kk + alpha * s;
-      dr[c] = d;
-      yr[c] = f2bf(bf2f(xr[c]) * __powf(d, -beta));
+      unsigned short o8[8];
+      float d8[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float s = 0.f;
+#pragma unroll
+        for (int w = -half; w <= half; ++w) {
+          float t = v[8 + j + w];
+          s += t * t;
+        }
+        float d = kk + alpha * s;
+        d8[j] = d;
+        o8[j] = f2bf(v[8 + j] * __powf(d, -beta));
+      }
+      if (cb + 8 <= C) {
+        *(short8*)(yr + cb) = *(short8*)o8;
+        *(f32x4*)(dr + cb) = *(f32x4*)d8;
+        *(f32x4*)(dr + cb + 4) = *(f32x4*)(d8 + 4);
+      } else {
+        for (int j = 0; cb + j < C; ++j) {
+          yr[cb + j] = o8[j];
+          dr[cb + j] = d8[j];
+        }
+      }
     }
   }
 }
@@ -245,18 +280,20 @@ __global__ void lrn_bwd_pre_k(const unsigned short* __restrict__ ey,
 }
 
 //   pass 2:  dx_c = dy_c·D_c^{-β} − 2αβ·x_c · Σ_{|j−c|≤half} t_j
+// (octet-vectorized like the forward; t halo zero-padded = clipping)
+template <int WIN>
 __global__ void lrn_bwd_k(const unsigned short* __restrict__ ey,
                           const unsigned short* __restrict__ x,
                           const float* __restrict__ denom,
                           const float* __restrict__ t,
                           unsigned short* __restrict__ ex, unsigned rows,
-                          int C, int lg, float alpha, float beta,
-                          int win) {
-  const int C2 = 1 << lg;
-  const int c0 = threadIdx.x & (C2 - 1);
-  const int rin = threadIdx.x >> lg;
-  const int rpb = blockDim.x >> lg;
-  const int half = win / 2;
+                          int C, int lgo, float alpha, float beta) {
+  const int O2 = 1 << lgo;
+  const int o0 = threadIdx.x & (O2 - 1);
+  const int rin = threadIdx.x >> lgo;
+  const int rpb = blockDim.x >> lgo;
+  constexpr int half = WIN / 2;
+  const int noct = (C + 7) >> 3;
   for (unsigned row = blockIdx.x * rpb + rin; row < rows;
        row += gridDim.x * rpb) {
     const unsigned short* er = ey + (long)row * C;
@@ -264,25 +301,86 @@ __global__ void lrn_bwd_k(const unsigned short* __restrict__ ey,
     const float* dr = denom + (long)row * C;
     const float* tr = t + (long)row * C;
     unsigned short* or_ = ex + (long)row * C;
-    for (int c = c0; c < C; c += C2) {
-      int j0 = max(0, c - half), j1 = min(C - 1, c + half);
-      float acc = 0.f;
-      for (int j = j0; j <= j1; ++j) acc += tr[j];
-      float v = bf2f(er[c]) * __powf(dr[c], -beta) -
-                2.f * alpha * beta * bf2f(xr[c]) * acc;
-      or_[c] = f2bf(v);
+    for (int oc = o0; oc < noct; oc += O2) {
+      const int cb = oc * 8;
+      float tv[24];
+#pragma unroll
+      for (int h = 0; h < 3; ++h) {
+        const int c8 = cb + (h - 1) * 8;
+        if (c8 >= 0 && c8 + 8 <= C) {
+          *(f32x4*)(tv + h * 8) = *(const f32x4*)(tr + c8);
+          *(f32x4*)(tv + h * 8 + 4) = *(const f32x4*)(tr + c8 + 4);
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            int c = c8 + j;
+            tv[h * 8 + j] = (c >= 0 && c < C) ? tr[c] : 0.f;
+          }
+        }
+      }
+      unsigned short e8[8], x8[8], o8[8];
+      float dd[8];
+      if (cb + 8 <= C) {
+        *(short8*)e8 = *(const short8*)(er + cb);
+        *(short8*)x8 = *(const short8*)(xr + cb);
+        *(f32x4*)dd = *(const f32x4*)(dr + cb);
+        *(f32x4*)(dd + 4) = *(const f32x4*)(dr + cb + 4);
+      } else {
+        for (int j = 0; j < 8; ++j) {
+          int c = cb + j;
+          e8[j] = c < C ? er[c] : 0;
+          x8[j] = c < C ? xr[c] : 0;
+          dd[j] = c < C ? dr[c] : 1.f;
+        }
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float acc = 0.f;
+#pragma unroll
+        for (int w = -half; w <= half; ++w) acc += tv[8 + j + w];
+        float v = bf2f(e8[j]) * __powf(dd[j], -beta) -
+                  2.f * alpha * beta * bf2f(x8[j]) * acc;
+        o8[j] = f2bf(v);
+      }
+      if (cb + 8 <= C) {
+        *(short8*)(or_ + cb) = *(short8*)o8;
+      } else {
+        for (int j = 0; cb + j < C; ++j) or_[cb + j] = o8[j];
+      }
     }
   }
+}
+
+// window sizes are small odd ints; compile the common ones (register-
+// resident halos need compile-time WIN)
+#define LRN_WIN_SWITCH(win, F)                                              \
+  switch (win) {                                                            \
+    case 3: F(3); break;                                                    \
+    case 5: F(5); break;                                                    \
+    case 7: F(7); break;                                                    \
+    case 9: F(9); break;                                                    \
+    default: throw std::runtime_error("LRN win must be 3/5/7/9");           \
+  }
+
+static inline int lrn_lgo(int C) {
+  int noct = (C + 7) >> 3;
+  int lg = 0;
+  while ((1 << lg) < noct && lg < 5) ++lg;
+  return lg;
 }
 
 void lrn_fwd_launch(const unsigned short* x, unsigned short* y,
                     float* denom, long rows, int C, float alpha,
                     float beta, float k, int win, hipStream_t s) {
-  const int lg = chan_lg(C);
-  const int rpb = 256 >> lg;
-  dim3 grid((unsigned)min((rows + rpb - 1) / rpb, (long)32768)), block(256);
-  hipLaunchKernelGGL(lrn_fwd_k, grid, block, 0, s, x, y, denom,
-                     (unsigned)rows, C, lg, alpha, beta, k, win);
+  const int lgo = lrn_lgo(C);
+  const int rpb = 256 >> lgo;
+  dim3 grid((unsigned)min((rows + rpb - 1) / rpb, (long)131072)),
+      block(256);
+#define GO(W)                                                               \
+  hipLaunchKernelGGL(lrn_fwd_k<W>, grid, block, 0, s, x, y, denom,          \
+                     (unsigned)rows, C, lgo, alpha, beta, k)
+  LRN_WIN_SWITCH(win, GO);
+#undef GO
   HIP_CHECK_LAST();
 }
 
@@ -295,10 +393,13 @@ void lrn_bwd_launch(const unsigned short* ey, const unsigned short* x,
   hipLaunchKernelGGL(lrn_bwd_pre_k, dim3(blocks), dim3(256), 0, s, ey, x,
                      denom, ws, total, beta);
   HIP_CHECK_LAST();
-  const int lg = chan_lg(C);
-  const int rpb = 256 >> lg;
-  dim3 grid((unsigned)min((rows + rpb - 1) / rpb, (long)32768));
-  hipLaunchKernelGGL(lrn_bwd_k, grid, dim3(256), 0, s, ey, x, denom, ws,
-                     ex, (unsigned)rows, C, lg, alpha, beta, win);
+  const int lgo = lrn_lgo(C);
+  const int rpb = 256 >> lgo;
+  dim3 grid((unsigned)min((rows + rpb - 1) / rpb, (long)131072));
+#define GO(W)                                                               \
+  hipLaunchKernelGGL(lrn_bwd_k<W>, grid, dim3(256), 0, s, ey, x, denom,     \
+                     ws, ex, (unsigned)rows, C, lgo, alpha, beta)
+  LRN_WIN_SWITCH(win, GO);
+#undef GO
   HIP_CHECK_LAST();
 }
