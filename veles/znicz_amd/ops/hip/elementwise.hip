@@ -127,9 +127,11 @@ __global__ void col_sum_k(const unsigned short* __restrict__ a,
         for (int j = 0; j < 8; ++j) s[j] += bf2f(v[j]);
       }
     } else {
-      for (long m = m0 + rsub; m < m1; m += rstep)
-        for (int j = 0; j < 8 && cb + j < N; ++j)
-          s[j] += bf2f(a[m * (long)N + cb + j]);
+      for (long m = m0 + rsub; m < m1; m += rstep) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          if (cb + j < N) s[j] += bf2f(a[m * (long)N + cb + j]);
+      }
     }
   }
   __shared__ float red[256][8];
@@ -141,7 +143,9 @@ __global__ void col_sum_k(const unsigned short* __restrict__ a,
     for (int r = 1; r < rstep; ++r)
 #pragma unroll
       for (int j = 0; j < 8; ++j) s[j] += red[oct + r * O2][j];
-    for (int j = 0; j < 8 && cb + j < N; ++j) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      if (cb + j >= N) continue;
       if (gridDim.x == 1)
         out[cb + j] = s[j];
       else

@@ -242,10 +242,12 @@ __global__ void lrn_fwd_k(const unsigned short* __restrict__ x,
         *(f32x4*)(dr + cb) = *(f32x4*)d8;
         *(f32x4*)(dr + cb + 4) = *(f32x4*)(d8 + 4);
       } else {
-        for (int j = 0; cb + j < C; ++j) {
-          yr[cb + j] = o8[j];
-          dr[cb + j] = d8[j];
-        }
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          if (cb + j < C) {
+            yr[cb + j] = o8[j];
+            dr[cb + j] = d8[j];
+          }
       }
     }
   }
@@ -326,6 +328,7 @@ __global__ void lrn_bwd_k(const unsigned short* __restrict__ ey,
         *(f32x4*)dd = *(const f32x4*)(dr + cb);
         *(f32x4*)(dd + 4) = *(const f32x4*)(dr + cb + 4);
       } else {
+#pragma unroll
         for (int j = 0; j < 8; ++j) {
           int c = cb + j;
           e8[j] = c < C ? er[c] : 0;
@@ -345,7 +348,9 @@ __global__ void lrn_bwd_k(const unsigned short* __restrict__ ey,
       if (cb + 8 <= C) {
         *(short8*)(or_ + cb) = *(short8*)o8;
       } else {
-        for (int j = 0; cb + j < C; ++j) or_[cb + j] = o8[j];
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          if (cb + j < C) or_[cb + j] = o8[j];
       }
     }
   }
