@@ -97,9 +97,8 @@ def bench_convs():
         us = timeit(lambda: ops.conv_wgrad(
             eo.view(N * OH * OW, K), x.view(N, -1), gw, desc, 0.0))
         report(nm + " wgrad", us, flops)
-        us = timeit(lambda: ops.col_sum(
-            eo.view(N * OH * OW, K),
-            torch.empty(K, device=DEV, dtype=torch.float32)))
+        bsum = torch.empty(K, device=DEV, dtype=torch.float32)
+        us = timeit(lambda: ops.col_sum(eo.view(N * OH * OW, K), bsum))
         report(nm + " bias(col_sum)", us, 0,
                N * OH * OW * K * 2)
 
@@ -135,11 +134,10 @@ def bench_lrns():
         ey = torch.randn_like(x)
         ex = torch.empty_like(x)
         nb = rows * C
-        us = timeit(lambda: ops.lrn_fwd(x.view(N, -1), y.view(N, -1),
-                                        denom, 1e-4, 0.75, 2.0, 5))
+        # NOTE: pass 4-D tensors — the binding reads C from size(-1)
+        us = timeit(lambda: ops.lrn_fwd(x, y, denom, 1e-4, 0.75, 2.0, 5))
         report(nm + " fwd", us, 0, nb * (2 + 2 + 4))
-        us = timeit(lambda: ops.lrn_bwd(ey.view(N, -1), x.view(N, -1),
-                                        None, denom, ex.view(N, -1),
+        us = timeit(lambda: ops.lrn_bwd(ey, x, None, denom, ex,
                                         1e-4, 0.75, 2.0, 5))
         report(nm + " bwd", us, 0, nb * (2 * 3 + 4 * 3 + 2))
 
