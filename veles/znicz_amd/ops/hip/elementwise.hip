@@ -100,6 +100,20 @@ void sgd_launch(float* w, const float* g, float* vel, unsigned short* wb,
   HIP_CHECK_LAST();
 }
 
+// zero-fill on the COMPUTE stream (hipMemsetAsync would bounce to the
+// SDMA engine — the cross-engine sync costs ~100 µs per call)
+__global__ void zero_f32_k(float* __restrict__ p, long n) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long q = i; q < n; q += stride) p[q] = 0.f;
+}
+
+void zero_f32_launch(float* p, long n, hipStream_t s) {
+  int b = (int)min((n + 255) / 256, (long)2048);
+  hipLaunchKernelGGL(zero_f32_k, dim3(b), dim3(256), 0, s, p, n);
+  HIP_CHECK_LAST();
+}
+
 // -- column sum (bias gradient): out[N] = Σ_m A[M][N], bf16 -> fp32 --------
 // Thread owns 8 consecutive cols (one b128 load per row — wave reads
 // 1 KiB contiguous); row-slab blocks reduce through LDS, then one fp32
@@ -177,7 +191,7 @@ void col_sum_launch(const unsigned short* a, float* out, int M, int N,
   if (N >= 2048) {
     int bx = (N + 255) / 256;
     int slabs = (int)min((long)(M + 63) / 64, (long)max(1, 1024 / bx));
-    if (slabs > 1) (void)hipMemsetAsync(out, 0, (size_t)N * 4, s);
+    if (slabs > 1) zero_f32_launch(out, N, s);
     int rps = (M + slabs - 1) / slabs;
     hipLaunchKernelGGL(col_sum_wide_k, dim3(bx, slabs), dim3(256), 0, s,
                        a, out, M, N, rps);
@@ -188,7 +202,7 @@ void col_sum_launch(const unsigned short* a, float* out, int M, int N,
   int lgo = 0;
   while ((1 << lgo) < noct && lgo < 8) ++lgo;
   int slabs = (int)min((long)(M + 31) / 32, (long)1024);
-  if (slabs > 1) (void)hipMemsetAsync(out, 0, (size_t)N * 4, s);
+  if (slabs > 1) zero_f32_launch(out, N, s);
   int rps = (M + slabs - 1) / slabs;
   hipLaunchKernelGGL(col_sum_k, dim3(slabs), dim3(256), 0, s, a,
                      out, M, N, rps, lgo);

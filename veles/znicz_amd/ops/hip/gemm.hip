@@ -427,7 +427,8 @@ void launch_gemm_tn(const LA& la, const LB& lb, float* C, int M, int N,
   const int splits = max(1, min(nk, 1024 / tiles));
   if (splits > 1) {
     if (beta == 0.f) {
-      (void)hipMemsetAsync(C, 0, (size_t)M * N * sizeof(float), stream);
+      void zero_f32_launch(float*, long, hipStream_t);
+      zero_f32_launch(C, (long)M * N, stream);
     } else if (beta != 1.f) {
       long n = (long)M * N;
       int b = (int)min((n + 255) / 256, (long)2048);
