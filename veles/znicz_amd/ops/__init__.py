@@ -239,3 +239,35 @@ def dropout_bwd(err, mask, ratio):
 
 def transpose2d(a, out):
     require().transpose2d(a, out)
+
+
+# misc (depooling / RBM / Kohonen / input-norm / Uniform) -------------------
+
+def gather_by_offset(src, offs, dst):
+    """dst[i] = src[offs[i]] (bf16, int32 absolute offsets)."""
+    require().gather_by_offset(src, offs, dst)
+
+
+def scatter_by_offset(src, offs, dst):
+    """dst zeroed then dst[offs[i]] = src[i] (unique offsets)."""
+    require().scatter_by_offset(src, offs, dst)
+
+
+def bernoulli(x, y, seed):
+    """y = (U < x) ? 1 : 0, counter-based PRNG (bf16 in/out)."""
+    require().bernoulli(x, y, int(seed))
+
+
+def fill_uniform(y, lo, hi, seed):
+    """fp32 uniform fill in [lo, hi) (reference Uniform unit)."""
+    require().fill_uniform(y, float(lo), float(hi), int(seed))
+
+
+def row_argmin(d, winners):
+    """winners[r] = argmin_c d[r, c] (fp32 rows, int32 out)."""
+    require().row_argmin(d, winners)
+
+
+def mean_disp(x, mean, rdisp, y):
+    """y = (x - mean) * rdisp, mean/rdisp broadcast over batch."""
+    require().mean_disp(x, mean, rdisp, y)

@@ -30,6 +30,17 @@ void sgd_launch(float*, const float*, float*, unsigned short*, long, float,
 void col_sum_launch(const unsigned short*, float*, int, int, hipStream_t);
 void transpose2d_launch(const unsigned short*, unsigned short*, int, int,
                         hipStream_t);
+void gather_off_launch(const unsigned short*, const int*, unsigned short*,
+                       long, hipStream_t);
+void scatter_off_launch(const unsigned short*, const int*, unsigned short*,
+                        long, long, hipStream_t);
+void bernoulli_launch(const unsigned short*, unsigned short*, long,
+                      unsigned long long, hipStream_t);
+void fill_uniform_launch(float*, long, float, float, unsigned long long,
+                         hipStream_t);
+void row_argmin_launch(const float*, int*, int, int, hipStream_t);
+void mean_disp_launch(const unsigned short*, const float*, const float*,
+                      unsigned short*, long, long, hipStream_t);
 void dropout_fwd_launch(const unsigned short*, unsigned short*,
                         unsigned char*, long, float, unsigned long long,
                         hipStream_t);
@@ -254,9 +265,49 @@ void dropout_bwd(torch::Tensor err, torch::Tensor mask, double ratio) {
                      err.numel(), (float)ratio, cur_stream());
 }
 
+void gather_by_offset(torch::Tensor src, torch::Tensor offs,
+                      torch::Tensor dst) {
+  gather_off_launch(ubf(src), offs.data_ptr<int>(), ubf_mut(dst),
+                    dst.numel(), cur_stream());
+}
+
+void scatter_by_offset(torch::Tensor src, torch::Tensor offs,
+                       torch::Tensor dst) {
+  scatter_off_launch(ubf(src), offs.data_ptr<int>(), ubf_mut(dst),
+                     src.numel(), dst.numel(), cur_stream());
+}
+
+void bernoulli(torch::Tensor x, torch::Tensor y, int64_t seed) {
+  bernoulli_launch(ubf(x), ubf_mut(y), x.numel(),
+                   (unsigned long long)seed, cur_stream());
+}
+
+void fill_uniform(torch::Tensor y, double lo, double hi, int64_t seed) {
+  fill_uniform_launch(y.data_ptr<float>(), y.numel(), (float)lo,
+                      (float)hi, (unsigned long long)seed, cur_stream());
+}
+
+void row_argmin(torch::Tensor d, torch::Tensor winners) {
+  row_argmin_launch(d.data_ptr<float>(), winners.data_ptr<int>(),
+                    d.size(0), d.size(1), cur_stream());
+}
+
+void mean_disp(torch::Tensor x, torch::Tensor mean, torch::Tensor rdisp,
+               torch::Tensor y) {
+  mean_disp_launch(ubf(x), mean.data_ptr<float>(),
+                   rdisp.data_ptr<float>(), ubf_mut(y), x.numel(),
+                   mean.numel(), cur_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("gather_by_offset", &gather_by_offset);
+  m.def("scatter_by_offset", &scatter_by_offset);
+  m.def("bernoulli", &bernoulli);
+  m.def("fill_uniform", &fill_uniform);
+  m.def("row_argmin", &row_argmin);
+  m.def("mean_disp", &mean_disp);
   m.doc() = "veles.znicz_amd hand-written CDNA4 (gfx950) kernels";
   m.def("gemm_nt", &gemm_nt, "C=act(A·Bᵀ+bias), bf16 MFMA");
   m.def("gemm_tn", &gemm_tn, "C=AᵀB (K-major wgrad core)");

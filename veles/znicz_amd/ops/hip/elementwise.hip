@@ -234,6 +234,159 @@ void transpose2d_launch(const unsigned short* in, unsigned short* out,
   HIP_CHECK_LAST();
 }
 
+// -- misc small kernels for cold units (depooling, RBM, Kohonen, input
+// normalization, Uniform) ---------------------------------------------------
+
+// dst[i] = src[offs[i]]  (depooling backward / unpooling gather)
+__global__ void gather_off_k(const unsigned short* __restrict__ src,
+                             const int* __restrict__ offs,
+                             unsigned short* __restrict__ dst, long n) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long p = i; p < n; p += stride) {
+    int o = offs[p];
+    dst[p] = o >= 0 ? src[o] : (unsigned short)0;
+  }
+}
+
+void gather_off_launch(const unsigned short* src, const int* offs,
+                       unsigned short* dst, long n, hipStream_t s) {
+  int b = (int)min((n + 255) / 256, (long)4096);
+  hipLaunchKernelGGL(gather_off_k, dim3(b), dim3(256), 0, s, src, offs,
+                     dst, n);
+  HIP_CHECK_LAST();
+}
+
+// dst[offs[i]] = src[i], dst pre-zeroed (depooling forward scatter; max-
+// pool offsets are unique per output cell so no atomics are needed)
+__global__ void scatter_off_k(const unsigned short* __restrict__ src,
+                              const int* __restrict__ offs,
+                              unsigned short* __restrict__ dst, long n) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long p = i; p < n; p += stride) {
+    int o = offs[p];
+    if (o >= 0) dst[o] = src[p];
+  }
+}
+
+__global__ void zero_bf16_k(unsigned short* __restrict__ p, long n) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long q = i; q < n; q += stride) p[q] = 0;
+}
+
+void scatter_off_launch(const unsigned short* src, const int* offs,
+                        unsigned short* dst, long n, long n_dst,
+                        hipStream_t s) {
+  int bz = (int)min((n_dst + 255) / 256, (long)4096);
+  hipLaunchKernelGGL(zero_bf16_k, dim3(bz), dim3(256), 0, s, dst, n_dst);
+  int b = (int)min((n + 255) / 256, (long)4096);
+  hipLaunchKernelGGL(scatter_off_k, dim3(b), dim3(256), 0, s, src, offs,
+                     dst, n);
+  HIP_CHECK_LAST();
+}
+
+// y = U(seed, i) < x  (Bernoulli sampling for RBM binarization)
+__global__ void bernoulli_k(const unsigned short* __restrict__ x,
+                            unsigned short* __restrict__ y, long n,
+                            unsigned long long seed) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long p = i; p < n; p += stride)
+    y[p] = uniform01(seed, (unsigned long long)p) < bf2f(x[p])
+               ? (unsigned short)0x3F80   // bf16 1.0
+               : (unsigned short)0;
+}
+
+void bernoulli_launch(const unsigned short* x, unsigned short* y, long n,
+                      unsigned long long seed, hipStream_t s) {
+  int b = (int)min((n + 255) / 256, (long)4096);
+  hipLaunchKernelGGL(bernoulli_k, dim3(b), dim3(256), 0, s, x, y, n, seed);
+  HIP_CHECK_LAST();
+}
+
+// uniform fill in [lo, hi) (the reference's Uniform accelerated unit)
+__global__ void fill_uniform_k(float* __restrict__ y, long n, float lo,
+                               float hi, unsigned long long seed) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long p = i; p < n; p += stride)
+    y[p] = lo + (hi - lo) * uniform01(seed, (unsigned long long)p);
+}
+
+void fill_uniform_launch(float* y, long n, float lo, float hi,
+                         unsigned long long seed, hipStream_t s) {
+  int b = (int)min((n + 255) / 256, (long)4096);
+  hipLaunchKernelGGL(fill_uniform_k, dim3(b), dim3(256), 0, s, y, n, lo,
+                     hi, seed);
+  HIP_CHECK_LAST();
+}
+
+// per-sample argmin over fp32 rows (Kohonen winner search): one wave per
+// row, lane-strided min + DPP-free shuffle reduce via LDS
+__global__ void row_argmin_k(const float* __restrict__ d,
+                             int* __restrict__ winners, int rows,
+                             int cols) {
+  __shared__ float bv[256];
+  __shared__ int bi[256];
+  const int row = blockIdx.x;
+  if (row >= rows) return;
+  const float* dr = d + (long)row * cols;
+  float best = 3.4e38f;
+  int besti = 0;
+  for (int c = threadIdx.x; c < cols; c += blockDim.x) {
+    float v = dr[c];
+    if (v < best) { best = v; besti = c; }
+  }
+  bv[threadIdx.x] = best;
+  bi[threadIdx.x] = besti;
+  __syncthreads();
+  for (int step = blockDim.x / 2; step > 0; step >>= 1) {
+    if (threadIdx.x < step) {
+      if (bv[threadIdx.x + step] < bv[threadIdx.x] ||
+          (bv[threadIdx.x + step] == bv[threadIdx.x] &&
+           bi[threadIdx.x + step] < bi[threadIdx.x])) {
+        bv[threadIdx.x] = bv[threadIdx.x + step];
+        bi[threadIdx.x] = bi[threadIdx.x + step];
+      }
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) winners[row] = bi[0];
+}
+
+void row_argmin_launch(const float* d, int* winners, int rows, int cols,
+                       hipStream_t s) {
+  hipLaunchKernelGGL(row_argmin_k, dim3(rows), dim3(256), 0, s, d,
+                     winners, rows, cols);
+  HIP_CHECK_LAST();
+}
+
+// y = (x - mean) * rdisp, mean/rdisp broadcast over the batch dim
+// (MeanDispNormalizer; S = sample size)
+__global__ void mean_disp_k(const unsigned short* __restrict__ x,
+                            const float* __restrict__ mean,
+                            const float* __restrict__ rdisp,
+                            unsigned short* __restrict__ y, long total,
+                            long S) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long p = i; p < total; p += stride) {
+    long sidx = p % S;
+    y[p] = f2bf((bf2f(x[p]) - mean[sidx]) * rdisp[sidx]);
+  }
+}
+
+void mean_disp_launch(const unsigned short* x, const float* mean,
+                      const float* rdisp, unsigned short* y, long total,
+                      long S, hipStream_t s) {
+  int b = (int)min((total + 255) / 256, (long)4096);
+  hipLaunchKernelGGL(mean_disp_k, dim3(b), dim3(256), 0, s, x, mean,
+                     rdisp, y, total, S);
+  HIP_CHECK_LAST();
+}
+
 // -- dropout ---------------------------------------------------------------
 __global__ void dropout_fwd_k(const unsigned short* __restrict__ x,
                               unsigned short* __restrict__ y,
