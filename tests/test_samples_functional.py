@@ -1,0 +1,60 @@
+"""Functional (integration) tests for the sample model zoo on the numpy
+backend — the reference's znicz/tests/functional pattern: run a real
+sample workflow a few epochs with fixed seeds, assert the loss/error
+trajectory improves, then resume-from-snapshot style checks are in
+test_mnist_functional.py.
+"""
+
+import numpy
+import pytest
+
+from veles import prng
+from veles.backends import NumpyDevice
+from veles.config import root
+from veles.dummy import DummyLauncher
+from veles.loader.base import TRAIN, VALID
+
+
+@pytest.mark.timeout(600)
+def test_mnist_ae_reconstructs():
+    prng.seed_all(2)
+    from veles.znicz_amd.models.mnist_ae import MnistAEWorkflow
+    root.mnist_ae.loader.update({"minibatch_size": 50, "n_train": 100,
+                                 "n_valid": 50})
+    root.mnist_ae.decision.update({"max_epochs": 4})
+    wf = MnistAEWorkflow(DummyLauncher())
+    wf.initialize(device=NumpyDevice())
+    wf.run()
+    assert wf.is_finished
+    first, last = wf.decision.mse_history[0], wf.decision.mse_history[-1]
+    assert last < 0.8 * first, (first, last)
+
+
+@pytest.mark.timeout(600)
+def test_cifar_short_numpy():
+    prng.seed_all(3)
+    from veles.znicz_amd.models.cifar import CifarWorkflow
+    root.cifar.loader.update({"minibatch_size": 64, "n_train": 128,
+                              "n_valid": 64, "on_device": False,
+                              "device_dtype": "fp32"})
+    root.cifar.decision.update({"max_epochs": 2})
+    wf = CifarWorkflow(DummyLauncher())
+    wf.initialize(device=NumpyDevice())
+    wf.run()
+    assert wf.is_finished
+    assert wf.loader.epoch_number == 2
+
+
+@pytest.mark.timeout(600)
+def test_wine_mlp():
+    """Reference znicz/samples/Wine — tiny 3-class MLP, must hit ~0
+    train error within a few epochs."""
+    prng.seed_all(4)
+    from veles.znicz_amd.models.wine import WineWorkflow
+    root.wine.decision.update({"max_epochs": 8})
+    wf = WineWorkflow(DummyLauncher())
+    wf.initialize(device=NumpyDevice())
+    wf.run()
+    assert wf.is_finished
+    assert wf.decision.epoch_n_err_pt[TRAIN] < 15.0, \
+        wf.decision.epoch_n_err_pt

@@ -142,6 +142,7 @@ class DecisionMSE(DecisionBase):
         self.min_validation_mse = None
         self.min_validation_mse_epoch = 0
         self.min_train_mse = None
+        self.mse_history = []           # train mse per epoch
 
     def on_last_minibatch(self, clazz):
         arr = self.minibatch_metrics
@@ -168,6 +169,7 @@ class DecisionMSE(DecisionBase):
                 self.improved <<= False
                 self.fails += 1
         if clazz == TRAIN:
+            self.mse_history.append(mse)
             if self.min_train_mse is None or mse < self.min_train_mse:
                 self.min_train_mse = mse
 

@@ -9,7 +9,7 @@ functional tests assert that trajectory.
 import numpy
 
 from veles import prng
-from veles.loader.fullbatch import FullBatchLoader
+from veles.loader.fullbatch import (FullBatchLoader, FullBatchLoaderMSE)
 
 
 class SyntheticClassLoader(FullBatchLoader):
@@ -46,3 +46,22 @@ class SyntheticClassLoader(FullBatchLoader):
         self.original_data.mem = data
         self.original_labels.mem = labels
         self.class_lengths = [self.n_test, self.n_valid, self.n_train]
+
+
+class SyntheticAELoader(FullBatchLoaderMSE):
+    """Autoencoder variant: regression targets = the input samples
+    (reference FullBatchLoaderMSE usage in MnistAE/VideoAE)."""
+
+    def __init__(self, workflow, sample_shape=(784,), n_classes=10,
+                 n_train=600, n_valid=100, n_test=0, noise=0.35,
+                 seed=0xC0FFEE, **kwargs):
+        super(SyntheticAELoader, self).__init__(workflow, **kwargs)
+        self.sample_shape = tuple(sample_shape)
+        self.n_classes = n_classes
+        self.n_train, self.n_valid, self.n_test = n_train, n_valid, n_test
+        self.noise = noise
+        self.seed = seed
+
+    def load_data(self):
+        SyntheticClassLoader.load_data(self)
+        self.original_targets.mem = self.original_data.mem.copy()
