@@ -58,3 +58,31 @@ def test_wine_mlp():
     assert wf.is_finished
     assert wf.decision.epoch_n_err_pt[TRAIN] < 15.0, \
         wf.decision.epoch_n_err_pt
+
+
+@pytest.mark.timeout(600)
+def test_rbm_cd1_improves_reconstruction():
+    prng.seed_all(5)
+    from veles.znicz_amd.models.rbm import RBMWorkflow
+    root.rbm.decision.update({"max_epochs": 3})
+    wf = RBMWorkflow(DummyLauncher())
+    wf.initialize(device=NumpyDevice())
+    wf.run()
+    assert wf.is_finished
+    hist = wf.decision.mse_history
+    assert hist[-1] < hist[0], hist
+    assert numpy.isfinite(wf.energy.energy)
+
+
+@pytest.mark.timeout(600)
+def test_kohonen_map_spreads():
+    prng.seed_all(6)
+    from veles.znicz_amd.models.kohonen_map import KohonenWorkflow
+    root.kohonen.decision.update({"max_epochs": 5})
+    wf = KohonenWorkflow(DummyLauncher())
+    wf.initialize(device=NumpyDevice())
+    wf.run()
+    assert wf.is_finished
+    wf.forward.output.map_read()
+    # the 4 clusters must spread over several SOM cells, not collapse
+    assert len(numpy.unique(wf.forward.output.mem)) >= 4

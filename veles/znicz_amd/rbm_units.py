@@ -184,7 +184,10 @@ class EnergyCalc(AcceleratedUnit):
         self.v.map_read()
         self.weights.map_read()
         v = self.v.mem.reshape(self.v.shape[0], -1)
-        x = v.dot(self.weights.mem)
+        w = self.weights.mem
+        if w.shape[0] != v.shape[1]:    # [nh, nv] layout -> transpose
+            w = w.T
+        x = v.dot(w)
         if self.hbias is not None and self.hbias:
             self.hbias.map_read()
             x = x + self.hbias.mem
