@@ -1,0 +1,167 @@
+"""GPU (MI355X) numerics for the extended unit set: each unit's hip_run
+against its own numpy_run on the same seed/data, plus short end-to-end
+runs of the AE / RBM / Kohonen sample workflows on the HIP path.
+"""
+
+import numpy
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+torch = pytest.importorskip("torch")
+if not torch.cuda.is_available():
+    pytest.skip("no GPU", allow_module_level=True)
+
+from veles import prng  # noqa: E402
+from veles.backends import NumpyDevice, ROCmDevice  # noqa: E402
+from veles.config import root  # noqa: E402
+from veles.dummy import DummyLauncher, DummyWorkflow  # noqa: E402
+from veles.memory import Array  # noqa: E402
+
+RNG = numpy.random.RandomState(21)
+
+
+def assert_close(got, want, atol=0.05, rtol=2e-2):
+    numpy.testing.assert_allclose(
+        numpy.asarray(got, numpy.float32),
+        numpy.asarray(want, numpy.float32), rtol=rtol, atol=atol)
+
+
+def run_both(make, data_dict, out_attrs=("output",)):
+    """Build+run a unit on numpy and rocm; compare mapped outputs."""
+    outs = {}
+    for kind in ("numpy", "rocm"):
+        dev = NumpyDevice() if kind == "numpy" else ROCmDevice(0)
+        wf = DummyWorkflow()
+        unit = make(wf)
+        for name, arr in data_dict.items():
+            a = Array(numpy.ascontiguousarray(arr).copy())
+            a.device_dtype = "bf16" if arr.dtype == numpy.float32 \
+                else None
+            setattr(unit, name, a)
+            a.initialize(dev)
+        assert unit.initialize(device=dev) is not False
+        unit.run()
+        row = []
+        for attr in out_attrs:
+            out = getattr(unit, attr)
+            out.map_read()
+            row.append(out.mem.copy())
+        outs[kind] = row
+    for g, w in zip(outs["rocm"], outs["numpy"]):
+        assert_close(g, w)
+    return outs
+
+
+def test_deconv_fwd_gpu():
+    from veles.znicz_amd.deconv import Deconv
+    data = RNG.uniform(-1, 1, (2, 5, 5, 16)).astype(numpy.float32)
+
+    def make(wf):
+        prng.get().seed(42)
+        return Deconv(wf, n_kernels=16, kx=3, ky=3, sliding=(2, 2),
+                      n_output_channels=8)
+    run_both(make, {"input": data})
+
+
+def test_mean_disp_gpu():
+    from veles.znicz_amd.mean_disp_normalizer import MeanDispNormalizer
+    data = RNG.uniform(-2, 2, (3, 6, 6, 8)).astype(numpy.float32)
+    mean = data.mean(axis=0)
+    rdisp = (1.0 / (data.std(axis=0) + 0.1)).astype(numpy.float32)
+
+    def make(wf):
+        u = MeanDispNormalizer(wf)
+        u.mean.mem = mean.copy()
+        u.rdisp.mem = rdisp.copy()
+        return u
+    run_both(make, {"input": data})
+
+
+def test_kohonen_forward_gpu():
+    from veles.znicz_amd.kohonen import KohonenForward
+    data = RNG.uniform(-1, 1, (32, 8)).astype(numpy.float32)
+    weights = RNG.uniform(-1, 1, (16, 8)).astype(numpy.float32)
+
+    outs = {}
+    for kind in ("numpy", "rocm"):
+        dev = NumpyDevice() if kind == "numpy" else ROCmDevice(0)
+        wf = DummyWorkflow()
+        u = KohonenForward(wf, shape=(4, 4))
+        u.input = Array(data.copy())
+        u.input.device_dtype = "bf16"
+        u.input.initialize(dev)
+        u.weights = Array(weights.copy())
+        u.weights.initialize(dev)
+        assert u.initialize(device=dev) is not False
+        u.run()
+        u.output.map_read()
+        outs[kind] = u.output.mem.copy()
+    numpy.testing.assert_array_equal(outs["rocm"], outs["numpy"])
+
+
+def test_bernoulli_and_row_argmin_ops():
+    from veles.znicz_amd import ops
+    x = torch.full((100000,), 0.3, dtype=torch.bfloat16, device="cuda")
+    y = torch.empty_like(x)
+    ops.bernoulli(x, y, 12345)
+    m = y.float().mean().item()
+    assert 0.28 < m < 0.32
+    d = torch.randn(64, 100, device="cuda", dtype=torch.float32)
+    w = torch.empty(64, dtype=torch.int32, device="cuda")
+    ops.row_argmin(d, w)
+    numpy.testing.assert_array_equal(w.cpu().numpy(),
+                                     d.argmin(dim=1).cpu().numpy())
+
+
+def test_gather_scatter_by_offset():
+    from veles.znicz_amd import ops
+    src = torch.randn(64, device="cuda").bfloat16()
+    offs = torch.randperm(256, device="cuda")[:64].int()
+    dst = torch.empty(256, device="cuda", dtype=torch.bfloat16)
+    ops.scatter_by_offset(src, offs, dst)
+    back = torch.empty(64, device="cuda", dtype=torch.bfloat16)
+    ops.gather_by_offset(dst, offs, back)
+    assert torch.equal(back, src)
+    assert dst.float().abs().sum() == src.float().abs().sum()
+
+
+@pytest.mark.timeout(300)
+def test_mnist_ae_trains_on_gpu():
+    prng.seed_all(2)
+    from veles.znicz_amd.models.mnist_ae import MnistAEWorkflow
+    root.mnist_ae.loader.update({"minibatch_size": 50, "n_train": 100,
+                                 "n_valid": 50})
+    root.mnist_ae.decision.update({"max_epochs": 4})
+    wf = MnistAEWorkflow(DummyLauncher())
+    wf.initialize(device=ROCmDevice(0))
+    wf.run()
+    assert wf.is_finished
+    hist = wf.decision.mse_history
+    assert hist[-1] < 0.85 * hist[0], hist
+
+
+@pytest.mark.timeout(300)
+def test_rbm_trains_on_gpu():
+    prng.seed_all(5)
+    from veles.znicz_amd.models.rbm import RBMWorkflow
+    root.rbm.decision.update({"max_epochs": 3})
+    wf = RBMWorkflow(DummyLauncher())
+    wf.initialize(device=ROCmDevice(0))
+    wf.run()
+    assert wf.is_finished
+    hist = wf.decision.mse_history
+    assert hist[-1] <= hist[0] * 1.02, hist
+
+
+@pytest.mark.timeout(300)
+def test_kohonen_trains_on_gpu():
+    prng.seed_all(6)
+    from veles.znicz_amd.models.kohonen_map import KohonenWorkflow
+    root.kohonen.decision.update({"max_epochs": 5})
+    wf = KohonenWorkflow(DummyLauncher())
+    wf.initialize(device=ROCmDevice(0))
+    wf.run()
+    assert wf.is_finished
+    wf.forward.output.map_read()
+    assert len(numpy.unique(wf.forward.output.mem)) >= 4
