@@ -120,7 +120,7 @@ void zero_f32_launch(float* p, long n, hipStream_t s) {
 // atomic per (octet, slab).  Fills 256 CUs on tall-skinny conv bias
 // shapes (M ≈ 10⁵..10⁶, N ≤ a few hundred).
 __global__ void col_sum_k(const unsigned short* __restrict__ a,
-                          float* __restrict__ out, int M, int N,
+                          float* __restrict__ ws, int M, int N,
                           int rows_per_slab, int lgo) {
   const int O2 = 1 << lgo;
   const int oct = threadIdx.x & (O2 - 1);
@@ -157,15 +157,25 @@ __global__ void col_sum_k(const unsigned short* __restrict__ a,
     for (int r = 1; r < rstep; ++r)
 #pragma unroll
       for (int j = 0; j < 8; ++j) s[j] += red[oct + r * O2][j];
+    // plain stores into the per-slab workspace row (NO atomics: 10^3
+    // blocks x few cache lines of fp32 fadds serialize at the L2 slice
+    // owning them -- measured as a flat ~190 us floor)
+    float* dst = ws + (long)blockIdx.x * N + cb;  // ws==out at 1 slab
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      if (cb + j >= N) continue;
-      if (gridDim.x == 1)
-        out[cb + j] = s[j];
-      else
-        unsafeAtomicAdd(&out[cb + j], s[j]);  // hw fadd (gfx950)
-    }
+    for (int j = 0; j < 8; ++j)
+      if (cb + j < N) dst[j] = s[j];
   }
+}
+
+// pass 2: out[col] = sum_s ws[s][col] -- [slabs, N] fp32, slabs <= 1024
+__global__ void col_sum_fin_k(const float* __restrict__ ws,
+                              float* __restrict__ out, int slabs, int N) {
+  int col = blockIdx.x * blockDim.x + threadIdx.x;
+  if (col >= N) return;
+  float s = 0.f;
+  const float* p = ws + col;
+  for (int r = 0; r < slabs; ++r, p += N) s += *p;
+  out[col] = s;
 }
 
 // wide-N fallback (N ≥ 2048 exceeds one block's octet span; those
@@ -202,10 +212,28 @@ void col_sum_launch(const unsigned short* a, float* out, int M, int N,
   int lgo = 0;
   while ((1 << lgo) < noct && lgo < 8) ++lgo;
   int slabs = (int)min((long)(M + 31) / 32, (long)1024);
-  if (slabs > 1) zero_f32_launch(out, N, s);
+  if (slabs == 1) {
+    hipLaunchKernelGGL(col_sum_k, dim3(1), dim3(256), 0, s, a,
+                       out, M, N, M, lgo);
+    HIP_CHECK_LAST();
+    return;
+  }
+  // persistent per-device workspace for the slab partials (<= 8 MiB)
+  static float* ws = nullptr;
+  static size_t ws_cap = 0;
+  size_t need = (size_t)slabs * N * sizeof(float);
+  if (need > ws_cap) {
+    if (ws) (void)hipFree(ws);
+    ws_cap = need < (8u << 20) ? (8u << 20) : need;
+    if (hipMalloc(&ws, ws_cap) != hipSuccess)
+      throw std::runtime_error("col_sum workspace alloc failed");
+  }
   int rps = (M + slabs - 1) / slabs;
   hipLaunchKernelGGL(col_sum_k, dim3(slabs), dim3(256), 0, s, a,
-                     out, M, N, rps, lgo);
+                     ws, M, N, rps, lgo);
+  HIP_CHECK_LAST();
+  hipLaunchKernelGGL(col_sum_fin_k, dim3((N + 255) / 256), dim3(256), 0,
+                     s, ws, out, slabs, N);
   HIP_CHECK_LAST();
 }
 
