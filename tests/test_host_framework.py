@@ -1,0 +1,141 @@
+"""Host-framework surface tests: CLI, import_file, graph export,
+genetics GA, ensemble, image/pickle loaders, interaction shell,
+veles.znicz pickle-compat aliases.
+"""
+
+import os
+import pickle
+import subprocess
+import sys
+
+import numpy
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def test_cli_dry_run():
+    out = subprocess.run(
+        [sys.executable, "-m", "veles",
+         "veles/znicz_amd/models/wine.py", "--dry-run", "-b", "numpy"],
+        cwd=REPO, capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stderr[-2000:]
+    assert "dry run:" in out.stdout
+
+
+def test_cli_overrides_and_run(tmp_path):
+    dot = tmp_path / "g.dot"
+    out = subprocess.run(
+        [sys.executable, "-m", "veles",
+         "veles/znicz_amd/models/wine.py",
+         "root.wine.decision.max_epochs=1",
+         "root.wine.loader.n_train=30", "root.wine.loader.n_valid=0",
+         "-b", "numpy", "--random-seed", "4",
+         "--dump-graph", str(dot)],
+        cwd=REPO, capture_output=True, text=True, timeout=600)
+    assert out.returncode == 0, out.stderr[-2000:]
+    assert dot.exists() and "digraph" in dot.read_text()
+
+
+def test_cli_config_file(tmp_path):
+    cfg = tmp_path / "cfg.py"
+    cfg.write_text("root.wine.decision.max_epochs = 1\n"
+                   "root.wine.loader.n_train = 30\n"
+                   "root.wine.loader.n_valid = 0\n")
+    out = subprocess.run(
+        [sys.executable, "-m", "veles",
+         "veles/znicz_amd/models/wine.py", str(cfg), "-b", "numpy"],
+        cwd=REPO, capture_output=True, text=True, timeout=600)
+    assert out.returncode == 0, out.stderr[-2000:]
+
+
+def test_genetics_ga():
+    from veles.config import root
+    from veles.genetics import Tune, find_tunes, GeneticsOptimizer
+    root.tga.lr = Tune(0.01, 0.001, 0.1)
+    tunes = find_tunes(root.tga, "root.tga")
+    assert len(tunes) == 1
+
+    def ev(cfg):
+        return -(cfg["root.tga.lr"] - 0.05) ** 2
+    ga = GeneticsOptimizer(tunes, ev, population=8, generations=6)
+    best = ga.run()
+    assert abs(best.genes[0] - 0.05) < 0.02
+    assert len(ga.history) == 6
+
+
+def test_ensemble():
+    from veles.ensemble import EnsembleTrainer, majority_vote
+    models = EnsembleTrainer(lambda i, seed: ("m%d" % i, seed),
+                             n_models=3).run()
+    assert len(models) == 3
+    votes = majority_vote([[0, 1, 1], [0, 1, 2], [1, 1, 2]])
+    numpy.testing.assert_array_equal(votes, [0, 1, 2])
+
+
+def test_image_loader(tmp_path):
+    from PIL import Image
+    for cls in ("cats", "dogs"):
+        d = tmp_path / "train" / cls
+        d.mkdir(parents=True)
+        for i in range(3):
+            arr = (numpy.random.RandomState(i).uniform(
+                0, 255, (12, 10, 3))).astype(numpy.uint8)
+            Image.fromarray(arr).save(d / ("%d.png" % i))
+    from veles.dummy import DummyWorkflow
+    from veles.backends import NumpyDevice
+    from veles.loader.image import FileImageLoader
+    ldr = FileImageLoader(
+        DummyWorkflow(), minibatch_size=6,
+        train_paths=[str(tmp_path / "train" / "*" / "*.png")],
+        scale=(8, 8))
+    ldr.initialize(device=NumpyDevice())
+    assert ldr.original_data.shape == (6, 8, 8, 8)  # padded to 8 chans
+    assert sorted(numpy.unique(ldr.original_labels.mem)) == [0, 1]
+    ldr.run()
+    assert ldr.minibatch_data.mem.shape[0] == 6
+
+
+def test_pickles_loader_and_saver(tmp_path):
+    from veles.dummy import DummyWorkflow
+    from veles.backends import NumpyDevice
+    from veles.loader.pickles import PicklesLoader, MinibatchesSaver
+    data = numpy.random.RandomState(0).uniform(
+        -1, 1, (20, 4)).astype(numpy.float32)
+    labels = (numpy.arange(20) % 3).astype(numpy.int32)
+    p = tmp_path / "train.pickle"
+    with open(p, "wb") as f:
+        pickle.dump((data, labels), f)
+    ldr = PicklesLoader(DummyWorkflow(), minibatch_size=10,
+                        train_path=str(p))
+    ldr.initialize(device=NumpyDevice())
+    assert ldr.class_lengths == [0, 0, 20]
+    ldr.run()
+
+    sv = MinibatchesSaver(DummyWorkflow(),
+                          file_name=str(tmp_path / "mb.pickle.gz"))
+    sv.input = ldr.minibatch_data
+    sv.labels = ldr.minibatch_labels
+    sv.minibatch_size = 10
+    sv.initialize()
+    sv.run()
+    sv.close()
+    import gzip
+    with gzip.open(tmp_path / "mb.pickle.gz") as f:
+        d, l = pickle.load(f)
+    assert d.shape == (10, 4) and l.shape == (10,)
+
+
+def test_shell_noop():
+    from veles.dummy import DummyWorkflow
+    from veles.interaction import Shell
+    Shell(DummyWorkflow()).run()        # enabled=False -> returns
+
+
+def test_znicz_alias_unpickles_reference_paths():
+    import veles.znicz  # registers aliases  # noqa: F401
+    import importlib
+    mod = importlib.import_module("veles.znicz.all2all")
+    assert mod.All2All.__module__ == "veles.znicz_amd.all2all"
+    s = pickle.dumps(mod.All2All)
+    assert pickle.loads(s) is mod.All2All

@@ -1,0 +1,145 @@
+"""CLI entry point.
+
+Parity target: reference ``veles/__main__.py`` + ``scripts/velescli.py``
+(``Main``): ``python -m veles <workflow.py> [<config.py> ...] [flags]``
+imports the workflow module by path, executes the config files (they
+mutate the global ``root``), then invokes the module's
+``run(load, main)`` — the sample convention kept from the reference.
+
+MI355X additions: ``--gpus N`` self-spawns one process per GPU via
+``torch.distributed.run`` (RCCL inside one node) instead of the
+reference's master/slave flags.
+"""
+
+import argparse
+import os
+import sys
+
+
+def parse_args(argv):
+    p = argparse.ArgumentParser(
+        prog="veles",
+        description="MI355X-native Veles/Znicz: unit-graph NN training")
+    p.add_argument("workflow", help="workflow .py file (defines "
+                                    "run(load, main))")
+    p.add_argument("configs", nargs="*", help="config .py files "
+                   "(executed; mutate veles.config.root) and/or "
+                   "key=value overrides")
+    p.add_argument("-b", "--backend", default=None,
+                   choices=[None, "rocm", "numpy"],
+                   help="compute backend (default: auto)")
+    p.add_argument("-s", "--snapshot", default=None,
+                   help="resume from this snapshot file")
+    p.add_argument("--gpus", type=int, default=0,
+                   help="self-spawn N data-parallel ranks (one per GPU)")
+    p.add_argument("--dry-run", action="store_true",
+                   help="build + initialize the workflow, then exit")
+    p.add_argument("--random-seed", type=int, default=None)
+    p.add_argument("--dump-graph", default=None, metavar="DOT",
+                   help="write the workflow DAG to this .dot file")
+    p.add_argument("-v", "--verbose", action="store_true")
+    return p.parse_args(argv)
+
+
+def apply_configs(args):
+    from veles.config import root
+    for item in args.configs:
+        if item.endswith(".py") and os.path.exists(item):
+            with open(item) as f:
+                code = f.read()
+            exec(compile(code, item, "exec"),
+                 {"root": root, "__file__": item})
+        elif "=" in item:
+            path, _, value = item.partition("=")
+            node = root
+            parts = path.split(".")
+            if parts[0] == "root":
+                parts = parts[1:]
+            for part in parts[:-1]:
+                node = getattr(node, part)
+            try:
+                value = eval(value, {}, {})
+            except Exception:
+                pass
+            setattr(node, parts[-1], value)
+        else:
+            raise SystemExit("config argument %r is neither a .py file "
+                             "nor key=value" % item)
+
+
+def spawn_ranks(args):
+    """Re-exec under torch.distributed.run with one rank per GPU."""
+    cmd = [sys.executable, "-m", "torch.distributed.run",
+           "--nnodes=1", "--nproc-per-node", str(args.gpus),
+           "--master-addr", "127.0.0.1", "--master-port", "29411",
+           "-m", "veles", args.workflow] + args.configs
+    if args.backend:
+        cmd += ["--backend", args.backend]
+    if args.snapshot:
+        cmd += ["--snapshot", args.snapshot]
+    if args.random_seed is not None:
+        cmd += ["--random-seed", str(args.random_seed)]
+    os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+    os.execvpe(cmd[0], cmd, os.environ)
+
+
+class Main(object):
+    def __init__(self, args):
+        self.args = args
+        self.launcher = None
+        self.workflow = None
+
+    def load(self, workflow_class, **kwargs):
+        from veles.launcher import Launcher
+        from veles.snapshotter import SnapshotterToFile
+        self.launcher = Launcher(backend=self.args.backend,
+                                 snapshot=self.args.snapshot)
+        if self.args.snapshot:
+            self.workflow = SnapshotterToFile.import_(self.args.snapshot)
+            self.workflow.workflow = self.launcher
+        else:
+            self.workflow = workflow_class(self.launcher, **kwargs)
+        return self.workflow, bool(self.args.snapshot)
+
+    def main(self, **kwargs):
+        if self.workflow is None:
+            raise RuntimeError("run(load, main): load() was not called")
+        if self.args.dump_graph:
+            from veles.graph_plotter import save_dot
+            save_dot(self.workflow, self.args.dump_graph)
+            print("wrote %s" % self.args.dump_graph)
+        if self.args.dry_run:
+            dev = self.launcher.make_device()
+            self.workflow.initialize(device=dev, **kwargs)
+            print("dry run: %d units initialized" % len(self.workflow))
+            return
+        self.launcher.boot(**kwargs)
+
+    def run_module(self):
+        from veles.import_file import import_file_as_module
+        mod = import_file_as_module(self.args.workflow)
+        # reference order: workflow module import (sets root defaults)
+        # FIRST, then config files / overrides on top
+        apply_configs(self.args)
+        if not hasattr(mod, "run"):
+            raise SystemExit("%s does not define run(load, main)" %
+                             self.args.workflow)
+        mod.run(self.load, self.main)
+        return 0
+
+
+def main(argv=None):
+    args = parse_args(argv if argv is not None else sys.argv[1:])
+    if args.gpus > 1 and "WORLD_SIZE" not in os.environ:
+        spawn_ranks(args)                      # does not return
+    if args.random_seed is not None:
+        from veles import prng
+        prng.seed_all(args.random_seed)
+    if args.backend:
+        from veles.config import root
+        root.common.engine.backend = args.backend
+    return Main(args).run_module()
+
+
+if __name__ == "__main__":
+    sys.exit(main())
