@@ -167,15 +167,31 @@ __global__ void col_sum_k(const unsigned short* __restrict__ a,
   }
 }
 
-// pass 2: out[col] = sum_s ws[s][col] -- [slabs, N] fp32, slabs <= 1024
+// pass 2: out[col] = sum_s ws[s][col] — [slabs, N] fp32.  One block per
+// 8-column octet; threads stripe the slab dim (32 independent loads in
+// flight per lane), LDS tree-reduce.  A naive col-per-thread loop here
+// serialized on vmcnt and cost ~0.2 µs PER SLAB (the measured flat
+// ~200 µs floor of the whole col_sum) — see tools/probe_colsum.hip.
 __global__ void col_sum_fin_k(const float* __restrict__ ws,
                               float* __restrict__ out, int slabs, int N) {
-  int col = blockIdx.x * blockDim.x + threadIdx.x;
-  if (col >= N) return;
+  const int cb = blockIdx.x * 8;
+  const int j = threadIdx.x & 7;        // column within octet
+  const int r0 = threadIdx.x >> 3;      // slab stripe (32 lanes)
+  const int col = cb + j;
   float s = 0.f;
-  const float* p = ws + col;
-  for (int r = 0; r < slabs; ++r, p += N) s += *p;
-  out[col] = s;
+  if (col < N) {
+    for (int r = r0; r < slabs; r += 32)
+      s += ws[(long)r * N + col];
+  }
+  __shared__ float red[256];
+  red[threadIdx.x] = s;
+  __syncthreads();
+#pragma unroll
+  for (int step = 128; step >= 8; step >>= 1) {
+    if (threadIdx.x < step) red[threadIdx.x] += red[threadIdx.x + step];
+    __syncthreads();
+  }
+  if (threadIdx.x < 8 && col < N) out[col] = red[threadIdx.x];
 }
 
 // wide-N fallback (N ≥ 2048 exceeds one block's octet span; those
@@ -211,7 +227,7 @@ void col_sum_launch(const unsigned short* a, float* out, int M, int N,
   const int noct = (N + 7) >> 3;
   int lgo = 0;
   while ((1 << lgo) < noct && lgo < 8) ++lgo;
-  int slabs = (int)min((long)(M + 31) / 32, (long)1024);
+  int slabs = (int)min((long)(M + 63) / 64, (long)512);
   if (slabs == 1) {
     hipLaunchKernelGGL(col_sum_k, dim3(1), dim3(256), 0, s, a,
                        out, M, N, M, lgo);
@@ -232,7 +248,7 @@ void col_sum_launch(const unsigned short* a, float* out, int M, int N,
   hipLaunchKernelGGL(col_sum_k, dim3(slabs), dim3(256), 0, s, a,
                      ws, M, N, rps, lgo);
   HIP_CHECK_LAST();
-  hipLaunchKernelGGL(col_sum_fin_k, dim3((N + 255) / 256), dim3(256), 0,
+  hipLaunchKernelGGL(col_sum_fin_k, dim3((N + 7) / 8), dim3(256), 0,
                      s, ws, out, slabs, N);
   HIP_CHECK_LAST();
 }
