@@ -21,6 +21,7 @@ __global__ void pool_fwd_k(const unsigned short* __restrict__ x,
   const int rin = threadIdx.x >> lg;
   const int rpb = blockDim.x >> lg;
   const unsigned rows = (unsigned)d.N * d.OH * d.OW;
+  const bool vec8 = (d.C & 7) == 0;
   for (unsigned row = blockIdx.x * rpb + rin; row < rows;
        row += gridDim.x * rpb) {
     unsigned t = row / d.OW;
@@ -28,9 +29,135 @@ __global__ void pool_fwd_k(const unsigned short* __restrict__ x,
     const int n = t / d.OH;
     const int oh = t - n * d.OH;
     const int ih0 = oh * d.SY, iw0 = ow * d.SX;
-    for (int c = c0; c < d.C; c += C2) {
-      const long idx = (long)row * d.C + c;
-      pool_fwd_one<MODE>(x, y, offs, d, idx, n, ih0, iw0, c);
+    if (vec8) {
+      // octet-per-thread: one b128 window load serves 8 channels
+      const int noct = d.C >> 3;
+      for (int oc = c0; oc < noct; oc += C2) {
+        const int cb = oc * 8;
+        const long idx0 = (long)row * d.C + cb;
+        float best[8], sum[8];
+        int bestoff[8];
+        int count = 0;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          best[j] = 0.f;
+          sum[j] = 0.f;
+          bestoff[j] = -1;
+        }
+        float besta[8];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) besta[j] = -1.f;
+        float st_t[8], st_a[8];
+        if (MODE == POOL_STOCHASTIC) {
+          float possum[8] = {};
+          for (int ky = 0; ky < d.KY; ++ky) {
+            int ih = ih0 + ky;
+            if (ih >= d.H) break;
+            for (int kx = 0; kx < d.KX; ++kx) {
+              int iw = iw0 + kx;
+              if (iw >= d.W) break;
+              unsigned short v[8];
+              *(short8*)v = *(const short8*)(
+                  x + (((long)n * d.H + ih) * d.W + iw) * d.C + cb);
+#pragma unroll
+              for (int j = 0; j < 8; ++j)
+                possum[j] += fmaxf(bf2f(v[j]), 0.f);
+            }
+          }
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            st_t[j] = possum[j] * uniform01(
+                d.seed, (unsigned long long)(idx0 + j));
+            st_a[j] = 0.f;
+          }
+        }
+        for (int ky = 0; ky < d.KY; ++ky) {
+          int ih = ih0 + ky;
+          if (ih >= d.H) break;
+          for (int kx = 0; kx < d.KX; ++kx) {
+            int iw = iw0 + kx;
+            if (iw >= d.W) break;
+            long off = (((long)n * d.H + ih) * d.W + iw) * d.C + cb;
+            unsigned short v[8];
+            *(short8*)v = *(const short8*)(x + off);
+            ++count;
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+              float vj = bf2f(v[j]);
+              if (MODE == POOL_AVG) {
+                sum[j] += vj;
+              } else if (MODE == POOL_MAX) {
+                if (bestoff[j] < 0 || vj > best[j]) {
+                  best[j] = vj;
+                  bestoff[j] = (int)off + j;
+                }
+              } else if (MODE == POOL_MAXABS) {
+                float aj = fabsf(vj);
+                if (bestoff[j] < 0 || aj > besta[j]) {
+                  best[j] = vj;
+                  besta[j] = aj;
+                  bestoff[j] = (int)off + j;
+                }
+              } else {
+                float pv = fmaxf(vj, 0.f);
+                st_a[j] += pv;
+                if (bestoff[j] < 0 && st_a[j] > st_t[j] && pv > 0.f) {
+                  best[j] = vj;
+                  bestoff[j] = (int)off + j;
+                }
+              }
+            }
+          }
+        }
+        unsigned short o8[8];
+        int i8[8];
+        if (MODE == POOL_AVG) {
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            o8[j] = f2bf(sum[j] / max(count, 1));
+            i8[j] = count;
+          }
+        } else {
+          if (MODE == POOL_STOCHASTIC) {
+            // all-nonpositive windows fall back to max
+            for (int ky = 0; ky < d.KY; ++ky) {
+              int ih = ih0 + ky;
+              if (ih >= d.H) break;
+              for (int kx = 0; kx < d.KX; ++kx) {
+                int iw = iw0 + kx;
+                if (iw >= d.W) break;
+                long off = (((long)n * d.H + ih) * d.W + iw) * d.C + cb;
+                unsigned short v[8];
+                *(short8*)v = *(const short8*)(x + off);
+#pragma unroll
+                for (int j = 0; j < 8; ++j) {
+                  float vj = bf2f(v[j]);
+                  if (st_a[j] <= 0.f &&
+                      (bestoff[j] < 0 || vj > best[j])) {
+                    best[j] = vj;
+                    bestoff[j] = (int)off + j;
+                  }
+                }
+              }
+            }
+          }
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            o8[j] = f2bf(best[j]);
+            i8[j] = bestoff[j];
+          }
+        }
+        *(short8*)(y + idx0) = *(short8*)o8;
+        if (offs) {
+          *(int4*)(offs + idx0) = *(int4*)i8;
+          *(int4*)(offs + idx0 + 4) = *(int4*)(i8 + 4);
+        }
+      }
+    } else {
+      for (int c = c0; c < d.C; c += C2) {
+        const long idx = (long)row * d.C + c;
+        pool_fwd_one<MODE>(x, y, offs, d, idx, n, ih0, iw0, c);
+      }
     }
   }
 }
@@ -120,37 +247,71 @@ __global__ void pool_bwd_k(const unsigned short* __restrict__ ey,
   const int rin = threadIdx.x >> lg;
   const int rpb = blockDim.x >> lg;
   const unsigned rows = (unsigned)d.N * d.H * d.W;
+  const bool vec8 = (d.C & 7) == 0;
   for (unsigned row = blockIdx.x * rpb + rin; row < rows;
        row += gridDim.x * rpb) {
     unsigned t = row / d.W;
     const int iw = row - t * d.W;
     const int n = t / d.H;
     const int ih = t - n * d.H;
-    // output cells whose window covers (ih, iw)
     const int oy0 = (ih - d.KY + 1 <= 0) ? 0 : (ih - d.KY) / d.SY + 1;
     const int oy1 = min(d.OH - 1, ih / d.SY);
     const int ox0 = (iw - d.KX + 1 <= 0) ? 0 : (iw - d.KX) / d.SX + 1;
     const int ox1 = min(d.OW - 1, iw / d.SX);
-    for (int c = c0; c < d.C; c += C2) {
-      const long idx = (long)row * d.C + c;
-      float g = 0.f;
-      for (int oy = oy0; oy <= oy1; ++oy)
-        for (int ox = ox0; ox <= ox1; ++ox) {
-          long o = (((long)n * d.OH + oy) * d.OW + ox) * d.C + c;
-          if (MODE == POOL_AVG) {
-            g += bf2f(ey[o]) / (float)max(offs[o], 1);
-          } else {
-            if (offs[o] == (int)idx) g += bf2f(ey[o]);
+    if (vec8) {
+      const int noct = d.C >> 3;
+      for (int oc = c0; oc < noct; oc += C2) {
+        const int cb = oc * 8;
+        const long idx0 = (long)row * d.C + cb;
+        float g[8] = {};
+        for (int oy = oy0; oy <= oy1; ++oy)
+          for (int ox = ox0; ox <= ox1; ++ox) {
+            long o = (((long)n * d.OH + oy) * d.OW + ox) * d.C + cb;
+            unsigned short e8[8];
+            *(short8*)e8 = *(const short8*)(ey + o);
+            if (MODE == POOL_AVG) {
+              int c8[8];
+              *(int4*)c8 = *(const int4*)(offs + o);
+              *(int4*)(c8 + 4) = *(const int4*)(offs + o + 4);
+#pragma unroll
+              for (int j = 0; j < 8; ++j)
+                g[j] += bf2f(e8[j]) / (float)max(c8[j], 1);
+            } else {
+              int w8[8];
+              *(int4*)w8 = *(const int4*)(offs + o);
+              *(int4*)(w8 + 4) = *(const int4*)(offs + o + 4);
+#pragma unroll
+              for (int j = 0; j < 8; ++j)
+                if (w8[j] == (int)(idx0 + j)) g[j] += bf2f(e8[j]);
+            }
           }
-        }
-      ex[idx] = f2bf(g);
+        unsigned short o8[8];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) o8[j] = f2bf(g[j]);
+        *(short8*)(ex + idx0) = *(short8*)o8;
+      }
+    } else {
+      for (int c = c0; c < d.C; c += C2) {
+        const long idx = (long)row * d.C + c;
+        float g = 0.f;
+        for (int oy = oy0; oy <= oy1; ++oy)
+          for (int ox = ox0; ox <= ox1; ++ox) {
+            long o = (((long)n * d.OH + oy) * d.OW + ox) * d.C + c;
+            if (MODE == POOL_AVG) {
+              g += bf2f(ey[o]) / (float)max(offs[o], 1);
+            } else {
+              if (offs[o] == (int)idx) g += bf2f(ey[o]);
+            }
+          }
+        ex[idx] = f2bf(g);
+      }
     }
   }
 }
 
 void pool_fwd_launch(const unsigned short* x, unsigned short* y, int* offs,
                      const PoolDesc& d, int mode, hipStream_t s) {
-  const int lg = chan_lg(d.C);
+  const int lg = (d.C & 7) == 0 ? chan_lg(d.C >> 3) : chan_lg(d.C);
   const int rpb = 256 >> lg;
   long rows = (long)d.N * d.OH * d.OW;
   dim3 grid((unsigned)min((rows + rpb - 1) / rpb, (long)32768)), block(256);
@@ -169,7 +330,7 @@ void pool_fwd_launch(const unsigned short* x, unsigned short* y, int* offs,
 void pool_bwd_launch(const unsigned short* ey, unsigned short* ex,
                      const int* offs, const PoolDesc& d, int mode,
                      hipStream_t s) {
-  const int lg = chan_lg(d.C);
+  const int lg = (d.C & 7) == 0 ? chan_lg(d.C >> 3) : chan_lg(d.C);
   const int rpb = 256 >> lg;
   long rows = (long)d.N * d.H * d.W;
   dim3 grid((unsigned)min((rows + rpb - 1) / rpb, (long)32768)), block(256);
