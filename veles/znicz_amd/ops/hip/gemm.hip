@@ -400,6 +400,161 @@ __global__ void scale_f32_k(float* __restrict__ p, long n, float beta) {
 }
 
 // ---------------------------------------------------------------------------
+// core TN v2 — m-contiguous 16-B staged loads (the v1 per-element
+// k-gather made hipcc branch around each 2-B load and wait vmcnt(0)
+// per element — the documented conditional-load trap; one select per
+// 16-B chunk here).  LDS image [BK][BM+2] u16 (row pad ⇒ the four
+// 16-lane k-octet groups of a fragment read hit distinct bank halves);
+// fragments assembled from 8 ds_read_u16.  Requires both operand
+// m-dims % 8 == 0 (true for every conv/FC shape in the framework;
+// launcher falls back to v1 otherwise).
+// ---------------------------------------------------------------------------
+
+#define TNP (BM + 2)                    // padded LDS row, elements
+#define TN2_LDS_OP (BK * TNP * 2)       // one operand image, bytes
+#define TN2_LDS_BUF (TN2_LDS_OP * 2)    // A + B
+#define TN2_LDS_TOTAL (TN2_LDS_BUF * 2) // double-buffered
+
+struct DenseTNv {
+  const unsigned short* base;
+  long ld;
+  int Klim, Mlim;
+  __device__ void load16(int k, int mo, unsigned short* v) const {
+    const bool ok = (k < Klim) & (mo * 8 + 8 <= Mlim);
+    const unsigned short* p = ok ? base + (long)k * ld + mo * 8 : base;
+    *(short8*)v = *(const short8*)p;     // unconditional 16-B load
+#pragma unroll
+    for (int j = 0; j < 8; ++j) v[j] = ok ? v[j] : (unsigned short)0;
+  }
+};
+
+struct Im2colTNv {
+  const unsigned short* x;
+  ConvDesc d;
+  int Klim, Mlim;
+  __device__ void load16(int k, int mo, unsigned short* v) const {
+    const int m = mo * 8;
+    int c = m % d.C, t = m / d.C;        // c stays in one pixel C-run
+    int kw = t % d.KW, kh = t / d.KW;
+    int ow = k % d.OW;
+    int t2 = k / d.OW;
+    int oh = t2 % d.OH, n = t2 / d.OH;
+    int ih = oh * d.SY - d.PT + kh;
+    int iw = ow * d.SX - d.PL + kw;
+    const bool ok = (k < Klim) & (m < Mlim) & (kh < d.KH) &
+                    ((unsigned)ih < (unsigned)d.H) &
+                    ((unsigned)iw < (unsigned)d.W);
+    const unsigned short* p =
+        ok ? x + (((long)n * d.H + ih) * d.W + iw) * d.C + c : x;
+    *(short8*)v = *(const short8*)p;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) v[j] = ok ? v[j] : (unsigned short)0;
+  }
+};
+
+template <typename LA, typename LB>
+__global__ __launch_bounds__(256, 2) void gemm_tn2_k(
+    LA la, LB lb, float* __restrict__ C, int M, int N, int K, float beta,
+    int tiles_n) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int tile = blockIdx.x;
+  const int tm = tile / tiles_n, tn = tile % tiles_n;
+  const int m0 = tm * BM, n0 = tn * BN;
+  const int wr = wave >> 1, wc = wave & 1;
+  const int nk = (K + BK - 1) / BK;
+  const int sp = gridDim.y, sid = blockIdx.y;
+  const int t_lo = (int)(((long)nk * sid) / sp);
+  const int t_hi = (int)(((long)nk * (sid + 1)) / sp);
+  const int moA = m0 >> 3, moB = n0 >> 3;
+
+  auto stage2 = [&](int buf, int kt) {
+    char* As = smem + buf * TN2_LDS_BUF;
+    char* Bs = As + TN2_LDS_OP;
+    const int k0t = kt * BK;
+    const int oct = tid & 15;
+    const int kk0 = tid >> 4;            // 16 k-rows per pass
+#pragma unroll
+    for (int pass = 0; pass < 4; ++pass) {
+      const int kk = pass * 16 + kk0;
+      unsigned short v[8];
+      la.load16(k0t + kk, moA + oct, v);
+      *(short8*)(As + (kk * TNP + oct * 8) * 2) = *(short8*)v;
+      lb.load16(k0t + kk, moB + oct, v);
+      *(short8*)(Bs + (kk * TNP + oct * 8) * 2) = *(short8*)v;
+    }
+  };
+
+  f32x4 acc[4][4] = {};
+  if (t_lo >= t_hi) return;
+  stage2(0, t_lo);
+  __syncthreads();
+  int buf = 0;
+  for (int t = t_lo; t < t_hi; ++t) {
+    if (t + 1 < t_hi) stage2(buf ^ 1, t + 1);
+    const unsigned short* As =
+        (const unsigned short*)(smem + buf * TN2_LDS_BUF);
+    const unsigned short* Bs = As + BK * TNP;
+    const int fr = lane & 15;
+    const int ko8 = (lane >> 4) * 8;     // k-octet base within 32
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      short8 af[4], bf[4];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int mrow = wr * 64 + i * 16 + fr;
+        const int kbase = ks * 32 + ko8;
+        unsigned short tmp[8];
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          tmp[j] = As[(kbase + j) * TNP + mrow];
+        af[i] = *(short8*)tmp;
+      }
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int nrow = wc * 64 + i * 16 + fr;
+        const int kbase = ks * 32 + ko8;
+        unsigned short tmp[8];
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          tmp[j] = Bs[(kbase + j) * TNP + nrow];
+        bf[i] = *(short8*)tmp;
+      }
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+    }
+    __syncthreads();
+    buf ^= 1;
+  }
+
+  const int colw = n0 + wc * 64 + (lane & 15);
+#pragma unroll
+  for (int ni = 0; ni < 4; ++ni) {
+    const int col = colw + ni * 16;
+    if (col >= N) continue;
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi) {
+      const int rbase = m0 + wr * 64 + mi * 16 + ((lane >> 4) << 2);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = rbase + r;
+        if (row >= M) continue;
+        float* p = C + (long)row * N + col;
+        if (sp > 1)
+          unsafeAtomicAdd(p, acc[mi][ni][r]);
+        else
+          *p = (beta == 0.f ? 0.f : beta * *p) + acc[mi][ni][r];
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // launch helpers (called from bindings.cpp)
 // ---------------------------------------------------------------------------
 
@@ -442,6 +597,30 @@ void launch_gemm_tn(const LA& la, const LB& lb, float* C, int M, int N,
   HIP_CHECK_LAST();
 }
 
+template <typename LA, typename LB>
+void launch_gemm_tn2(const LA& la, const LB& lb, float* C, int M, int N,
+                     int K, float beta, hipStream_t stream) {
+  const int tn = cdiv(N, BN);
+  const int tiles = cdiv(M, BM) * tn;
+  const int nk = cdiv(K, BK);
+  const int splits = max(1, min(nk, 1024 / tiles));
+  if (splits > 1) {
+    if (beta == 0.f) {
+      void zero_f32_launch(float*, long, hipStream_t);
+      zero_f32_launch(C, (long)M * N, stream);
+    } else if (beta != 1.f) {
+      long n = (long)M * N;
+      int b = (int)min((n + 255) / 256, (long)2048);
+      hipLaunchKernelGGL(scale_f32_k, dim3(b), dim3(256), 0, stream, C, n,
+                         beta);
+    }
+  }
+  dim3 grid(tiles, splits), block(256);
+  hipLaunchKernelGGL((gemm_tn2_k<LA, LB>), grid, block, TN2_LDS_TOTAL,
+                     stream, la, lb, C, M, N, K, beta, tn);
+  HIP_CHECK_LAST();
+}
+
 // explicit entry points used by bindings.cpp --------------------------------
 
 void gemm_nt_dense(const unsigned short* A, const unsigned short* B,
@@ -458,6 +637,11 @@ void gemm_nt_dense(const unsigned short* A, const unsigned short* B,
 void gemm_tn_dense(const unsigned short* A, const unsigned short* B,
                    float* C, int M, int N, int K, float beta,
                    hipStream_t s) {
+  if ((M & 7) == 0 && (N & 7) == 0) {
+    DenseTNv la{A, M, K, M}, lb{B, N, K, N};
+    launch_gemm_tn2(la, lb, C, M, N, K, beta, s);
+    return;
+  }
   DenseTN la{A, M, K, M}, lb{B, N, K, N};
   launch_gemm_tn(la, lb, C, M, N, K, beta, s);
 }
@@ -493,6 +677,12 @@ void conv_wgrad_gemm(const unsigned short* eo, const unsigned short* x,
   const int Kdim = d.N * d.OH * d.OW;
   const int M = d.K;                   // n_kernels
   const int N = d.KH * d.KW * d.C;
+  if ((M & 7) == 0 && (d.C & 7) == 0) {
+    DenseTNv la{eo, d.K, Kdim, d.K};
+    Im2colTNv lb{x, d, Kdim, N};
+    launch_gemm_tn2(la, lb, gw, M, N, Kdim, beta, s);
+    return;
+  }
   DenseTN la{eo, d.K, Kdim, d.K};
   Im2colTN lb{x, d, Kdim, N};
   launch_gemm_tn(la, lb, gw, M, N, Kdim, beta, s);
