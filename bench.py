@@ -86,7 +86,8 @@ def main():
     dist_ctx = None
     if world > 1:
         from veles.znicz_amd.parallel import DistributedContext
-        dist_ctx = DistributedContext.create(backend="nccl")
+        dist_ctx = DistributedContext.create(
+            backend=os.environ.get("VELES_DIST_BACKEND", "nccl"))
 
     wf, batch = build_workflow(args, rank, world, device)
     if dist_ctx is not None:

@@ -72,6 +72,13 @@ class DistributedContext(object):
         import torch.distributed as dist
         if self.world == 1:
             return None
+        if self.backend == "gloo" and tensor.is_cuda:
+            # test-lane fallback (2 ranks sharing 1 GPU): gloo reduces
+            # host copies — correctness only, never the production path
+            host = tensor.detach().cpu()
+            dist.all_reduce(host)
+            tensor.copy_(host.to(tensor.device))
+            return None
         return dist.all_reduce(tensor, async_op=async_op)
 
     def barrier(self):
