@@ -84,6 +84,17 @@ __device__ __forceinline__ float uniform01(unsigned long long seed,
   return (float)(z >> 40) * (1.0f / 16777216.0f);  // 24-bit mantissa
 }
 
+// fast x^(-beta) for LRN: the AlexNet beta = 3/4 hits the exact-
+// instruction path (v_rsq + v_sqrt + v_mul) instead of exp2/log2
+__device__ __forceinline__ float pow_nbeta(float d, float beta,
+                                           bool b34) {
+  if (b34) {
+    float r = __builtin_amdgcn_rsqf(d);       // d^-1/2
+    return r * __builtin_amdgcn_sqrtf(r);     // d^-3/4
+  }
+  return __powf(d, -beta);
+}
+
 // host-side: log2 of the pow2 lane->channel tile (≤256) for NHWC kernels:
 // lanes cover channels with shifts/masks only — no per-element div/mod.
 static inline int chan_lg(int C) {

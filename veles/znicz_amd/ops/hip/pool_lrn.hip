@@ -353,7 +353,8 @@ template <int WIN>
 __global__ void lrn_fwd_k(const unsigned short* __restrict__ x,
                           unsigned short* __restrict__ y,
                           float* __restrict__ denom, unsigned rows, int C,
-                          int lgo, float alpha, float beta, float kk) {
+                          int lgo, float alpha, float beta, float kk,
+                          bool b34) {
   const int O2 = 1 << lgo;               // pow2 octet slots per row
   const int o0 = threadIdx.x & (O2 - 1);
   const int rin = threadIdx.x >> lgo;
@@ -396,7 +397,7 @@ __global__ void lrn_fwd_k(const unsigned short* __restrict__ x,
         }
         float d = kk + alpha * s;
         d8[j] = d;
-        o8[j] = f2bf(v[8 + j] * __powf(d, -beta));
+        o8[j] = f2bf(v[8 + j] * pow_nbeta(d, beta, b34));
       }
       if (cb + 8 <= C) {
         *(short8*)(yr + cb) = *(short8*)o8;
@@ -420,7 +421,7 @@ __global__ void lrn_bwd_pre_k(const unsigned short* __restrict__ ey,
                               const unsigned short* __restrict__ x,
                               const float* __restrict__ denom,
                               float* __restrict__ t, long total,
-                              float beta) {
+                              float beta, bool b34) {
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   long stride = (long)gridDim.x * blockDim.x;
   for (long p = i * 8; p < total; p += stride * 8) {
@@ -432,12 +433,15 @@ __global__ void lrn_bwd_pre_k(const unsigned short* __restrict__ ey,
 #pragma unroll
       for (int j = 0; j < 8; ++j)
         o[j] = bf2f(e8[j]) * bf2f(x8[j]) *
-               __powf(denom[p + j], -beta - 1.f);
+               pow_nbeta(denom[p + j], beta, b34) *
+               __builtin_amdgcn_rcpf(denom[p + j]);
       *(f32x4*)(t + p) = *(f32x4*)o;
       *(f32x4*)(t + p + 4) = *(f32x4*)(o + 4);
     } else {
       for (long q = p; q < total; ++q)
-        t[q] = bf2f(ey[q]) * bf2f(x[q]) * __powf(denom[q], -beta - 1.f);
+        t[q] = bf2f(ey[q]) * bf2f(x[q]) *
+               pow_nbeta(denom[q], beta, b34) *
+               __builtin_amdgcn_rcpf(denom[q]);
     }
   }
 }
@@ -450,7 +454,8 @@ __global__ void lrn_bwd_k(const unsigned short* __restrict__ ey,
                           const float* __restrict__ denom,
                           const float* __restrict__ t,
                           unsigned short* __restrict__ ex, unsigned rows,
-                          int C, int lgo, float alpha, float beta) {
+                          int C, int lgo, float alpha, float beta,
+                          bool b34) {
   const int O2 = 1 << lgo;
   const int o0 = threadIdx.x & (O2 - 1);
   const int rin = threadIdx.x >> lgo;
@@ -502,7 +507,7 @@ __global__ void lrn_bwd_k(const unsigned short* __restrict__ ey,
         float acc = 0.f;
 #pragma unroll
         for (int w = -half; w <= half; ++w) acc += tv[8 + j + w];
-        float v = bf2f(e8[j]) * __powf(dd[j], -beta) -
+        float v = bf2f(e8[j]) * pow_nbeta(dd[j], beta, b34) -
                   2.f * alpha * beta * bf2f(x8[j]) * acc;
         o8[j] = f2bf(v);
       }
@@ -542,9 +547,10 @@ void lrn_fwd_launch(const unsigned short* x, unsigned short* y,
   const int rpb = 256 >> lgo;
   dim3 grid((unsigned)min((rows + rpb - 1) / rpb, (long)131072)),
       block(256);
+  const bool b34 = beta == 0.75f;
 #define GO(W)                                                               \
   hipLaunchKernelGGL(lrn_fwd_k<W>, grid, block, 0, s, x, y, denom,          \
-                     (unsigned)rows, C, lgo, alpha, beta, k)
+                     (unsigned)rows, C, lgo, alpha, beta, k, b34)
   LRN_WIN_SWITCH(win, GO);
 #undef GO
   HIP_CHECK_LAST();
@@ -556,15 +562,16 @@ void lrn_bwd_launch(const unsigned short* ey, const unsigned short* x,
                     hipStream_t s) {
   long total = rows * C;
   int blocks = (int)min((total + 8 * 256 - 1) / (8 * 256), (long)4096);
+  const bool b34 = beta == 0.75f;
   hipLaunchKernelGGL(lrn_bwd_pre_k, dim3(blocks), dim3(256), 0, s, ey, x,
-                     denom, ws, total, beta);
+                     denom, ws, total, beta, b34);
   HIP_CHECK_LAST();
   const int lgo = lrn_lgo(C);
   const int rpb = 256 >> lgo;
   dim3 grid((unsigned)min((rows + rpb - 1) / rpb, (long)131072));
 #define GO(W)                                                               \
   hipLaunchKernelGGL(lrn_bwd_k<W>, grid, dim3(256), 0, s, ey, x, denom,     \
-                     ws, ex, (unsigned)rows, C, lgo, alpha, beta)
+                     ws, ex, (unsigned)rows, C, lgo, alpha, beta, b34)
   LRN_WIN_SWITCH(win, GO);
 #undef GO
   HIP_CHECK_LAST();
