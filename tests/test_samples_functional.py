@@ -86,3 +86,58 @@ def test_kohonen_map_spreads():
     wf.forward.output.map_read()
     # the 4 clusters must spread over several SOM cells, not collapse
     assert len(numpy.unique(wf.forward.output.mem)) >= 4
+
+
+@pytest.mark.timeout(600)
+def test_approximator_regression():
+    prng.seed_all(7)
+    from veles.znicz_amd.models.approximator import ApproximatorWorkflow
+    root.approximator.decision.update({"max_epochs": 6})
+    wf = ApproximatorWorkflow(DummyLauncher())
+    wf.initialize(device=NumpyDevice())
+    wf.run()
+    assert wf.is_finished
+    hist = wf.decision.mse_history
+    assert hist[-1] < 0.6 * hist[0], hist
+
+
+@pytest.mark.timeout(600)
+def test_lines_learns_orientation():
+    prng.seed_all(8)
+    from veles.znicz_amd.models.lines import LinesWorkflow
+    root.lines.decision.update({"max_epochs": 9})
+    wf = LinesWorkflow(DummyLauncher())
+    wf.initialize(device=NumpyDevice())
+    wf.run()
+    assert wf.is_finished
+    assert wf.decision.epoch_n_err_pt[TRAIN] < 20.0, \
+        wf.decision.epoch_n_err_pt
+
+
+@pytest.mark.timeout(600)
+def test_channels_classifier():
+    prng.seed_all(9)
+    from veles.znicz_amd.models.channels import ChannelsWorkflow
+    root.channels.decision.update({"max_epochs": 6})
+    root.channels.loader.update({"n_train": 200, "n_valid": 40})
+    wf = ChannelsWorkflow(DummyLauncher())
+    wf.initialize(device=NumpyDevice())
+    wf.run()
+    assert wf.is_finished
+    # position-jittered logos are genuinely hard for 6 epochs — assert
+    # steady learning, not mastery
+    assert wf.decision.epoch_n_err_pt[TRAIN] < 80.0, \
+        wf.decision.epoch_n_err_pt
+
+
+@pytest.mark.timeout(600)
+def test_video_ae():
+    prng.seed_all(10)
+    from veles.znicz_amd.models.video_ae import VideoAEWorkflow
+    root.video_ae.decision.update({"max_epochs": 6})
+    wf = VideoAEWorkflow(DummyLauncher())
+    wf.initialize(device=NumpyDevice())
+    wf.run()
+    assert wf.is_finished
+    hist = wf.decision.mse_history
+    assert hist[-1] < 0.85 * hist[0], hist
