@@ -139,3 +139,32 @@ def test_znicz_alias_unpickles_reference_paths():
     assert mod.All2All.__module__ == "veles.znicz_amd.all2all"
     s = pickle.dumps(mod.All2All)
     assert pickle.loads(s) is mod.All2All
+
+
+def test_uniform_unit_numpy():
+    from veles import prng
+    from veles.backends import NumpyDevice
+    from veles.dummy import DummyWorkflow
+    from veles.prng.uniform import Uniform
+    prng.get().seed(2)
+    u = Uniform(DummyWorkflow(), shape=(1000,), low=-1.0, high=1.0)
+    assert u.initialize(device=NumpyDevice()) is not False
+    u.run()
+    v = u.output.mem
+    assert -1.0 <= v.min() and v.max() <= 1.0
+    assert abs(v.mean()) < 0.1
+
+
+def test_web_status():
+    import json
+    import urllib.request
+    from veles.web_status import StatusServer
+    srv = StatusServer(port=0).start()
+    try:
+        srv.publish(epoch=3, err_pt=[1.0, 2.0, 3.0])
+        with urllib.request.urlopen(
+                "http://127.0.0.1:%d/" % srv.port, timeout=10) as r:
+            data = json.loads(r.read())
+        assert data["epoch"] == 3 and data["err_pt"] == [1.0, 2.0, 3.0]
+    finally:
+        srv.stop()
