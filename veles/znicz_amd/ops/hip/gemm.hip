@@ -411,7 +411,11 @@ __global__ void scale_f32_k(float* __restrict__ p, long n, float beta) {
 // ---------------------------------------------------------------------------
 
 #define TNP (BM + 2)                    // padded LDS row, elements
-#define TN2_LDS_OP (BK * TNP * 2)       // one operand image, bytes
+#define TBK 32                          // tn2 K-step: half of BK so the
+                                        // LDS fits 3-4 blocks/CU and the
+                                        // stage latency hides behind
+                                        // more resident waves
+#define TN2_LDS_OP (TBK * TNP * 2)      // one operand image, bytes
 #define TN2_LDS_BUF (TN2_LDS_OP * 2)    // A + B
 #define TN2_LDS_TOTAL (TN2_LDS_BUF * 2) // double-buffered
 
@@ -463,7 +467,7 @@ __global__ __launch_bounds__(256, 2) void gemm_tn2_k(
   const int tm = tile / tiles_n, tn = tile % tiles_n;
   const int m0 = tm * BM, n0 = tn * BN;
   const int wr = wave >> 1, wc = wave & 1;
-  const int nk = (K + BK - 1) / BK;
+  const int nk = (K + TBK - 1) / TBK;
   const int sp = gridDim.y, sid = blockIdx.y;
   const int t_lo = (int)(((long)nk * sid) / sp);
   const int t_hi = (int)(((long)nk * (sid + 1)) / sp);
@@ -472,11 +476,11 @@ __global__ __launch_bounds__(256, 2) void gemm_tn2_k(
   auto stage2 = [&](int buf, int kt) {
     char* As = smem + buf * TN2_LDS_BUF;
     char* Bs = As + TN2_LDS_OP;
-    const int k0t = kt * BK;
+    const int k0t = kt * TBK;
     const int oct = tid & 15;
     const int kk0 = tid >> 4;            // 16 k-rows per pass
 #pragma unroll
-    for (int pass = 0; pass < 4; ++pass) {
+    for (int pass = 0; pass < TBK / 16; ++pass) {
       const int kk = pass * 16 + kk0;
       unsigned short v[8];
       la.load16(k0t + kk, moA + oct, v);
@@ -495,11 +499,11 @@ __global__ __launch_bounds__(256, 2) void gemm_tn2_k(
     if (t + 1 < t_hi) stage2(buf ^ 1, t + 1);
     const unsigned short* As =
         (const unsigned short*)(smem + buf * TN2_LDS_BUF);
-    const unsigned short* Bs = As + BK * TNP;
+    const unsigned short* Bs = As + TBK * TNP;
     const int fr = lane & 15;
     const int ko8 = (lane >> 4) * 8;     // k-octet base within 32
 #pragma unroll
-    for (int ks = 0; ks < 2; ++ks) {
+    for (int ks = 0; ks < TBK / 32; ++ks) {
       short8 af[4], bf[4];
 #pragma unroll
       for (int i = 0; i < 4; ++i) {
@@ -602,7 +606,7 @@ void launch_gemm_tn2(const LA& la, const LB& lb, float* C, int M, int N,
                      int K, float beta, hipStream_t stream) {
   const int tn = cdiv(N, BN);
   const int tiles = cdiv(M, BM) * tn;
-  const int nk = cdiv(K, BK);
+  const int nk = cdiv(K, TBK);
   const int splits = max(1, min(nk, 1024 / tiles));
   if (splits > 1) {
     if (beta == 0.f) {
