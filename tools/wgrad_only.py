@@ -1,0 +1,14 @@
+import sys, os
+sys.path.insert(0, os.getcwd())
+import torch
+from veles.znicz_amd import ops
+N,H,W,C,K,KH,KW,SY,SX,PT,PL = 256,27,27,96,256,5,5,1,1,2,2
+OH=(H+2*PT-KH)//SY+1; OW=(W+2*PL-KW)//SX+1
+desc=[N,H,W,C,K,KH,KW,OH,OW,SY,SX,PT,PL]
+x=torch.randn(N,H,W,C,device="cuda").bfloat16()
+eo=torch.randn(N,OH,OW,K,device="cuda").bfloat16()
+gw=torch.zeros(K,KH*KW*C,device="cuda",dtype=torch.float32)
+for _ in range(30):
+    ops.conv_wgrad(eo.view(-1,K), x.view(N,-1), gw, desc, 0.0)
+torch.cuda.synchronize()
+print("done")
