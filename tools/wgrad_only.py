@@ -1,5 +1,5 @@
 import sys, os
-sys.path.insert(0, os.getcwd())
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch
 from veles.znicz_amd import ops
 N,H,W,C,K,KH,KW,SY,SX,PT,PL = 256,27,27,96,256,5,5,1,1,2,2
