@@ -30,6 +30,13 @@ def parse_args():
                    help="per-GPU minibatch (0 = model default)")
     p.add_argument("--unit-times", action="store_true",
                    help="print the per-unit time table to stderr")
+    p.add_argument("--target-loss", type=float, default=0.0,
+                   help="also run time-to-loss: train until the epoch "
+                        "train error %% drops below this, report secs")
+    p.add_argument("--kill-rank-after", type=int, default=0,
+                   help="fault injection: rank (world-1) aborts after "
+                        "N steps — exercises the emergency-snapshot / "
+                        "failure-detection path")
     return p.parse_args()
 
 
@@ -127,7 +134,14 @@ def main():
         step()
     barrier_sync()
     t0 = time.perf_counter()
-    for _ in range(args.steps):
+    for it in range(args.steps):
+        if args.kill_rank_after and rank == world - 1 and \
+                it == args.kill_rank_after:
+            # simulated rank death: the launcher path
+            # catches the collective failure and snapshots
+            print(json.dumps({"fault_injected": True, "rank": rank,
+                              "at_step": it}), flush=True)
+            os._exit(13)
         step()
     barrier_sync()
     elapsed = time.perf_counter() - t0
@@ -163,6 +177,25 @@ def main():
                        "input": list(sample[1:]),
                        "parallelism": "dp%d" % world},
         }))
+    if args.target_loss > 0:
+        tl0 = time.perf_counter()
+        from veles.loader.base import TRAIN
+        steps_done = 0
+        while (dec.epoch_n_err_pt[TRAIN] is None or
+               dec.epoch_n_err_pt[TRAIN] > args.target_loss) and \
+                steps_done < 100000 and not bool(dec.complete):
+            step()
+            steps_done += 1
+        barrier_sync()
+        if rank == 0:
+            print(json.dumps({
+                "metric": "time-to-loss %s (train err%% <= %g)" % (
+                    args.model, args.target_loss),
+                "value": round(time.perf_counter() - tl0, 3),
+                "unit": "seconds", "n_gpus": world,
+                "higher_is_better": False,
+                "final_err_pt": dec.epoch_n_err_pt[TRAIN],
+                "steps": steps_done}))
     if dist_ctx is not None:
         dist_ctx.finalize()
 

@@ -168,3 +168,43 @@ def test_web_status():
         assert data["epoch"] == 3 and data["err_pt"] == [1.0, 2.0, 3.0]
     finally:
         srv.stop()
+
+
+def _boom():
+    raise RuntimeError("injected rank failure")
+
+
+def test_emergency_snapshot(tmp_path):
+    """a failing run writes an emergency snapshot that can
+    be re-imported."""
+    from veles import prng
+    from veles.config import root
+    from veles.launcher import Launcher
+    from veles.snapshotter import SnapshotterToFile
+    root.common.dirs.snapshots = str(tmp_path)
+    prng.seed_all(4)
+    from veles.znicz_amd.models.wine import WineWorkflow
+    root.wine.decision.update({"max_epochs": 1})
+    root.wine.loader.update({"n_train": 30, "n_valid": 0})
+    launcher = Launcher(backend="numpy")
+    wf = WineWorkflow(launcher)
+
+    # inject a picklable failure into the decision unit
+    orig_run = wf.decision.run
+    wf.decision.run = _boom
+    with pytest.raises(RuntimeError, match="injected"):
+        launcher.boot()
+    del wf.decision.run                  # drop the instance override
+    snaps = list(tmp_path.glob("emergency_r0*.pickle.gz"))
+    assert len(snaps) == 1
+    wf2 = SnapshotterToFile.import_(str(snaps[0]))
+    assert type(wf2).__name__ == "WineWorkflow"
+    # resumed workflow re-initializes and completes with the fault gone
+    if "run" in wf2.decision.__dict__:
+        del wf2.decision.run
+    wf2.workflow = Launcher(backend="numpy")
+    wf2.workflow.workflow = wf2
+    wf2.initialize(device=__import__(
+        "veles.backends", fromlist=["NumpyDevice"]).NumpyDevice())
+    wf2.run()
+    assert wf2.is_finished

@@ -165,3 +165,38 @@ def test_kohonen_trains_on_gpu():
     assert wf.is_finished
     wf.forward.output.map_read()
     assert len(numpy.unique(wf.forward.output.mem)) >= 4
+
+
+@pytest.mark.timeout(300)
+def test_snapshot_resume_on_gpu(tmp_path):
+    """Snapshot a partially-trained GPU workflow, unpickle, re-initialize
+    on the device and continue — Arrays must round-trip host-only."""
+    import pickle
+    from veles.config import root
+    prng.seed_all(77)
+    from veles.znicz_amd.models.mnist import MnistWorkflow
+    root.mnist.decision.update({"max_epochs": 2})
+    root.mnist.loader.update({"n_train": 300, "n_valid": 60,
+                              "minibatch_size": 60})
+    wf = MnistWorkflow(DummyLauncher())
+    wf.initialize(device=ROCmDevice(0))
+    wf.run()
+    assert wf.is_finished
+    w_before = None
+    for f in wf.forwards:
+        f.weights.map_read()
+        w_before = f.weights.mem.copy()
+    blob = pickle.dumps(wf)
+    assert len(blob) > 1000
+    wf2 = pickle.loads(blob)
+    numpy.testing.assert_array_equal(
+        wf2.forwards[-1].weights.mem, w_before)
+    # resume for 2 more epochs on the device
+    wf2.workflow = DummyLauncher()
+    wf2.workflow.workflow = wf2
+    wf2.decision.max_epochs = 4
+    wf2.decision.complete <<= False
+    wf2.initialize(device=ROCmDevice(0))
+    wf2.run()
+    assert wf2.is_finished
+    assert int(wf2.loader.epoch_number) == 4

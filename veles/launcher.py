@@ -85,8 +85,35 @@ class Launcher(Logger):
             if wf.loader is not None:
                 wf.loader.shard(self.dist.rank, self.dist.world)
         wf.initialize(device=device, **kwargs)
-        wf.run()
+        try:
+            wf.run()
+        except BaseException as e:
+            # failure detection: a dead rank aborts the
+            # synchronous collective on the others — write an emergency
+            # snapshot so the job can relaunch (possibly at a smaller
+            # world) from the last consistent weights, then re-raise.
+            self.emergency_snapshot(reason=repr(e))
+            raise
         return wf
+
+    def emergency_snapshot(self, reason=""):
+        wf = self.workflow
+        if wf is None or self.stealth:
+            return None
+        try:
+            import os as _os
+            from veles.snapshotter import SnapshotterToFile
+            snap = SnapshotterToFile(
+                wf, prefix="emergency_r%d" % self.rank, compression="gz")
+            snap.suffix = "abort"
+            _os.makedirs(snap.directory, exist_ok=True)
+            path = snap.export()
+            self.warning("emergency snapshot after failure (%s): %s",
+                         reason, path)
+            return path
+        except Exception as e2:                      # never mask the
+            self.warning("emergency snapshot failed: %r", e2)  # original
+            return None
 
     def stop(self):
         if self._stopped:

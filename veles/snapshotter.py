@@ -87,8 +87,17 @@ class SnapshotterToFile(SnapshotterBase):
         fname = "%s%s.%d.pickle%s" % (self.prefix, suffix, self._counter,
                                       ext)
         path = os.path.join(self.directory, fname)
-        with opener(path, "wb") as f:
-            pickle.dump(wf, f, protocol=self.WRITE_PROTOCOL)
+        tmp = path + ".tmp"
+        try:
+            with opener(tmp, "wb") as f:
+                pickle.dump(wf, f, protocol=self.WRITE_PROTOCOL)
+        except BaseException:
+            try:
+                os.remove(tmp)           # never leave a truncated file
+            except OSError:
+                pass
+            raise
+        os.replace(tmp, path)
         self.destination = path
         self.info("snapshotted to %s", path)
         return path
