@@ -410,7 +410,10 @@ __global__ void scale_f32_k(float* __restrict__ p, long n, float beta) {
 // launcher falls back to v1 otherwise).
 // ---------------------------------------------------------------------------
 
-#define TNP (BM + 2)                    // padded LDS row, elements
+#define TNP (BM + 16)                   // padded LDS row: stride ≡ 8
+                                        // dwords (mod 64) ⇒ the 4 rows
+                                        // of a tr_b16 block hit
+                                        // disjoint bank octets
 #define TBK 32                          // tn2 K-step: half of BK so the
                                         // LDS fits 3-4 blocks/CU and the
                                         // stage latency hides behind
@@ -500,30 +503,48 @@ __global__ __launch_bounds__(256, 2) void gemm_tn2_k(
     const unsigned short* As =
         (const unsigned short*)(smem + buf * TN2_LDS_BUF);
     const unsigned short* Bs = As + TBK * TNP;
-    const int fr = lane & 15;
+    // hardware-transpose fragment reads (gfx950 ds_read_b64_tr_b16):
+    // lanes 4a..4a+3 of each 16-lane group fetch the 4 8-B chunks of
+    // image row kb+a cols [m0f, m0f+16); the group transpose hands
+    // lane j its m-column across the 4 k's — 2 reads per fragment
+    // instead of 8 scalar u16 reads (probe: tools/probe_trb16.hip).
+    const int gl = lane & 15;
+    const int a4 = gl >> 2;              // k-row within the quad
+    const int c4 = gl & 3;               // 8-B chunk within the row
     const int ko8 = (lane >> 4) * 8;     // k-octet base within 32
+    const unsigned As_b = (unsigned)(unsigned long long)As;
+    const unsigned Bs_b = (unsigned)(unsigned long long)Bs;
 #pragma unroll
     for (int ks = 0; ks < TBK / 32; ++ks) {
+      unsigned long long al[4], ah[4], bl[4], bh[4];
+      const int kb = ks * 32 + ko8;
+      const unsigned rowb = (unsigned)((kb + a4) * TNP + c4 * 4) * 2;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const unsigned ao = As_b + rowb +
+                            (unsigned)(wr * 64 + i * 16) * 2;
+        asm volatile("ds_read_b64_tr_b16 %0, %1"
+                     : "=v"(al[i]) : "v"(ao));
+        asm volatile("ds_read_b64_tr_b16 %0, %1"
+                     : "=v"(ah[i]) : "v"(ao + 4 * TNP * 2));
+        const unsigned bo = Bs_b + rowb +
+                            (unsigned)(wc * 64 + i * 16) * 2;
+        asm volatile("ds_read_b64_tr_b16 %0, %1"
+                     : "=v"(bl[i]) : "v"(bo));
+        asm volatile("ds_read_b64_tr_b16 %0, %1"
+                     : "=v"(bh[i]) : "v"(bo + 4 * TNP * 2));
+      }
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
       short8 af[4], bf[4];
 #pragma unroll
       for (int i = 0; i < 4; ++i) {
-        const int mrow = wr * 64 + i * 16 + fr;
-        const int kbase = ks * 32 + ko8;
-        unsigned short tmp[8];
-#pragma unroll
-        for (int j = 0; j < 8; ++j)
-          tmp[j] = As[(kbase + j) * TNP + mrow];
-        af[i] = *(short8*)tmp;
-      }
-#pragma unroll
-      for (int i = 0; i < 4; ++i) {
-        const int nrow = wc * 64 + i * 16 + fr;
-        const int kbase = ks * 32 + ko8;
-        unsigned short tmp[8];
-#pragma unroll
-        for (int j = 0; j < 8; ++j)
-          tmp[j] = Bs[(kbase + j) * TNP + nrow];
-        bf[i] = *(short8*)tmp;
+        unsigned long long q[2];
+        q[0] = al[i];
+        q[1] = ah[i];
+        af[i] = *(short8*)q;
+        q[0] = bl[i];
+        q[1] = bh[i];
+        bf[i] = *(short8*)q;
       }
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi)
