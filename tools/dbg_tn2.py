@@ -5,7 +5,7 @@ from veles.znicz_amd import ops
 RNG = numpy.random.RandomState(1)
 def dev_bf16(a): return torch.from_numpy(numpy.ascontiguousarray(a)).to("cuda", torch.bfloat16)
 def bfr(a): return dev_bf16(a).float().cpu().numpy()
-for (M,N,K) in [(64,64,32),(64,64,64),(128,128,32),(64,64,96),(256,128,512)]:
+for (M,N,K) in [(64,64,32),(64,64,64),(64,64,96),(64,64,128),(64,64,160),(64,64,192),(64,64,256),(1024,16384,64),(256,128,512)]:
     a = RNG.uniform(-1,1,(K,M)).astype(numpy.float32)
     b = RNG.uniform(-1,1,(K,N)).astype(numpy.float32)
     out = torch.zeros((M,N), dtype=torch.float32, device="cuda")
