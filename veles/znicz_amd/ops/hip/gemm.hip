@@ -503,48 +503,39 @@ __global__ __launch_bounds__(256, 2) void gemm_tn2_k(
     const unsigned short* As =
         (const unsigned short*)(smem + buf * TN2_LDS_BUF);
     const unsigned short* Bs = As + TBK * TNP;
-    // hardware-transpose fragment reads (gfx950 ds_read_b64_tr_b16):
-    // lanes 4a..4a+3 of each 16-lane group fetch the 4 8-B chunks of
-    // image row kb+a cols [m0f, m0f+16); the group transpose hands
-    // lane j its m-column across the 4 k's — 2 reads per fragment
-    // instead of 8 scalar u16 reads (probe: tools/probe_trb16.hip).
+    // hardware-transpose fragment reads (gfx950 ds_read_b64_tr_b16 via
+    // the compiler-tracked builtin — raw asm with truncated generic
+    // pointers + hand waitcnt raced non-deterministically under
+    // multi-block residency): lanes 4a..4a+3 of each 16-lane group
+    // fetch the 4 8-B chunks of image row kb+a cols [m0f, m0f+16); the
+    // group transpose hands lane j its m-column across the 4 k's — 2
+    // reads per fragment vs 8 scalar u16 (probe: tools/probe_trb16.hip)
+    typedef __attribute__((ext_vector_type(4))) short v4s;
+    typedef __attribute__((address_space(3))) v4s* lds_v4s_ptr;
     const int gl = lane & 15;
     const int a4 = gl >> 2;              // k-row within the quad
     const int c4 = gl & 3;               // 8-B chunk within the row
     const int ko8 = (lane >> 4) * 8;     // k-octet base within 32
-    const unsigned As_b = (unsigned)(unsigned long long)As;
-    const unsigned Bs_b = (unsigned)(unsigned long long)Bs;
 #pragma unroll
     for (int ks = 0; ks < TBK / 32; ++ks) {
-      unsigned long long al[4], ah[4], bl[4], bh[4];
       const int kb = ks * 32 + ko8;
-      const unsigned rowb = (unsigned)((kb + a4) * TNP + c4 * 4) * 2;
-#pragma unroll
-      for (int i = 0; i < 4; ++i) {
-        const unsigned ao = As_b + rowb +
-                            (unsigned)(wr * 64 + i * 16) * 2;
-        asm volatile("ds_read_b64_tr_b16 %0, %1"
-                     : "=v"(al[i]) : "v"(ao));
-        asm volatile("ds_read_b64_tr_b16 %0, %1"
-                     : "=v"(ah[i]) : "v"(ao + 4 * TNP * 2));
-        const unsigned bo = Bs_b + rowb +
-                            (unsigned)(wc * 64 + i * 16) * 2;
-        asm volatile("ds_read_b64_tr_b16 %0, %1"
-                     : "=v"(bl[i]) : "v"(bo));
-        asm volatile("ds_read_b64_tr_b16 %0, %1"
-                     : "=v"(bh[i]) : "v"(bo + 4 * TNP * 2));
-      }
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      const int rowe = (kb + a4) * TNP + c4 * 4;
       short8 af[4], bf[4];
 #pragma unroll
       for (int i = 0; i < 4; ++i) {
-        unsigned long long q[2];
-        q[0] = al[i];
-        q[1] = ah[i];
-        af[i] = *(short8*)q;
-        q[0] = bl[i];
-        q[1] = bh[i];
-        bf[i] = *(short8*)q;
+        const unsigned short* ap = As + rowe + (wr * 64 + i * 16);
+        v4s lo = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+            (lds_v4s_ptr)ap);
+        v4s hi = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+            (lds_v4s_ptr)(ap + 4 * TNP));
+        *(v4s*)&af[i] = lo;
+        *((v4s*)&af[i] + 1) = hi;
+        const unsigned short* bp = Bs + rowe + (wc * 64 + i * 16);
+        lo = __builtin_amdgcn_ds_read_tr16_b64_v4i16((lds_v4s_ptr)bp);
+        hi = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+            (lds_v4s_ptr)(bp + 4 * TNP));
+        *(v4s*)&bf[i] = lo;
+        *((v4s*)&bf[i] + 1) = hi;
       }
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi)
