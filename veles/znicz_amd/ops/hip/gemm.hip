@@ -142,25 +142,32 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_k(
 
   const int nk = (K + BK - 1) / BK;
 
+  // Slot swizzle: slot s of row r holds k-octet s ^ (r&7) ^ ((r>>3)&1).
+  // The glds DMA destination is lane-linear (cannot pad rows), so the
+  // b128 fragment reads of 16 consecutive rows would hit 128-B-strided
+  // addresses ≡ 0 (mod 64 dw) — an 8-way bank conflict measured at
+  // 13-16% of NT wave cycles.  Swizzling WHICH k-octet each slot loads
+  // spreads the 16 lanes of a fragment read over distinct banks.
   auto stage = [&](int buf, int kt) {
     char* As = smem + buf * NT_LDS_BUF;
     char* Bs = As + NT_LDS_A;
     const int k8base = kt * (BK / 8);
     const int rl = lane >> 3;          // row within 8-row chunk
-    const int k8 = lane & 7;           // 16-B chunk within the 64-k row
-    const int kc0 = (k8base + k8) * 8;
-    const bool straddle = (kc0 < K) & (kc0 + 8 > K);  // K%8 tail chunk
+    const int k8 = lane & 7;           // 16-B slot within the 64-k row
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
       const int chunk = wave * 4 + i;  // 16 chunks x 8 rows = 128 rows
       const int row = chunk * 8 + rl;
+      const int k8e = k8 ^ rl ^ (chunk & 1);   // swizzled source octet
+      const int kc0 = (k8base + k8e) * 8;
+      const bool straddle = (kc0 < K) & (kc0 + 8 > K);
       if (!straddle) {
-        const void* sa = la.addr(min(m0 + row, M - 1), k8base + k8);
+        const void* sa = la.addr(min(m0 + row, M - 1), k8base + k8e);
         __builtin_amdgcn_global_load_lds(
             (const __attribute__((address_space(1))) void*)sa,
             (__attribute__((address_space(3))) void*)(As + chunk * 1024),
             16, 0, 0);
-        const void* sb = lb.addr(min(n0 + row, N - 1), k8base + k8);
+        const void* sb = lb.addr(min(n0 + row, N - 1), k8base + k8e);
         __builtin_amdgcn_global_load_lds(
             (const __attribute__((address_space(1))) void*)sb,
             (__attribute__((address_space(3))) void*)(Bs + chunk * 1024),
@@ -190,18 +197,22 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_k(
     const char* As = smem + buf * NT_LDS_BUF;
     const char* Bs = As + NT_LDS_A;
     const int fr = lane & 15;          // fragment row/col
-    const int fo = (lane >> 4) * 16;   // k-octet byte offset
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
+      const int ko = ks * 4 + (lane >> 4);   // k-octet 0..7
       short8 af[4], bf[4];
 #pragma unroll
-      for (int i = 0; i < 4; ++i)
-        af[i] = *(const short8*)(As + (wr * 64 + i * 16 + fr) * 128 +
-                                 ks * 64 + fo);
+      for (int i = 0; i < 4; ++i) {
+        const int row = wr * 64 + i * 16 + fr;
+        const int slot = ko ^ (row & 7) ^ ((row >> 3) & 1);
+        af[i] = *(const short8*)(As + row * 128 + slot * 16);
+      }
 #pragma unroll
-      for (int i = 0; i < 4; ++i)
-        bf[i] = *(const short8*)(Bs + (wc * 64 + i * 16 + fr) * 128 +
-                                 ks * 64 + fo);
+      for (int i = 0; i < 4; ++i) {
+        const int row = wc * 64 + i * 16 + fr;
+        const int slot = ko ^ (row & 7) ^ ((row >> 3) & 1);
+        bf[i] = *(const short8*)(Bs + row * 128 + slot * 16);
+      }
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
