@@ -124,6 +124,41 @@ struct DgradNT {
   }
 };
 
+// stride-1 dgrad specialization: SY == SX == 1 removes the per-address
+// stride divisibility checks and divisions of the generic DgradNT.
+struct DgradNT1 {
+  const unsigned short* eo;
+  const unsigned short* zero;
+  ConvDesc d;
+  __device__ const void* addr(int row, int k8) const {
+    int k = k8 * 8;
+    int co = k % d.K, t = k / d.K;
+    int kw = t % d.KW, kh = t / d.KW;
+    int iw = row % d.W;
+    t = row / d.W;
+    int ih = t % d.H, n = t / d.H;
+    int oh = ih + d.PT - kh;
+    int ow = iw + d.PL - kw;
+    if (kh >= d.KH || (unsigned)oh >= (unsigned)d.OH ||
+        (unsigned)ow >= (unsigned)d.OW)
+      return zero;
+    return eo + (((long)n * d.OH + oh) * d.OW + ow) * d.K + co;
+  }
+  __device__ unsigned short elem(int row, int k) const {
+    int co = k % d.K, t = k / d.K;
+    int kw = t % d.KW, kh = t / d.KW;
+    int iw = row % d.W;
+    t = row / d.W;
+    int ih = t % d.H, n = t / d.H;
+    int oh = ih + d.PT - kh;
+    int ow = iw + d.PL - kw;
+    if (kh >= d.KH || (unsigned)oh >= (unsigned)d.OH ||
+        (unsigned)ow >= (unsigned)d.OW)
+      return 0;
+    return eo[(((long)n * d.OH + oh) * d.OW + ow) * d.K + co];
+  }
+};
+
 // ---------------------------------------------------------------------------
 // core NT
 // ---------------------------------------------------------------------------
@@ -693,8 +728,14 @@ void conv_dgrad_gemm(const unsigned short* eo, const unsigned short* wdg,
   const int M = d.N * d.H * d.W;       // input pixels
   const int N = d.C;                   // input channels
   const int K = d.KH * d.KW * d.K;     // co-inner gather dim
-  DgradNT la{eo, zero, d};
   DenseNT lb{wdg, zero, K, K};         // wdg: [C][KH][KW][K]
+  if (d.SY == 1 && d.SX == 1) {
+    DgradNT1 la{eo, zero, d};
+    launch_gemm_nt(la, lb, ei, (const float*)nullptr, M, N, K,
+                   ACT_LINEAR, s);
+    return;
+  }
+  DgradNT la{eo, zero, d};
   launch_gemm_nt(la, lb, ei, (const float*)nullptr, M, N, K, ACT_LINEAR, s);
 }
 
