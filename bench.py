@@ -175,6 +175,10 @@ def main():
             "config": {"model": args.model,
                        "global_batch": batch * world,
                        "input": list(sample[1:]),
+                       # RGB padded to 8 channels for the NHWC HIP conv
+                       # DMA constraint — the kernels do MORE work than
+                       # the 3-channel baseline config requires
+                       "channels_padded_from": 3,
                        "parallelism": "dp%d" % world},
         }))
     if args.target_loss > 0:
