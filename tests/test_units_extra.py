@@ -471,3 +471,42 @@ def test_diversity():
     w[3] = w[0] * 2.0 + 0.01            # near-duplicate of kernel 0
     groups = get_similar_kernels(w, threshold=0.9)
     assert any(set(g) >= {0, 3} for g in groups)
+
+
+def test_standard_workflow_optional_links(tmp_path):
+    """link_lr_adjuster / link_error_plotter / link_weights_plotter /
+    link_image_saver — the reference's optional StandardWorkflow surface
+    wired into a real training run."""
+    from veles import prng
+    from veles.backends import NumpyDevice
+    from veles.config import root
+    from veles.dummy import DummyLauncher
+    from veles.znicz_amd import lr_adjust as la
+    import veles.plotter as vp
+    root.common.dirs.plots = str(tmp_path / "plots")
+    vp.GraphicsServer._instance = None
+    prng.seed_all(4)
+    from veles.znicz_amd.models.wine import WineWorkflow
+    root.wine.decision.update({"max_epochs": 2})
+    root.wine.loader.update({"n_train": 60, "n_valid": 30})
+    wf = WineWorkflow(DummyLauncher())
+    adj = wf.link_lr_adjuster(wf.decision,
+                              lr_policy=la.exp_policy(0.05, 0.9))
+    ep = wf.link_error_plotter(adj)
+    wp = wf.link_weights_plotter(ep)
+    sv = wf.link_image_saver(
+        wp, out_dirs=[str(tmp_path / d) for d in ("t", "v", "tr")])
+    sv.link_attrs(wf.loader, ("input", "minibatch_data"),
+                  ("labels", "minibatch_labels"),
+                  ("indices", "minibatch_indices"),
+                  "minibatch_class", "minibatch_size")
+    sv.link_attrs(wf.forwards[-1], "max_idx")
+    wf.repeater.link_from(sv)
+    wf.initialize(device=NumpyDevice())
+    wf.run()
+    assert wf.is_finished
+    assert adj.iteration > 0
+    # lr decayed from the policy
+    assert wf.gds[0].learning_rate < 0.05
+    sink = tmp_path / "plots" / "metrics.jsonl"
+    assert sink.exists() and sink.read_text().strip()
