@@ -15,7 +15,6 @@ time.  ``build_program``/``get_kernel``/``execute_kernel`` are kept as
 thin shims for API parity.
 """
 
-from veles.backends import NumpyDevice
 from veles.memory import Array
 from veles.units import Unit
 from veles.workflow import Workflow

@@ -13,8 +13,6 @@ all-reduced across ranks with one tiny collective (the replacement for
 the master-side merge in reference §3.3).
 """
 
-import numpy
-
 from veles.loader.base import TRAIN, VALID, TEST, CLASS_NAME
 from veles.mutable import Bool
 from veles.units import Unit

@@ -5,8 +5,6 @@ Parity target: reference ``znicz/samples/Kohonen`` /
 clouds with hits/U-matrix plotters.
 """
 
-import numpy
-
 from veles.config import root
 from veles.znicz_amd.decision import DecisionBase
 from veles.znicz_amd.kohonen import KohonenForward, KohonenTrainer

@@ -6,8 +6,6 @@ graph is wired explicitly (the AE units need offset / shape cross-links
 that the declarative ``layers=[...]`` schema does not express).
 """
 
-import numpy
-
 from veles.config import root
 from veles.znicz_amd.conv import Conv
 from veles.znicz_amd.deconv import Deconv, GDDeconv

@@ -8,7 +8,6 @@ phase statistics, CD weight update, reconstruction-MSE decision.
 import numpy
 
 from veles.config import root
-from veles.memory import Array
 from veles.znicz_amd.activation import ActivationSigmoid
 from veles.znicz_amd.all2all import All2AllSigmoid
 from veles.znicz_amd.decision import DecisionMSE

@@ -8,7 +8,6 @@ functional tests assert that trajectory.
 
 import numpy
 
-from veles import prng
 from veles.loader.fullbatch import (FullBatchLoader, FullBatchLoaderMSE)
 
 

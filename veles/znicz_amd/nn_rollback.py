@@ -6,8 +6,6 @@ restore the best snapshot of every registered weights/bias Array and
 multiply the GD units' learning rates by ``lr_cut_ratio``.
 """
 
-import numpy
-
 from veles.units import Unit
 
 

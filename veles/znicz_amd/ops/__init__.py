@@ -11,8 +11,6 @@ Build: ``python veles/znicz_amd/ops/setup.py build_ext --inplace`` (or
 ``__graft_entry__.build()`` at the repo root).
 """
 
-import os
-
 _ext = None
 _ext_err = None
 
@@ -168,9 +166,7 @@ def attach_conv_copies(weights_arr):
 
 
 def refresh_conv_copies(weights_arr):
-    import torch
     w = weights_arr.devmem          # fp32 master, already updated
-    k = w.shape[0]
     weights_arr.bf16_dgrad.copy_(
         weights_arr.bf16.view(w.shape).permute(3, 1, 2, 0))
 

@@ -11,8 +11,6 @@ Layer spec (reference schema): ``{"type": "<mapping>", "->": {forward
 kwargs}, "<-": {gd kwargs}, ...shared kwargs...}``.
 """
 
-from veles.loader.base import TRAIN
-from veles.mutable import Bool
 from veles.units import Unit
 from veles.znicz_amd.decision import DecisionGD, DecisionMSE
 from veles.znicz_amd.evaluator import EvaluatorSoftmax, EvaluatorMSE
