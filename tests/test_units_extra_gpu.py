@@ -200,3 +200,33 @@ def test_snapshot_resume_on_gpu(tmp_path):
     wf2.run()
     assert wf2.is_finished
     assert int(wf2.loader.epoch_number) == 4
+
+
+def test_confusion_matrix_device_accum():
+    """GPU confusion accumulation matches the numpy path."""
+    from veles.znicz_amd import ops
+    C, batch = 7, 500
+    rng = numpy.random.RandomState(3)
+    preds = rng.randint(0, C, batch).astype(numpy.int32)
+    labels = rng.randint(0, C, batch).astype(numpy.int32)
+    cm = torch.zeros(C, C, dtype=torch.int32, device="cuda")
+    ops.confusion_accum(torch.from_numpy(preds).cuda(),
+                        torch.from_numpy(labels).cuda(), cm, batch)
+    want = numpy.zeros((C, C), numpy.int32)
+    for p, l in zip(preds, labels):
+        want[p, l] += 1
+    numpy.testing.assert_array_equal(cm.cpu().numpy(), want)
+
+
+def test_mnist_confusion_on_gpu():
+    prng.seed_all(11)
+    from veles.znicz_amd.models.mnist import MnistWorkflow
+    root.mnist.decision.update({"max_epochs": 1})
+    root.mnist.loader.update({"n_train": 300, "n_valid": 0,
+                              "minibatch_size": 60})
+    wf = MnistWorkflow(DummyLauncher())
+    wf.initialize(device=ROCmDevice(0))
+    wf.run()
+    cm = wf.evaluator.confusion_matrix
+    cm.map_read()
+    assert cm.mem.sum() == 300       # every trained sample counted

@@ -7,8 +7,8 @@ err = y − target, metrics).
 
 MI355X path: one fused kernel (``ops.eval_softmax``) computes the error
 tensor (bf16) and the device-side n_err counter in a single pass; the
-confusion matrix is accumulated on host from ``max_idx`` (epoch-rate,
-not step-rate, so it never syncs the hot loop).
+confusion matrix is accumulated by a device-side atomic kernel, so the
+hot loop never syncs.
 """
 
 import numpy
@@ -70,6 +70,8 @@ class EvaluatorSoftmax(EvaluatorBase):
             self.max_err_output_sum.reset(numpy.zeros(
                 1, dtype=numpy.float32))
         self.init_array(self.n_err, device_dtype="i32")
+        if self.compute_confusion_matrix and self.confusion_matrix:
+            self.init_array(self.confusion_matrix, device_dtype="i32")
 
     def numpy_run(self):
         self.input.map_read()
@@ -110,15 +112,12 @@ class EvaluatorSoftmax(EvaluatorBase):
         self.err_output.touch_device()
         self.n_err.touch_device()
         if self.compute_confusion_matrix and self.confusion_matrix and \
-                self.max_idx is not None:
-            # host-side, cheap (reads small int arrays)
-            self.max_idx.map_read()
-            self.labels.map_read()
-            mi = self.max_idx.mem
-            lb = self.labels.mem
-            for i in range(batch):
-                if lb[i] >= 0:
-                    self.confusion_matrix.mem[int(mi[i]), int(lb[i])] += 1
+                self.max_idx is not None and \
+                self.max_idx.devmem is not None:
+            # device-side accumulation — the hot loop never syncs
+            ops.confusion_accum(self.max_idx.devmem, labels_dev,
+                                self.confusion_matrix.devmem, batch)
+            self.confusion_matrix.touch_device()
 
 
 class EvaluatorMSE(EvaluatorBase):

@@ -186,6 +186,11 @@ def eval_softmax(probs, labels, err_out, n_err, batch):
     require().eval_softmax(probs, labels, err_out, n_err, int(batch))
 
 
+def confusion_accum(max_idx, labels, cm, batch):
+    """cm[pred, label] += 1 on device (int32) — no host sync."""
+    require().confusion_accum(max_idx, labels, cm, int(batch))
+
+
 def eval_mse(y, target, err_out, mse, metrics, batch, root):
     require().eval_mse(y, target, err_out, mse, metrics, int(batch),
                        int(root))

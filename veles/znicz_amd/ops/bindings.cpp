@@ -41,6 +41,8 @@ void fill_uniform_launch(float*, long, float, float, unsigned long long,
 void row_argmin_launch(const float*, int*, int, int, hipStream_t);
 void mean_disp_launch(const unsigned short*, const float*, const float*,
                       unsigned short*, long, long, hipStream_t);
+void confusion_launch(const int*, const int*, int*, int, int,
+                      hipStream_t);
 void dropout_fwd_launch(const unsigned short*, unsigned short*,
                         unsigned char*, long, float, unsigned long long,
                         hipStream_t);
@@ -299,9 +301,17 @@ void mean_disp(torch::Tensor x, torch::Tensor mean, torch::Tensor rdisp,
                    mean.numel(), cur_stream());
 }
 
+void confusion_accum(torch::Tensor max_idx, torch::Tensor labels,
+                     torch::Tensor cm, int64_t batch) {
+  confusion_launch(max_idx.data_ptr<int>(), labels.data_ptr<int>(),
+                   cm.data_ptr<int>(), (int)batch, cm.size(0),
+                   cur_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("confusion_accum", &confusion_accum);
   m.def("gather_by_offset", &gather_by_offset);
   m.def("scatter_by_offset", &scatter_by_offset);
   m.def("bernoulli", &bernoulli);

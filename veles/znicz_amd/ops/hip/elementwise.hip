@@ -431,6 +431,26 @@ void mean_disp_launch(const unsigned short* x, const float* mean,
   HIP_CHECK_LAST();
 }
 
+// confusion-matrix accumulation: cm[pred, label] += 1, device-side so
+// the hot loop never syncs (int32 atomics on a tiny [C,C] matrix)
+__global__ void confusion_k(const int* __restrict__ max_idx,
+                            const int* __restrict__ labels,
+                            int* __restrict__ cm, int batch, int C) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= batch) return;
+  int lbl = labels[i];
+  if (lbl < 0 || lbl >= C) return;
+  int pred = max_idx[i];
+  atomicAdd(&cm[pred * C + lbl], 1);
+}
+
+void confusion_launch(const int* max_idx, const int* labels, int* cm,
+                      int batch, int C, hipStream_t s) {
+  hipLaunchKernelGGL(confusion_k, dim3((batch + 255) / 256), dim3(256),
+                     0, s, max_idx, labels, cm, batch, C);
+  HIP_CHECK_LAST();
+}
+
 // -- dropout ---------------------------------------------------------------
 __global__ void dropout_fwd_k(const unsigned short* __restrict__ x,
                               unsigned short* __restrict__ y,
