@@ -248,10 +248,9 @@ void lrn_bwd(torch::Tensor ey, torch::Tensor x,
              int64_t win) {
   const int C = x.size(-1);
   (void)y; (void)k;
-  auto ws = torch::empty_like(denom);  // fp32 t_j workspace (pass 1)
-  lrn_bwd_launch(ubf(ey), ubf(x), denom.data_ptr<float>(),
-                 ws.data_ptr<float>(), ubf_mut(ex), x.numel() / C, C,
-                 (float)alpha, (float)beta, (int)win, cur_stream());
+  lrn_bwd_launch(ubf(ey), ubf(x), denom.data_ptr<float>(), nullptr,
+                 ubf_mut(ex), x.numel() / C, C, (float)alpha,
+                 (float)beta, (int)win, cur_stream());
 }
 
 void dropout_fwd(torch::Tensor x, torch::Tensor y, torch::Tensor mask,

@@ -415,44 +415,14 @@ __global__ void lrn_fwd_k(const unsigned short* __restrict__ x,
   }
 }
 
-// backward in two fully-parallel passes (ws = fp32 workspace, numel total):
-//   pass 1:  t_j = dy_j·x_j·D_j^{-β-1}   (vectorized 8-wide)
-__global__ void lrn_bwd_pre_k(const unsigned short* __restrict__ ey,
-                              const unsigned short* __restrict__ x,
-                              const float* __restrict__ denom,
-                              float* __restrict__ t, long total,
-                              float beta, bool b34) {
-  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  long stride = (long)gridDim.x * blockDim.x;
-  for (long p = i * 8; p < total; p += stride * 8) {
-    if (p + 8 <= total) {
-      unsigned short e8[8], x8[8];
-      *(short8*)e8 = *(const short8*)(ey + p);
-      *(short8*)x8 = *(const short8*)(x + p);
-      float o[8];
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        o[j] = bf2f(e8[j]) * bf2f(x8[j]) *
-               pow_nbeta(denom[p + j], beta, b34) *
-               __builtin_amdgcn_rcpf(denom[p + j]);
-      *(f32x4*)(t + p) = *(f32x4*)o;
-      *(f32x4*)(t + p + 4) = *(f32x4*)(o + 4);
-    } else {
-      for (long q = p; q < total; ++q)
-        t[q] = bf2f(ey[q]) * bf2f(x[q]) *
-               pow_nbeta(denom[q], beta, b34) *
-               __builtin_amdgcn_rcpf(denom[q]);
-    }
-  }
-}
-
-//   pass 2:  dx_c = dy_c·D_c^{-β} − 2αβ·x_c · Σ_{|j−c|≤half} t_j
-// (octet-vectorized like the forward; t halo zero-padded = clipping)
+//   fused backward:  dx_c = dy_c·D_c^{-β} − 2αβ·x_c · Σ_{|j−c|≤half} t_j
+// with t_j = dy_j·x_j·D_j^{-β-1} recomputed in registers for the octet
+// plus its halo — no fp32 workspace round-trip (a separate pre-pass
+// cost a full extra read+write of the tensor).
 template <int WIN>
 __global__ void lrn_bwd_k(const unsigned short* __restrict__ ey,
                           const unsigned short* __restrict__ x,
                           const float* __restrict__ denom,
-                          const float* __restrict__ t,
                           unsigned short* __restrict__ ex, unsigned rows,
                           int C, int lgo, float alpha, float beta,
                           bool b34) {
@@ -467,48 +437,50 @@ __global__ void lrn_bwd_k(const unsigned short* __restrict__ ey,
     const unsigned short* er = ey + (long)row * C;
     const unsigned short* xr = x + (long)row * C;
     const float* dr = denom + (long)row * C;
-    const float* tr = t + (long)row * C;
     unsigned short* or_ = ex + (long)row * C;
     for (int oc = o0; oc < noct; oc += O2) {
       const int cb = oc * 8;
-      float tv[24];
+      float ev[24], xv[24], dv[24];
 #pragma unroll
       for (int h = 0; h < 3; ++h) {
         const int c8 = cb + (h - 1) * 8;
         if (c8 >= 0 && c8 + 8 <= C) {
-          *(f32x4*)(tv + h * 8) = *(const f32x4*)(tr + c8);
-          *(f32x4*)(tv + h * 8 + 4) = *(const f32x4*)(tr + c8 + 4);
+          unsigned short e8[8], x8[8];
+          *(short8*)e8 = *(const short8*)(er + c8);
+          *(short8*)x8 = *(const short8*)(xr + c8);
+          *(f32x4*)(dv + h * 8) = *(const f32x4*)(dr + c8);
+          *(f32x4*)(dv + h * 8 + 4) = *(const f32x4*)(dr + c8 + 4);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            ev[h * 8 + j] = bf2f(e8[j]);
+            xv[h * 8 + j] = bf2f(x8[j]);
+          }
         } else {
 #pragma unroll
           for (int j = 0; j < 8; ++j) {
             int c = c8 + j;
-            tv[h * 8 + j] = (c >= 0 && c < C) ? tr[c] : 0.f;
+            bool ok = c >= 0 && c < C;
+            ev[h * 8 + j] = ok ? bf2f(er[c]) : 0.f;
+            xv[h * 8 + j] = ok ? bf2f(xr[c]) : 0.f;
+            dv[h * 8 + j] = ok ? dr[c] : 1.f;
           }
         }
       }
-      unsigned short e8[8], x8[8], o8[8];
-      float dd[8];
-      if (cb + 8 <= C) {
-        *(short8*)e8 = *(const short8*)(er + cb);
-        *(short8*)x8 = *(const short8*)(xr + cb);
-        *(f32x4*)dd = *(const f32x4*)(dr + cb);
-        *(f32x4*)(dd + 4) = *(const f32x4*)(dr + cb + 4);
-      } else {
+      float tv[8 + 2 * half];
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          int c = cb + j;
-          e8[j] = c < C ? er[c] : 0;
-          x8[j] = c < C ? xr[c] : 0;
-          dd[j] = c < C ? dr[c] : 1.f;
-        }
+      for (int j = 0; j < 8 + 2 * half; ++j) {
+        const int q = 8 - half + j;          // window [cb-half, cb+7+half]
+        tv[j] = ev[q] * xv[q] * pow_nbeta(dv[q], beta, b34) *
+                __builtin_amdgcn_rcpf(dv[q]);
       }
+      unsigned short o8[8];
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         float acc = 0.f;
 #pragma unroll
-        for (int w = -half; w <= half; ++w) acc += tv[8 + j + w];
-        float v = bf2f(e8[j]) * pow_nbeta(dd[j], beta, b34) -
-                  2.f * alpha * beta * bf2f(x8[j]) * acc;
+        for (int w = 0; w < WIN; ++w) acc += tv[j + w];
+        float v = ev[8 + j] * pow_nbeta(dv[8 + j], beta, b34) -
+                  2.f * alpha * beta * xv[8 + j] * acc;
         o8[j] = f2bf(v);
       }
       if (cb + 8 <= C) {
@@ -560,18 +532,14 @@ void lrn_bwd_launch(const unsigned short* ey, const unsigned short* x,
                     const float* denom, float* ws, unsigned short* ex,
                     long rows, int C, float alpha, float beta, int win,
                     hipStream_t s) {
-  long total = rows * C;
-  int blocks = (int)min((total + 8 * 256 - 1) / (8 * 256), (long)4096);
+  (void)ws;                       // fused kernel needs no workspace
   const bool b34 = beta == 0.75f;
-  hipLaunchKernelGGL(lrn_bwd_pre_k, dim3(blocks), dim3(256), 0, s, ey, x,
-                     denom, ws, total, beta, b34);
-  HIP_CHECK_LAST();
   const int lgo = lrn_lgo(C);
   const int rpb = 256 >> lgo;
   dim3 grid((unsigned)min((rows + rpb - 1) / rpb, (long)131072));
 #define GO(W)                                                               \
   hipLaunchKernelGGL(lrn_bwd_k<W>, grid, dim3(256), 0, s, ey, x, denom,     \
-                     ws, ex, (unsigned)rows, C, lgo, alpha, beta, b34)
+                     ex, (unsigned)rows, C, lgo, alpha, beta, b34)
   LRN_WIN_SWITCH(win, GO);
 #undef GO
   HIP_CHECK_LAST();
