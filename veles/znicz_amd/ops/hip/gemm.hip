@@ -175,7 +175,13 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_k(
   const int m0 = tm * BM, n0 = tn * BN;
   const int wr = wave >> 1, wc = wave & 1;   // 2x2 waves of 64x64
 
-  const int nk = (K + BK - 1) / BK;
+  // split-K (gridDim.y slices; fp32-out atomic epilogue) fills the
+  // chip on short-and-wide FC shapes (batch 256 x N<=4096: 16-64 MN
+  // tiles) — bias/activation run in a separate tiny pass.
+  const int nk_all = (K + BK - 1) / BK;
+  const int sp = gridDim.y, sid = blockIdx.y;
+  const int t_lo = (int)(((long)nk_all * sid) / sp);
+  const int nk = (int)(((long)nk_all * (sid + 1)) / sp);
 
   // Slot swizzle: slot s of row r holds k-octet s ^ (r&7) ^ ((r>>3)&1).
   // The glds DMA destination is lane-linear (cannot pad rows), so the
@@ -224,10 +230,11 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_k(
 
   f32x4 acc[4][4] = {};
 
-  stage(0, 0);
+  if (t_lo >= nk) return;
+  stage(0, t_lo);
   __syncthreads();
   int buf = 0;
-  for (int t = 0; t < nk; ++t) {
+  for (int t = t_lo; t < nk; ++t) {
     if (t + 1 < nk) stage(buf ^ 1, t + 1);
     const char* As = smem + buf * NT_LDS_BUF;
     const char* Bs = As + NT_LDS_A;
@@ -259,7 +266,8 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_k(
     buf ^= 1;
   }
 
-  // epilogue: bias + activation + store (row = (l>>4)*4+r, col = l&15)
+  // epilogue: bias + activation + store (row = (l>>4)*4+r, col = l&15);
+  // split-K partials accumulate raw fp32 (bias/act applied later)
   const int colw = n0 + wc * 64 + (lane & 15);
 #pragma unroll
   for (int ni = 0; ni < 4; ++ni) {
@@ -273,6 +281,11 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_k(
       for (int r = 0; r < 4; ++r) {
         const int row = rbase + r;
         if (row >= M) continue;
+        if (sp > 1) {
+          unsafeAtomicAdd((float*)&C[(long)row * N + col],
+                          acc[mi][ni][r]);
+          continue;
+        }
         float v = act_fwd<ACT>(acc[mi][ni][r] + bv);
         if constexpr (sizeof(OUT_T) == 2)
           C[(long)row * N + col] = (OUT_T)f2bf(v);
@@ -280,6 +293,24 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_k(
           C[(long)row * N + col] = (OUT_T)v;
       }
     }
+  }
+}
+
+// split-K finisher: out = act(ws + bias), bf16 or fp32 out
+template <typename OUT_T, int ACT>
+__global__ void bias_act_cast_k(const float* __restrict__ ws,
+                                OUT_T* __restrict__ out,
+                                const float* __restrict__ bias, long n,
+                                int N) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long p = i; p < n; p += stride) {
+    float v = ws[p] + (bias ? bias[p % N] : 0.f);
+    v = act_fwd<ACT>(v);
+    if constexpr (sizeof(OUT_T) == 2)
+      out[p] = (OUT_T)f2bf(v);
+    else
+      out[p] = (OUT_T)v;
   }
 }
 
@@ -690,6 +721,45 @@ void gemm_nt_dense(const unsigned short* A, const unsigned short* B,
                    const unsigned short* zero, int M, int N, int K,
                    int act, hipStream_t s) {
   DenseNT la{A, zero, K, K}, lb{B, zero, K, K};
+  const int tiles = cdiv(M, BM) * cdiv(N, BN);
+  const int nk = cdiv(K, BK);
+  const int splits = max(1, min(nk, 512 / tiles));
+  if (splits > 1) {
+    // fp32 workspace + atomic partials, then one bias/act/cast pass
+    static float* ws = nullptr;
+    static size_t cap = 0;
+    size_t need = (size_t)cdiv(M, BM) * BM * cdiv(N, BN) * BN *
+                  sizeof(float);
+    if (need > cap) {
+      if (ws) (void)hipFree(ws);
+      cap = need < (32u << 20) ? (32u << 20) : need;
+      if (hipMalloc(&ws, cap) != hipSuccess)
+        throw std::runtime_error("gemm_nt split workspace alloc");
+    }
+    void zero_f32_launch(float*, long, hipStream_t);
+    zero_f32_launch(ws, (long)M * N, s);
+    dim3 grid(tiles, splits), block(256);
+    hipLaunchKernelGGL((gemm_nt_k<DenseNT, DenseNT, float, ACT_LINEAR>),
+                       grid, block, NT_LDS_TOTAL, s, la, lb, ws,
+                       (const float*)nullptr, M, N, K, cdiv(N, BN));
+    HIP_CHECK_LAST();
+    long n = (long)M * N;
+    int b = (int)min((n + 255) / 256, (long)2048);
+#define FIN(A_)                                                              \
+    do {                                                                     \
+      if (c_is_f32)                                                          \
+        hipLaunchKernelGGL((bias_act_cast_k<float, A_>), dim3(b),            \
+                           dim3(256), 0, s, ws, (float*)C, bias, n, N);      \
+      else                                                                   \
+        hipLaunchKernelGGL((bias_act_cast_k<unsigned short, A_>), dim3(b),   \
+                           dim3(256), 0, s, ws, (unsigned short*)C, bias,    \
+                           n, N);                                            \
+    } while (0)
+    ACT_SWITCH(act, FIN);
+#undef FIN
+    HIP_CHECK_LAST();
+    return;
+  }
   if (c_is_f32)
     launch_gemm_nt(la, lb, (float*)C, bias, M, N, K, act, s);
   else
