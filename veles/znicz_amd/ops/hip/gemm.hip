@@ -159,6 +159,17 @@ struct DgradNT1 {
   }
 };
 
+static __device__ __forceinline__ int xcd_tile(int b, int ntiles) {
+  // blocks go to XCD (b % 8); remap so tiles of one XCD are contiguous
+  // (per-XCD L2 reuse of the shared B/A panels) — speed-only, any
+  // mapping is correct.
+  if (ntiles < 64) return b;
+  const int per = ntiles >> 3;            // tiles per XCD (floor)
+  const int cut = per << 3;
+  if (b >= cut) return b;                 // remainder keeps identity
+  return (b & 7) * per + (b >> 3);
+}
+
 // ---------------------------------------------------------------------------
 // core NT
 // ---------------------------------------------------------------------------
@@ -170,7 +181,7 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_k(
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const int tid = threadIdx.x;
   const int wave = tid >> 6, lane = tid & 63;
-  const int tile = blockIdx.x;
+  const int tile = xcd_tile(blockIdx.x, gridDim.x);
   const int tm = tile / tiles_n, tn = tile % tiles_n;
   const int m0 = tm * BM, n0 = tn * BN;
   const int wr = wave >> 1, wc = wave & 1;   // 2x2 waves of 64x64
@@ -382,7 +393,7 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_k(
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const int tid = threadIdx.x;
   const int wave = tid >> 6, lane = tid & 63;
-  const int tile = blockIdx.x;
+  const int tile = xcd_tile(blockIdx.x, gridDim.x);
   const int tm = tile / tiles_n, tn = tile % tiles_n;
   const int m0 = tm * BM, n0 = tn * BN;
   const int wr = wave >> 1, wc = wave & 1;
@@ -543,7 +554,7 @@ __global__ __launch_bounds__(256, 2) void gemm_tn2_k(
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const int tid = threadIdx.x;
   const int wave = tid >> 6, lane = tid & 63;
-  const int tile = blockIdx.x;
+  const int tile = xcd_tile(blockIdx.x, gridDim.x);
   const int tm = tile / tiles_n, tn = tile % tiles_n;
   const int m0 = tm * BM, n0 = tn * BN;
   const int wr = wave >> 1, wc = wave & 1;
