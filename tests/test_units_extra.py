@@ -510,3 +510,39 @@ def test_standard_workflow_optional_links(tmp_path):
     assert wf.gds[0].learning_rate < 0.05
     sink = tmp_path / "plots" / "metrics.jsonl"
     assert sink.exists() and sink.read_text().strip()
+
+
+def test_accumulate_gradient_fc_and_conv():
+    """accumulate_gradient=True sums per-step gradients instead of
+    overwriting (reference znicz gd option)."""
+    from veles.znicz_amd.all2all import All2All
+    from veles.znicz_amd.gd import GradientDescent
+    wf = DummyWorkflow()
+    dev = make_dev()
+    prng.get().seed(9)
+    fwd = All2All(wf, output_sample_shape=5)
+    fwd.input = Array(RNG.uniform(-1, 1, (4, 6)).astype(numpy.float32))
+    fwd.input.initialize(dev)
+    assert fwd.initialize(device=dev) is not False
+    fwd.run()
+    gd = GradientDescent(wf, apply_gradient=False,
+                         accumulate_gradient=True, need_err_input=False)
+    gd.input = fwd.input
+    gd.output = fwd.output
+    gd.weights = fwd.weights
+    gd.bias = fwd.bias
+    gd.err_output = Array(RNG.uniform(-1, 1, (4, 5)).astype(
+        numpy.float32))
+    gd.err_output.initialize(dev)
+    assert gd.initialize(device=dev) is not False
+    gd.batch_size = 4
+    gd.run()
+    gd.gradient_weights.map_read()
+    g1 = gd.gradient_weights.mem.copy()
+    b1 = gd.gradient_bias.mem.copy()
+    gd.run()
+    gd.gradient_weights.map_read()
+    numpy.testing.assert_allclose(gd.gradient_weights.mem, 2 * g1,
+                                  rtol=1e-5)
+    numpy.testing.assert_allclose(gd.gradient_bias.mem, 2 * b1,
+                                  rtol=1e-5)

@@ -230,3 +230,45 @@ def test_mnist_confusion_on_gpu():
     cm = wf.evaluator.confusion_matrix
     cm.map_read()
     assert cm.mem.sum() == 300       # every trained sample counted
+
+
+def test_accumulate_gradient_gpu():
+    """HIP path: gradient_weights/bias double after two accumulated
+    backward passes (beta=1 wgrad + accumulating col_sum)."""
+    from veles.znicz_amd.all2all import All2All
+    from veles.znicz_amd.gd import GradientDescent
+    wf = DummyWorkflow()
+    dev = ROCmDevice(0)
+    prng.get().seed(9)
+    fwd = All2All(wf, output_sample_shape=40)
+    fwd.input = Array(RNG.uniform(-1, 1, (16, 24)).astype(numpy.float32))
+    fwd.input.device_dtype = "bf16"
+    fwd.input.initialize(dev)
+    assert fwd.initialize(device=dev) is not False
+    fwd.run()
+    gd = GradientDescent(wf, apply_gradient=False,
+                         accumulate_gradient=True, need_err_input=False)
+    gd.input = fwd.input
+    gd.output = fwd.output
+    gd.weights = fwd.weights
+    gd.bias = fwd.bias
+    gd.err_output = Array(RNG.uniform(-1, 1, (16, 40)).astype(
+        numpy.float32))
+    gd.err_output.device_dtype = "bf16"
+    gd.err_output.initialize(dev)
+    assert gd.initialize(device=dev) is not False
+    gd.batch_size = 16
+    gd.run()
+    gd.gradient_weights.map_read()
+    gd.gradient_bias.map_read()
+    g1 = gd.gradient_weights.mem.copy()
+    b1 = gd.gradient_bias.mem.copy()
+    gd.gradient_weights.unmap()
+    gd.gradient_bias.unmap()
+    gd.run()
+    gd.gradient_weights.map_read()
+    gd.gradient_bias.map_read()
+    numpy.testing.assert_allclose(gd.gradient_weights.mem, 2 * g1,
+                                  rtol=2e-2, atol=1e-3)
+    numpy.testing.assert_allclose(gd.gradient_bias.mem, 2 * b1,
+                                  rtol=2e-2, atol=1e-3)

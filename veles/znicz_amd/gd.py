@@ -94,13 +94,15 @@ class GradientDescent(GradientDescentBase):
             ops.gemm_nt(eo, w_dgrad, ei)
             self.err_input.touch_device()
         x = ops.as2d_bf16(self.input)
+        beta = 1.0 if self.accumulate_gradient else 0.0
         if self.weights_transposed:   # gw[in, out] = xᵀ · eo
-            ops.gemm_tn(x, eo, self.gradient_weights.devmem)
+            ops.gemm_tn(x, eo, self.gradient_weights.devmem, beta=beta)
         else:                         # gw[out, in] = eoᵀ · x
-            ops.gemm_tn(eo, x, self.gradient_weights.devmem)
+            ops.gemm_tn(eo, x, self.gradient_weights.devmem, beta=beta)
         self.gradient_weights.touch_device()
         if self.include_bias and self.bias is not None and self.bias:
-            ops.col_sum(eo, self.gradient_bias.devmem)
+            ops.col_sum(eo, self.gradient_bias.devmem,
+                        accumulate=self.accumulate_gradient)
             self.gradient_bias.touch_device()
         self._hip_apply(batch)
 

@@ -120,11 +120,13 @@ class GradientDescentConv(GradientDescentBase):
                            self.err_input.devmem, desc)
             self.err_input.touch_device()
         ops.conv_wgrad(eo, self.input.devmem,
-                       self.gradient_weights.devmem, desc)
+                       self.gradient_weights.devmem, desc,
+                       beta=1.0 if self.accumulate_gradient else 0.0)
         self.gradient_weights.touch_device()
         if self.include_bias and self.bias is not None and self.bias:
             K = fwd.n_kernels
-            ops.col_sum(eo.view(-1, K), self.gradient_bias.devmem)
+            ops.col_sum(eo.view(-1, K), self.gradient_bias.devmem,
+                        accumulate=self.accumulate_gradient)
             self.gradient_bias.touch_device()
         self._hip_apply(batch)
 

@@ -192,8 +192,12 @@ class GradientDescentBase(NNLayerBase):
         """grad_w: already summed over batch (not normalized)."""
         batch = max(1, self.current_batch_size)
         g = grad_w / batch
-        self.gradient_weights.map_invalidate()
-        self.gradient_weights.mem[...] = g
+        if self.accumulate_gradient and self.gradient_weights:
+            self.gradient_weights.map_write()
+            self.gradient_weights.mem += g
+        else:
+            self.gradient_weights.map_invalidate()
+            self.gradient_weights.mem[...] = g
         if self.apply_gradient:
             self.weights.map_write()
             self.accumulated_gradient_weights.map_write()
@@ -207,8 +211,12 @@ class GradientDescentBase(NNLayerBase):
             return
         batch = max(1, self.current_batch_size)
         g = grad_b / batch
-        self.gradient_bias.map_invalidate()
-        self.gradient_bias.mem[...] = g
+        if self.accumulate_gradient and self.gradient_bias:
+            self.gradient_bias.map_write()
+            self.gradient_bias.mem += g
+        else:
+            self.gradient_bias.map_invalidate()
+            self.gradient_bias.mem[...] = g
         if self.apply_gradient:
             self.bias.map_write()
             self.accumulated_gradient_bias.map_write()

@@ -130,9 +130,9 @@ def act_fwd_inplace(x, activation):
     require().act_fwd(x, int(activation))
 
 
-def col_sum(a, out):
-    """out[N] (fp32) = Σ_rows a[M,N] (bf16)."""
-    require().col_sum(a, out)
+def col_sum(a, out, accumulate=False):
+    """out[N] (fp32) = (out +) Σ_rows a[M,N] (bf16)."""
+    require().col_sum(a, out, bool(accumulate))
 
 
 def sgd_step(param_arr, grad, vel, lr, decay, l1_vs_l2, moment, scale):

@@ -27,7 +27,8 @@ void act_bwd_launch(unsigned short*, const unsigned short*, long, int,
                     hipStream_t);
 void sgd_launch(float*, const float*, float*, unsigned short*, long, float,
                 float, float, float, float, hipStream_t);
-void col_sum_launch(const unsigned short*, float*, int, int, hipStream_t);
+void col_sum_launch(const unsigned short*, float*, int, int, bool,
+                    hipStream_t);
 void transpose2d_launch(const unsigned short*, unsigned short*, int, int,
                         hipStream_t);
 void gather_off_launch(const unsigned short*, const int*, unsigned short*,
@@ -184,9 +185,9 @@ void sgd_step(torch::Tensor w, torch::Tensor g, torch::Tensor vel,
   }
 }
 
-void col_sum(torch::Tensor a, torch::Tensor out) {
+void col_sum(torch::Tensor a, torch::Tensor out, bool accumulate) {
   col_sum_launch(ubf(a), out.data_ptr<float>(), a.size(0), a.size(1),
-                 cur_stream());
+                 accumulate, cur_stream());
 }
 
 void transpose2d(torch::Tensor a, torch::Tensor out) {

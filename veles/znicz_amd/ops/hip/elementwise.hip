@@ -121,7 +121,7 @@ void zero_f32_launch(float* p, long n, hipStream_t s) {
 // shapes (M ≈ 10⁵..10⁶, N ≤ a few hundred).
 __global__ void col_sum_k(const unsigned short* __restrict__ a,
                           float* __restrict__ ws, int M, int N,
-                          int rows_per_slab, int lgo) {
+                          int rows_per_slab, int lgo, int accumulate) {
   const int O2 = 1 << lgo;
   const int oct = threadIdx.x & (O2 - 1);
   const int rsub = threadIdx.x >> lgo;
@@ -163,7 +163,8 @@ __global__ void col_sum_k(const unsigned short* __restrict__ a,
     float* dst = ws + (long)blockIdx.x * N + cb;  // ws==out at 1 slab
 #pragma unroll
     for (int j = 0; j < 8; ++j)
-      if (cb + j < N) dst[j] = s[j];
+      if (cb + j < N)
+        dst[j] = (accumulate && gridDim.x == 1 ? dst[j] : 0.f) + s[j];
   }
 }
 
@@ -173,7 +174,8 @@ __global__ void col_sum_k(const unsigned short* __restrict__ a,
 // serialized on vmcnt and cost ~0.2 µs PER SLAB (the measured flat
 // ~200 µs floor of the whole col_sum) — see tools/probe_colsum.hip.
 __global__ void col_sum_fin_k(const float* __restrict__ ws,
-                              float* __restrict__ out, int slabs, int N) {
+                              float* __restrict__ out, int slabs, int N,
+                              int accumulate) {
   const int cb = blockIdx.x * 8;
   const int j = threadIdx.x & 7;        // column within octet
   const int r0 = threadIdx.x >> 3;      // slab stripe (32 lanes)
@@ -191,14 +193,15 @@ __global__ void col_sum_fin_k(const float* __restrict__ ws,
     if (threadIdx.x < step) red[threadIdx.x] += red[threadIdx.x + step];
     __syncthreads();
   }
-  if (threadIdx.x < 8 && col < N) out[col] = red[threadIdx.x];
+  if (threadIdx.x < 8 && col < N)
+    out[col] = (accumulate ? out[col] : 0.f) + red[threadIdx.x];
 }
 
 // wide-N fallback (N ≥ 2048 exceeds one block's octet span; those
 // matrices are short FC-bias shapes where column parallelism suffices)
 __global__ void col_sum_wide_k(const unsigned short* __restrict__ a,
                                float* __restrict__ out, int M, int N,
-                               int rows_per_slab) {
+                               int rows_per_slab, int accumulate) {
   int col = blockIdx.x * blockDim.x + threadIdx.x;
   if (col >= N) return;
   long m0 = (long)blockIdx.y * rows_per_slab;
@@ -207,20 +210,20 @@ __global__ void col_sum_wide_k(const unsigned short* __restrict__ a,
   const unsigned short* p = a + m0 * N + col;
   for (long m = m0; m < m1; ++m, p += N) s += bf2f(*p);
   if (gridDim.y == 1)
-    out[col] = s;
+    out[col] = (accumulate ? out[col] : 0.f) + s;
   else
     unsafeAtomicAdd(&out[col], s);
 }
 
 void col_sum_launch(const unsigned short* a, float* out, int M, int N,
-                    hipStream_t s) {
+                    bool accumulate, hipStream_t s) {
   if (N >= 2048) {
     int bx = (N + 255) / 256;
     int slabs = (int)min((long)(M + 63) / 64, (long)max(1, 1024 / bx));
-    if (slabs > 1) zero_f32_launch(out, N, s);
+    if (slabs > 1 && !accumulate) zero_f32_launch(out, N, s);
     int rps = (M + slabs - 1) / slabs;
     hipLaunchKernelGGL(col_sum_wide_k, dim3(bx, slabs), dim3(256), 0, s,
-                       a, out, M, N, rps);
+                       a, out, M, N, rps, (int)accumulate);
     HIP_CHECK_LAST();
     return;
   }
@@ -230,7 +233,7 @@ void col_sum_launch(const unsigned short* a, float* out, int M, int N,
   int slabs = (int)min((long)(M + 63) / 64, (long)512);
   if (slabs == 1) {
     hipLaunchKernelGGL(col_sum_k, dim3(1), dim3(256), 0, s, a,
-                       out, M, N, M, lgo);
+                       out, M, N, M, lgo, (int)accumulate);
     HIP_CHECK_LAST();
     return;
   }
@@ -246,10 +249,10 @@ void col_sum_launch(const unsigned short* a, float* out, int M, int N,
   }
   int rps = (M + slabs - 1) / slabs;
   hipLaunchKernelGGL(col_sum_k, dim3(slabs), dim3(256), 0, s, a,
-                     ws, M, N, rps, lgo);
+                     ws, M, N, rps, lgo, 0);
   HIP_CHECK_LAST();
   hipLaunchKernelGGL(col_sum_fin_k, dim3((N + 7) / 8), dim3(256), 0,
-                     s, ws, out, slabs, N);
+                     s, ws, out, slabs, N, (int)accumulate);
   HIP_CHECK_LAST();
 }
 
