@@ -272,3 +272,41 @@ def test_accumulate_gradient_gpu():
                                   rtol=2e-2, atol=1e-3)
     numpy.testing.assert_allclose(gd.gradient_bias.mem, 2 * b1,
                                   rtol=2e-2, atol=1e-3)
+
+
+@pytest.mark.timeout(600)
+def test_remaining_samples_on_gpu():
+    """Every sample family steps on the HIP path (lines / channels /
+    approximator / video_ae / wine / vgg one short run each)."""
+    specs = [
+        ("veles.znicz_amd.models.lines", "LinesWorkflow", "lines", 2),
+        ("veles.znicz_amd.models.channels", "ChannelsWorkflow",
+         "channels", 2),
+        ("veles.znicz_amd.models.approximator", "ApproximatorWorkflow",
+         "approximator", 2),
+        ("veles.znicz_amd.models.video_ae", "VideoAEWorkflow",
+         "video_ae", 2),
+        ("veles.znicz_amd.models.wine", "WineWorkflow", "wine", 2),
+    ]
+    import importlib
+    for mod, cls, cfg, epochs in specs:
+        prng.seed_all(12)
+        m = importlib.import_module(mod)
+        getattr(root, cfg).decision.update({"max_epochs": epochs})
+        wf = getattr(m, cls)(DummyLauncher())
+        wf.initialize(device=ROCmDevice(0))
+        wf.run()
+        assert wf.is_finished, cfg
+
+
+@pytest.mark.timeout(600)
+def test_vgg_short_on_gpu():
+    prng.seed_all(13)
+    from veles.znicz_amd.models.vgg import VGG16Workflow
+    root.vgg16.loader.update({"minibatch_size": 16, "n_train": 32,
+                              "n_valid": 16})
+    root.vgg16.decision.update({"max_epochs": 1})
+    wf = VGG16Workflow(DummyLauncher())
+    wf.initialize(device=ROCmDevice(0))
+    wf.run()
+    assert wf.is_finished

@@ -125,7 +125,8 @@ class Unit(Logger, Distributable):
                 mine = theirs = arg
             # an existing instance attribute would shadow the link
             self.__dict__.pop(mine, None)
-            table[mine] = LinkableAttribute(other, theirs, two_way=True)
+            table[mine] = LinkableAttribute(other, theirs,
+                                            two_way=True)
         return self
 
     def unlink_attrs(self, *names):
