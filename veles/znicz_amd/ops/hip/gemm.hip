@@ -34,141 +34,7 @@
 // operand loaders (per-lane global source address generators)
 // ---------------------------------------------------------------------------
 
-// dense row-major [rows][ld], K-innermost; k past Klim reads the zero page
-struct DenseNT {
-  const unsigned short* base;
-  const unsigned short* zero;
-  long ld;
-  int Klim;
-  __device__ const void* addr(int row, int k8) const {
-    int k = k8 * 8;
-    return k < Klim ? (const void*)(base + (long)row * ld + k)
-                    : (const void*)zero;
-  }
-  // single-element access for the K%8 tail chunk
-  __device__ unsigned short elem(int row, int k) const {
-    return k < Klim ? base[(long)row * ld + k] : (unsigned short)0;
-  }
-};
-
-// implicit im2col over NHWC input: row=(n,oh,ow), kdim=(kh,kw,c) c-inner.
-// Requires C % 8 == 0 so each 16-B chunk stays inside one pixel's C-run.
-struct Im2colNT {
-  const unsigned short* x;
-  const unsigned short* zero;
-  ConvDesc d;
-  __device__ const void* addr(int row, int k8) const {
-    int k = k8 * 8;
-    int c = k % d.C, t = k / d.C;
-    int kw = t % d.KW, kh = t / d.KW;
-    int ow = row % d.OW;
-    t = row / d.OW;
-    int oh = t % d.OH, n = t / d.OH;
-    int ih = oh * d.SY - d.PT + kh;
-    int iw = ow * d.SX - d.PL + kw;
-    if (kh >= d.KH || (unsigned)ih >= (unsigned)d.H ||
-        (unsigned)iw >= (unsigned)d.W)
-      return zero;
-    return x + (((long)n * d.H + ih) * d.W + iw) * d.C + c;
-  }
-  __device__ unsigned short elem(int row, int k) const {
-    // only hit when K%8 != 0 (conv K-dims are C-padded so never here)
-    int c = k % d.C, t = k / d.C;
-    int kw = t % d.KW, kh = t / d.KW;
-    int ow = row % d.OW;
-    t = row / d.OW;
-    int oh = t % d.OH, n = t / d.OH;
-    int ih = oh * d.SY - d.PT + kh;
-    int iw = ow * d.SX - d.PL + kw;
-    if (kh >= d.KH || (unsigned)ih >= (unsigned)d.H ||
-        (unsigned)iw >= (unsigned)d.W)
-      return 0;
-    return x[(((long)n * d.H + ih) * d.W + iw) * d.C + c];
-  }
-};
-
-// conv dgrad gather over NHWC err_output: row=(n,ih,iw) of the FORWARD
-// input, kdim=(kh,kw,co) co-inner; strided positions read the zero page.
-struct DgradNT {
-  const unsigned short* eo;
-  const unsigned short* zero;
-  ConvDesc d;
-  __device__ const void* addr(int row, int k8) const {
-    int k = k8 * 8;
-    int co = k % d.K, t = k / d.K;
-    int kw = t % d.KW, kh = t / d.KW;
-    int iw = row % d.W;
-    t = row / d.W;
-    int ih = t % d.H, n = t / d.H;
-    int ohn = ih + d.PT - kh;
-    int own = iw + d.PL - kw;
-    if (kh >= d.KH || ohn < 0 || own < 0 || ohn % d.SY || own % d.SX)
-      return zero;
-    int oh = ohn / d.SY, ow = own / d.SX;
-    if (oh >= d.OH || ow >= d.OW) return zero;
-    return eo + (((long)n * d.OH + oh) * d.OW + ow) * d.K + co;
-  }
-  __device__ unsigned short elem(int row, int k) const {
-    int co = k % d.K, t = k / d.K;
-    int kw = t % d.KW, kh = t / d.KW;
-    int iw = row % d.W;
-    t = row / d.W;
-    int ih = t % d.H, n = t / d.H;
-    int ohn = ih + d.PT - kh;
-    int own = iw + d.PL - kw;
-    if (kh >= d.KH || ohn < 0 || own < 0 || ohn % d.SY || own % d.SX)
-      return 0;
-    int oh = ohn / d.SY, ow = own / d.SX;
-    if (oh >= d.OH || ow >= d.OW) return 0;
-    return eo[(((long)n * d.OH + oh) * d.OW + ow) * d.K + co];
-  }
-};
-
-// stride-1 dgrad specialization: SY == SX == 1 removes the per-address
-// stride divisibility checks and divisions of the generic DgradNT.
-struct DgradNT1 {
-  const unsigned short* eo;
-  const unsigned short* zero;
-  ConvDesc d;
-  __device__ const void* addr(int row, int k8) const {
-    int k = k8 * 8;
-    int co = k % d.K, t = k / d.K;
-    int kw = t % d.KW, kh = t / d.KW;
-    int iw = row % d.W;
-    t = row / d.W;
-    int ih = t % d.H, n = t / d.H;
-    int oh = ih + d.PT - kh;
-    int ow = iw + d.PL - kw;
-    if (kh >= d.KH || (unsigned)oh >= (unsigned)d.OH ||
-        (unsigned)ow >= (unsigned)d.OW)
-      return zero;
-    return eo + (((long)n * d.OH + oh) * d.OW + ow) * d.K + co;
-  }
-  __device__ unsigned short elem(int row, int k) const {
-    int co = k % d.K, t = k / d.K;
-    int kw = t % d.KW, kh = t / d.KW;
-    int iw = row % d.W;
-    t = row / d.W;
-    int ih = t % d.H, n = t / d.H;
-    int oh = ih + d.PT - kh;
-    int ow = iw + d.PL - kw;
-    if (kh >= d.KH || (unsigned)oh >= (unsigned)d.OH ||
-        (unsigned)ow >= (unsigned)d.OW)
-      return 0;
-    return eo[(((long)n * d.OH + oh) * d.OW + ow) * d.K + co];
-  }
-};
-
-static __device__ __forceinline__ int xcd_tile(int b, int ntiles) {
-  // blocks go to XCD (b % 8); remap so tiles of one XCD are contiguous
-  // (per-XCD L2 reuse of the shared B/A panels) — speed-only, any
-  // mapping is correct.
-  if (ntiles < 64) return b;
-  const int per = ntiles >> 3;            // tiles per XCD (floor)
-  const int cut = per << 3;
-  if (b >= cut) return b;                 // remainder keeps identity
-  return (b & 7) * per + (b >> 3);
-}
+#include "loaders.h"
 
 // ---------------------------------------------------------------------------
 // core NT
@@ -789,12 +655,23 @@ void gemm_tn_dense(const unsigned short* A, const unsigned short* B,
   launch_gemm_tn(la, lb, C, M, N, K, beta, s);
 }
 
+void conv_fwd_gemm8(const unsigned short*, const unsigned short*, void*,
+                    bool, const float*, const unsigned short*,
+                    const ConvDesc&, int, hipStream_t);
+void conv_dgrad_gemm8(const unsigned short*, const unsigned short*,
+                      unsigned short*, const unsigned short*,
+                      const ConvDesc&, hipStream_t);
+
 void conv_fwd_gemm(const unsigned short* x, const unsigned short* w,
                    void* out, bool out_f32, const float* bias,
                    const unsigned short* zero, const ConvDesc& d, int act,
                    hipStream_t s) {
   const int M = d.N * d.OH * d.OW;
   const int K = d.KH * d.KW * d.C;
+  if (M >= 2048 && d.K >= 192 && (K & 7) == 0) {
+    conv_fwd_gemm8(x, w, out, out_f32, bias, zero, d, act, s);
+    return;
+  }
   Im2colNT la{x, zero, d};
   DenseNT lb{w, zero, K, K};
   if (out_f32)
@@ -809,6 +686,10 @@ void conv_dgrad_gemm(const unsigned short* eo, const unsigned short* wdg,
   const int M = d.N * d.H * d.W;       // input pixels
   const int N = d.C;                   // input channels
   const int K = d.KH * d.KW * d.K;     // co-inner gather dim
+  if (M >= 2048 && N >= 192 && (K & 7) == 0) {
+    conv_dgrad_gemm8(eo, wdg, ei, zero, d, s);
+    return;
+  }
   DenseNT lb{wdg, zero, K, K};         // wdg: [C][KH][KW][K]
   if (d.SY == 1 && d.SX == 1) {
     DgradNT1 la{eo, zero, d};

@@ -1,0 +1,142 @@
+// Per-lane global source-address generators for the NT GEMM cores
+// (shared by the 128x128 2-barrier core in gemm.hip and the 256x256
+// 8-phase core in gemm8.hip).
+#pragma once
+#include "common.h"
+
+// dense row-major [rows][ld], K-innermost; k past Klim reads the zero page
+struct DenseNT {
+  const unsigned short* base;
+  const unsigned short* zero;
+  long ld;
+  int Klim;
+  __device__ const void* addr(int row, int k8) const {
+    int k = k8 * 8;
+    return k < Klim ? (const void*)(base + (long)row * ld + k)
+                    : (const void*)zero;
+  }
+  // single-element access for the K%8 tail chunk
+  __device__ unsigned short elem(int row, int k) const {
+    return k < Klim ? base[(long)row * ld + k] : (unsigned short)0;
+  }
+};
+
+// implicit im2col over NHWC input: row=(n,oh,ow), kdim=(kh,kw,c) c-inner.
+// Requires C % 8 == 0 so each 16-B chunk stays inside one pixel's C-run.
+struct Im2colNT {
+  const unsigned short* x;
+  const unsigned short* zero;
+  ConvDesc d;
+  __device__ const void* addr(int row, int k8) const {
+    int k = k8 * 8;
+    int c = k % d.C, t = k / d.C;
+    int kw = t % d.KW, kh = t / d.KW;
+    int ow = row % d.OW;
+    t = row / d.OW;
+    int oh = t % d.OH, n = t / d.OH;
+    int ih = oh * d.SY - d.PT + kh;
+    int iw = ow * d.SX - d.PL + kw;
+    if (kh >= d.KH || (unsigned)ih >= (unsigned)d.H ||
+        (unsigned)iw >= (unsigned)d.W)
+      return zero;
+    return x + (((long)n * d.H + ih) * d.W + iw) * d.C + c;
+  }
+  __device__ unsigned short elem(int row, int k) const {
+    // only hit when K%8 != 0 (conv K-dims are C-padded so never here)
+    int c = k % d.C, t = k / d.C;
+    int kw = t % d.KW, kh = t / d.KW;
+    int ow = row % d.OW;
+    t = row / d.OW;
+    int oh = t % d.OH, n = t / d.OH;
+    int ih = oh * d.SY - d.PT + kh;
+    int iw = ow * d.SX - d.PL + kw;
+    if (kh >= d.KH || (unsigned)ih >= (unsigned)d.H ||
+        (unsigned)iw >= (unsigned)d.W)
+      return 0;
+    return x[(((long)n * d.H + ih) * d.W + iw) * d.C + c];
+  }
+};
+
+// conv dgrad gather over NHWC err_output: row=(n,ih,iw) of the FORWARD
+// input, kdim=(kh,kw,co) co-inner; strided positions read the zero page.
+struct DgradNT {
+  const unsigned short* eo;
+  const unsigned short* zero;
+  ConvDesc d;
+  __device__ const void* addr(int row, int k8) const {
+    int k = k8 * 8;
+    int co = k % d.K, t = k / d.K;
+    int kw = t % d.KW, kh = t / d.KW;
+    int iw = row % d.W;
+    t = row / d.W;
+    int ih = t % d.H, n = t / d.H;
+    int ohn = ih + d.PT - kh;
+    int own = iw + d.PL - kw;
+    if (kh >= d.KH || ohn < 0 || own < 0 || ohn % d.SY || own % d.SX)
+      return zero;
+    int oh = ohn / d.SY, ow = own / d.SX;
+    if (oh >= d.OH || ow >= d.OW) return zero;
+    return eo + (((long)n * d.OH + oh) * d.OW + ow) * d.K + co;
+  }
+  __device__ unsigned short elem(int row, int k) const {
+    int co = k % d.K, t = k / d.K;
+    int kw = t % d.KW, kh = t / d.KW;
+    int iw = row % d.W;
+    t = row / d.W;
+    int ih = t % d.H, n = t / d.H;
+    int ohn = ih + d.PT - kh;
+    int own = iw + d.PL - kw;
+    if (kh >= d.KH || ohn < 0 || own < 0 || ohn % d.SY || own % d.SX)
+      return 0;
+    int oh = ohn / d.SY, ow = own / d.SX;
+    if (oh >= d.OH || ow >= d.OW) return 0;
+    return eo[(((long)n * d.OH + oh) * d.OW + ow) * d.K + co];
+  }
+};
+
+// stride-1 dgrad specialization: SY == SX == 1 removes the per-address
+// stride divisibility checks and divisions of the generic DgradNT.
+struct DgradNT1 {
+  const unsigned short* eo;
+  const unsigned short* zero;
+  ConvDesc d;
+  __device__ const void* addr(int row, int k8) const {
+    int k = k8 * 8;
+    int co = k % d.K, t = k / d.K;
+    int kw = t % d.KW, kh = t / d.KW;
+    int iw = row % d.W;
+    t = row / d.W;
+    int ih = t % d.H, n = t / d.H;
+    int oh = ih + d.PT - kh;
+    int ow = iw + d.PL - kw;
+    if (kh >= d.KH || (unsigned)oh >= (unsigned)d.OH ||
+        (unsigned)ow >= (unsigned)d.OW)
+      return zero;
+    return eo + (((long)n * d.OH + oh) * d.OW + ow) * d.K + co;
+  }
+  __device__ unsigned short elem(int row, int k) const {
+    int co = k % d.K, t = k / d.K;
+    int kw = t % d.KW, kh = t / d.KW;
+    int iw = row % d.W;
+    t = row / d.W;
+    int ih = t % d.H, n = t / d.H;
+    int oh = ih + d.PT - kh;
+    int ow = iw + d.PL - kw;
+    if (kh >= d.KH || (unsigned)oh >= (unsigned)d.OH ||
+        (unsigned)ow >= (unsigned)d.OW)
+      return 0;
+    return eo[(((long)n * d.OH + oh) * d.OW + ow) * d.K + co];
+  }
+};
+
+static __device__ __forceinline__ int xcd_tile(int b, int ntiles) {
+  // blocks go to XCD (b % 8); remap so tiles of one XCD are contiguous
+  // (per-XCD L2 reuse of the shared B/A panels) — speed-only, any
+  // mapping is correct.
+  if (ntiles < 64) return b;
+  const int per = ntiles >> 3;            // tiles per XCD (floor)
+  const int cut = per << 3;
+  if (b >= cut) return b;                 // remainder keeps identity
+  return (b & 7) * per + (b >> 3);
+}
+

@@ -22,6 +22,7 @@ ext = CUDAExtension(
     sources=[
         "bindings.cpp",
         "hip/gemm.hip",
+        "hip/gemm8.hip",
         "hip/elementwise.hip",
         "hip/softmax.hip",
         "hip/pool_lrn.hip",
