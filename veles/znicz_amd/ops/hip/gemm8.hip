@@ -137,19 +137,19 @@ __global__ __launch_bounds__(O_THREADS, 1) void gemm_nt8_k(
       // all 12 fragment reads BEFORE the barrier: their LDS latency
       // hides behind the waves' barrier skew (the template's phase
       // order); the compiler's own lgkmcnt waits guard the MFMA uses
-      short8 af[4], bf[2][2];
+      short8 af0[4], af1[4], bf[2][2];
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks)
 #pragma unroll
         for (int j = 0; j < 2; ++j)
           bf[ks][j] = rd_frag(Bb, b_row0 + (qn + j) * 16, ks);
 #pragma unroll
-      for (int ks = 0; ks < 2; ++ks)
-#pragma unroll
-        for (int i = 0; i < 2; ++i)
-          af[ks * 2 + i] = rd_frag(Ab, (qm + ks * 2 + i) * 16, 0);
-      // NOTE af[] here holds rows (qm..qm+3) at ks=0; the ks=1 halves
-      // are read inside the MFMA section to bound live registers
+      for (int i = 0; i < 4; ++i) {
+        af0[i] = rd_frag(Ab, (qm + i) * 16, 0);
+        af1[i] = rd_frag(Ab, (qm + i) * 16, 1);
+      }
+      if (q == 3)   // retire reads before the next tile overwrites buf
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
       __builtin_amdgcn_s_barrier();
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
@@ -157,16 +157,13 @@ __global__ __launch_bounds__(O_THREADS, 1) void gemm_nt8_k(
 #pragma unroll
         for (int j = 0; j < 2; ++j)
           acc[qm + i][qn + j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              af[i], bf[0][j], acc[qm + i][qn + j], 0, 0, 0);
-#pragma unroll
-      for (int i = 0; i < 4; ++i)
-        af[i] = rd_frag(Ab, (qm + i) * 16, 1);
+              af0[i], bf[0][j], acc[qm + i][qn + j], 0, 0, 0);
 #pragma unroll
       for (int i = 0; i < 4; ++i)
 #pragma unroll
         for (int j = 0; j < 2; ++j)
           acc[qm + i][qn + j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              af[i], bf[1][j], acc[qm + i][qn + j], 0, 0, 0);
+              af1[i], bf[1][j], acc[qm + i][qn + j], 0, 0, 0);
       __builtin_amdgcn_s_setprio(0);
     }
     buf ^= 1;
