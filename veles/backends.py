@@ -9,9 +9,10 @@ reference/CI path; every op unit has a ``numpy_run``) and
 :class:`ROCmDevice` (the production path: PyTorch-ROCm tensors + the
 hand-written HIP/CDNA4 kernels in ``veles.znicz_amd.ops``).  No OpenCL,
 no CUDA shims, no multi-backend dispatch.  The reference's per-device
-``BLOCK_SIZE`` autotune JSON becomes a static gfx950 tile table in
-``veles/znicz_amd/ops/tiles.py`` (regenerated by ``tools/bench_tiles.py``
-with rocprof evidence).
+``BLOCK_SIZE`` autotune JSON becomes the fixed gfx950 tile constants in
+``veles/znicz_amd/ops/hip/gemm.hip`` (BM/BN/BK/TBK/TNP — chosen with
+``tools/bench_kernels.py`` sweeps and the rocprof/PMC evidence committed
+under ``profiles/``).
 """
 
 import os
