@@ -115,15 +115,29 @@ def main():
     gds_rev = [g for g in reversed(wf.gds)]
     comm_sync = wf.comm_sync
 
-    def step():
-        loader.run()
-        for f in fwds:
-            f.run()
-        ev.run()
-        dec.run()
-        for g in gds_rev:
-            g.run()
-        comm_sync.run()
+    # hipGraph replay of the device step (veles/znicz_amd/hipgraph.py):
+    # default on for single-rank runs; multi-rank needs RCCL's graph
+    # capture, so it is opt-in there (VELES_HIPGRAPH=1).  VELES_HIPGRAPH=0
+    # forces eager.  Capture failure falls back to eager automatically.
+    graph_env = os.environ.get("VELES_HIPGRAPH", "auto")
+    use_graph = graph_env != "0" and (
+        world == 1 or graph_env == "1")
+    if use_graph:
+        from veles.znicz_amd.hipgraph import GraphStepper
+        stepper = GraphStepper(wf)
+
+        def step():
+            stepper.step()
+    else:
+        def step():
+            loader.run()
+            for f in fwds:
+                f.run()
+            ev.run()
+            dec.run()
+            for g in gds_rev:
+                g.run()
+            comm_sync.run()
 
     def barrier_sync():
         if dist_ctx is not None:

@@ -208,3 +208,27 @@ def test_emergency_snapshot(tmp_path):
         "veles.backends", fromlist=["NumpyDevice"]).NumpyDevice())
     wf2.run()
     assert wf2.is_finished
+
+
+def test_graph_stepper_cpu_fallback():
+    """GraphStepper on a CPU device must disable capture and keep the
+    exact eager semantics (incl. gd_skip on valid minibatches)."""
+    import numpy
+    from veles import prng
+    from veles.backends import NumpyDevice
+    from veles.config import root
+    from veles.dummy import DummyLauncher
+    from veles.znicz_amd.hipgraph import GraphStepper
+
+    prng.seed_all(7)
+    from veles.znicz_amd.models.wine import WineWorkflow
+    root.wine.decision.update({"max_epochs": 100})
+    wf = WineWorkflow(DummyLauncher())
+    wf.initialize(device=NumpyDevice())
+    st = GraphStepper(wf)
+    assert st._disabled
+    w0 = wf.forwards[0].weights.mem.copy()
+    for _ in range(6):
+        st.step()
+    w1 = wf.forwards[0].weights.mem
+    assert not numpy.allclose(w0, w1)   # training progressed

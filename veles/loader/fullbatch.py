@@ -52,14 +52,32 @@ class FullBatchLoader(Loader):
             self.original_data.initialize(device)
             self.original_labels.initialize(device)
 
+    def _dev_indices(self, idx, device):
+        """Upload indices through persistent pinned + device buffers —
+        no per-step allocation, async H2D (and the device buffer has a
+        stable address, which a captured hipGraph gather could read)."""
+        import torch
+        size = len(idx)
+        if getattr(self, "_idx_pin", None) is None or \
+                self._idx_pin.shape[0] < self.max_minibatch_size:
+            self._idx_pin = torch.empty(
+                self.max_minibatch_size, dtype=torch.int64,
+                pin_memory=torch.cuda.is_available())
+            self._idx_dev = torch.empty(
+                self.max_minibatch_size, dtype=torch.int64, device=device)
+        self._idx_pin[:size].copy_(
+            torch.from_numpy(idx.astype(numpy.int64)))
+        self._idx_dev[:size].copy_(self._idx_pin[:size],
+                                   non_blocking=True)
+        return self._idx_dev[:size]
+
     def fill_minibatch(self):
         size = self.minibatch_size
         idx = self.minibatch_indices.mem[:size]
         if self.on_device and self.original_data.on_gpu:
             import torch
-            dev_idx = torch.from_numpy(
-                numpy.ascontiguousarray(idx.astype(numpy.int64))).to(
-                self.original_data.devmem.device, non_blocking=True)
+            dev_idx = self._dev_indices(
+                idx, self.original_data.devmem.device)
             # device-side gather straight into the minibatch buffer
             torch.index_select(
                 self.original_data.devmem, 0, dev_idx,
@@ -121,9 +139,8 @@ class FullBatchLoaderMSE(FullBatchLoader):
         idx = self.minibatch_indices.mem[:size]
         if self.on_device and self.original_targets.on_gpu:
             import torch
-            dev_idx = torch.from_numpy(
-                numpy.ascontiguousarray(idx.astype(numpy.int64))).to(
-                self.original_targets.devmem.device, non_blocking=True)
+            dev_idx = self._dev_indices(
+                idx, self.original_targets.devmem.device)
             torch.index_select(
                 self.original_targets.devmem, 0, dev_idx,
                 out=self.minibatch_targets.devmem[:size])

@@ -216,11 +216,17 @@ class Array(object):
                 self._dev_valid = True
         return self
 
+    #: when hipgraph.py is capturing a step, every Array a kernel writes
+    #: registers itself here so replays can re-mark them device-dirty
+    _capture_log = None
+
     def touch_device(self):
         """Mark that a kernel wrote devmem: host becomes stale."""
         if self.on_gpu:
             self._dev_valid = True
             self._host_valid = False
+            if Array._capture_log is not None:
+                Array._capture_log.add(self)
         return self
 
     # -- pickling (snapshot format) -------------------------------------------

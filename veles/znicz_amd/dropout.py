@@ -51,6 +51,14 @@ class DropoutForward(Forward):
             self.mask.reset(numpy.ones(self.input.shape, numpy.uint8))
         self.init_array(self.output)
         self.init_array(self.mask, device_dtype="u8")
+        if device is not None and getattr(device, "is_gpu", False):
+            import torch
+            # device-resident step counter: the mask kernel reads it and
+            # a leading bump kernel advances it, so a hipGraph replay of
+            # the step draws a fresh mask every time (hipgraph.py)
+            self._ctr_dev = torch.full(
+                (1,), int(self.counter), dtype=torch.int64,
+                device=device.torch_device)
 
     def numpy_run(self):
         self.input.map_read()
@@ -79,7 +87,8 @@ class DropoutForward(Forward):
             ops.dropout_fwd(self.input.devmem.view(-1),
                             self.output.devmem.view(-1),
                             self.mask.devmem.view(-1),
-                            self.dropout_ratio, self.seed, self.counter)
+                            self.dropout_ratio, self.seed, self.counter,
+                            getattr(self, "_ctr_dev", None))
         self.output.touch_device()
         self.mask.touch_device()
 

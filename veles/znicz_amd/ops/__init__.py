@@ -228,9 +228,11 @@ def lrn_bwd(err_out, x, out, denom, err_in, alpha, beta, k, n):
                       float(beta), float(k), int(n))
 
 
-def dropout_fwd(x, out, mask, ratio, seed, counter):
+def dropout_fwd(x, out, mask, ratio, seed, counter, ctr_buf=None):
+    """ctr_buf (int64[1] device tensor): device-resident step counter,
+    bumped by a leading kernel — makes the launch hipGraph-replayable."""
     require().dropout_fwd(x, out, mask, float(ratio), int(seed),
-                          int(counter))
+                          int(counter), ctr_buf)
 
 
 def dropout_bwd(err, mask, ratio):

@@ -46,7 +46,7 @@ void confusion_launch(const int*, const int*, int*, int, int,
                       hipStream_t);
 void dropout_fwd_launch(const unsigned short*, unsigned short*,
                         unsigned char*, long, float, unsigned long long,
-                        hipStream_t);
+                        unsigned long long*, hipStream_t);
 void dropout_bwd_launch(unsigned short*, const unsigned char*, long, float,
                         hipStream_t);
 // from softmax.hip
@@ -255,11 +255,19 @@ void lrn_bwd(torch::Tensor ey, torch::Tensor x,
 }
 
 void dropout_fwd(torch::Tensor x, torch::Tensor y, torch::Tensor mask,
-                 double ratio, int64_t seed, int64_t counter) {
-  unsigned long long s =
-      (unsigned long long)seed * 0x100000001B3ull + (unsigned long long)counter;
+                 double ratio, int64_t seed, int64_t counter,
+                 c10::optional<torch::Tensor> ctr_buf) {
+  // with ctr_buf the step counter lives on the device (int64[1]) and the
+  // kernel sequence is hipGraph-replayable; otherwise the host counter
+  // is folded into the seed
+  unsigned long long s = (unsigned long long)seed * 0x100000001B3ull;
+  unsigned long long* ctr = nullptr;
+  if (ctr_buf.has_value())
+    ctr = (unsigned long long*)ctr_buf->data_ptr<int64_t>();
+  else
+    s += (unsigned long long)counter;
   dropout_fwd_launch(ubf(x), ubf_mut(y), mask.data_ptr<unsigned char>(),
-                     x.numel(), (float)ratio, s, cur_stream());
+                     x.numel(), (float)ratio, s, ctr, cur_stream());
 }
 
 void dropout_bwd(torch::Tensor err, torch::Tensor mask, double ratio) {

@@ -455,11 +455,21 @@ void confusion_launch(const int* max_idx, const int* labels, int* cm,
 }
 
 // -- dropout ---------------------------------------------------------------
+// step counter lives in DEVICE memory so the dropout launch sequence is
+// hipGraph-replayable: the bump kernel precedes the mask kernel on the
+// same stream, so every replay draws a fresh mask (a by-value counter
+// would freeze inside a captured graph).
+__global__ void ctr_bump_k(unsigned long long* __restrict__ c) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) ++*c;
+}
+
 __global__ void dropout_fwd_k(const unsigned short* __restrict__ x,
                               unsigned short* __restrict__ y,
                               unsigned char* __restrict__ mask, long n,
                               float ratio, float scale,
-                              unsigned long long seed) {
+                              unsigned long long seed,
+                              const unsigned long long* __restrict__ ctr) {
+  if (ctr) seed += *ctr;
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   long stride = (long)gridDim.x * blockDim.x;
   for (long p = i; p < n; p += stride) {
@@ -480,11 +490,14 @@ __global__ void dropout_bwd_k(unsigned short* __restrict__ err,
 
 void dropout_fwd_launch(const unsigned short* x, unsigned short* y,
                         unsigned char* mask, long n, float ratio,
-                        unsigned long long seed, hipStream_t s) {
+                        unsigned long long seed, unsigned long long* ctr,
+                        hipStream_t s) {
   int blocks = (int)min((n + 255) / 256, (long)4096);
   float scale = 1.f / (1.f - ratio);
+  if (ctr)
+    hipLaunchKernelGGL(ctr_bump_k, dim3(1), dim3(64), 0, s, ctr);
   hipLaunchKernelGGL(dropout_fwd_k, dim3(blocks), dim3(256), 0, s, x, y,
-                     mask, n, ratio, scale, seed);
+                     mask, n, ratio, scale, seed, ctr);
   HIP_CHECK_LAST();
 }
 
