@@ -166,6 +166,13 @@ def main():
         torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
         elapsed = float(t.item())
 
+    if use_graph and rank == 0:
+        print("graph: captured=%s disabled=%s%s" % (
+            stepper._graph is not None, stepper._disabled,
+            "\n" + stepper.capture_error
+            if getattr(stepper, "capture_error", None) else ""),
+            file=sys.stderr)
+
     if args.unit_times and rank == 0:
         wf.print_stats()
 

@@ -137,8 +137,10 @@ class GraphStepper(object):
         except Exception as e:
             # capture is an optimization; any failure (a unit that syncs,
             # an unsupported collective) falls back to eager for good
+            import traceback
             self._disabled = True
             self._graph = None
+            self.capture_error = traceback.format_exc()
             import logging
             logging.getLogger("GraphStepper").warning(
                 "hipGraph capture failed, falling back to eager: %s", e)
