@@ -48,6 +48,9 @@ def test_gemm_tn_matrix(M, N, K):
     (256, 4096, 9216),   # fc6: split-K route
     (96, 200, 128),      # small with bias path
     (256, 1000, 4096),   # fc8
+    (96, 48, 64),        # narrow-N BNT=64 route, tail cols
+    (64, 20, 64),        # narrow-N BNT=32 route, tail cols
+    (300, 64, 256),      # narrow-N BNT=64 exact, M tail
 ])
 def test_gemm_nt_fc_matrix(M, N, K):
     a = RNG.uniform(-1, 1, (M, K)).astype(numpy.float32)
@@ -69,6 +72,9 @@ def test_gemm_nt_fc_matrix(M, N, K):
     (16, 13, 13, 384, 384, 3, 1, 1),
     (8, 30, 30, 64, 200, 3, 1, 1),
     (4, 23, 23, 8, 96, 11, 4, 0),      # conv1-like strided
+    (16, 32, 32, 8, 32, 5, 1, 2),      # cifar conv1: narrow-N fwd
+    (8, 16, 16, 32, 32, 5, 1, 2),      # cifar conv2: narrow fwd+dgrad
+    (8, 8, 8, 32, 64, 5, 1, 2),        # cifar conv3: BNT=64 fwd
 ])
 def test_conv_fwd_dgrad_vs_torch(N, H, W, C, K, KH, SY, PT):
     KW, SX, PL = KH, SY, PT

@@ -174,6 +174,139 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_k(
   }
 }
 
+// ---------------------------------------------------------------------------
+// narrow-N NT core (N ≤ 64: CIFAR-class convs with 32/64 kernels, conv
+// dgrad into 8..64-channel inputs).  The 128-wide tile computes BN=128
+// columns regardless of N, so N=32 wastes 4× of the MFMA work; this
+// variant tiles BNT∈{32,64} with the 4 waves stacked 4M×1N (each wave:
+// 32 rows × BNT cols), keeping the same glds staging + slot swizzle.
+// ---------------------------------------------------------------------------
+
+template <typename LA, typename LB, typename OUT_T, int ACT, int BNT>
+__global__ __launch_bounds__(256, 2) void gemm_ntn_k(
+    LA la, LB lb, OUT_T* __restrict__ C, const float* __restrict__ bias,
+    int M, int N, int K, int tiles_n) {
+  constexpr int LDS_B = BNT * BK * 2;
+  constexpr int LDS_BUF = NT_LDS_A + LDS_B;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int tile = xcd_tile(blockIdx.x, gridDim.x);
+  const int tm = tile / tiles_n, tn = tile % tiles_n;
+  const int m0 = tm * BM, n0 = tn * BNT;
+
+  const int nk = (K + BK - 1) / BK;
+
+  auto stage = [&](int buf, int kt) {
+    char* As = smem + buf * LDS_BUF;
+    char* Bs = As + NT_LDS_A;
+    const int k8base = kt * (BK / 8);
+    const int rl = lane >> 3;
+    const int k8 = lane & 7;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {       // A: 16 chunks × 8 rows
+      const int chunk = wave * 4 + i;
+      const int row = chunk * 8 + rl;
+      const int k8e = k8 ^ rl ^ (chunk & 1);
+      const int kc0 = (k8base + k8e) * 8;
+      const bool straddle = (kc0 < K) & (kc0 + 8 > K);
+      if (!straddle) {
+        const void* sa = la.addr(min(m0 + row, M - 1), k8base + k8e);
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) void*)sa,
+            (__attribute__((address_space(3))) void*)(As + chunk * 1024),
+            16, 0, 0);
+      } else {
+        unsigned short va[8];
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          va[j] = la.elem(min(m0 + row, M - 1), kc0 + j);
+        *(short8*)(As + chunk * 1024 + lane * 16) = *(short8*)va;
+      }
+    }
+#pragma unroll
+    for (int i = 0; i < BNT / 32; ++i) {  // B: BNT/8 chunks × 8 rows
+      const int chunk = wave * (BNT / 32) + i;
+      const int row = chunk * 8 + rl;
+      const int k8e = k8 ^ rl ^ (chunk & 1);
+      const int kc0 = (k8base + k8e) * 8;
+      const bool straddle = (kc0 < K) & (kc0 + 8 > K);
+      if (!straddle) {
+        const void* sb = lb.addr(min(n0 + row, N - 1), k8base + k8e);
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) void*)sb,
+            (__attribute__((address_space(3))) void*)(Bs + chunk * 1024),
+            16, 0, 0);
+      } else {
+        unsigned short vb[8];
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          vb[j] = lb.elem(min(n0 + row, N - 1), kc0 + j);
+        *(short8*)(Bs + chunk * 1024 + lane * 16) = *(short8*)vb;
+      }
+    }
+  };
+
+  f32x4 acc[2][BNT / 16] = {};
+
+  stage(0, 0);
+  __syncthreads();
+  int buf = 0;
+  for (int t = 0; t < nk; ++t) {
+    if (t + 1 < nk) stage(buf ^ 1, t + 1);
+    const char* As = smem + buf * LDS_BUF;
+    const char* Bs = As + NT_LDS_A;
+    const int fr = lane & 15;
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      const int ko = ks * 4 + (lane >> 4);
+      short8 af[2], bf[BNT / 16];
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        const int row = wave * 32 + i * 16 + fr;
+        const int slot = ko ^ (row & 7) ^ ((row >> 3) & 1);
+        af[i] = *(const short8*)(As + row * 128 + slot * 16);
+      }
+#pragma unroll
+      for (int i = 0; i < BNT / 16; ++i) {
+        const int row = i * 16 + fr;
+        const int slot = ko ^ (row & 7) ^ ((row >> 3) & 1);
+        bf[i] = *(const short8*)(Bs + row * 128 + slot * 16);
+      }
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < BNT / 16; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+    }
+    __syncthreads();
+    buf ^= 1;
+  }
+
+  const int colw = n0 + (lane & 15);
+#pragma unroll
+  for (int ni = 0; ni < BNT / 16; ++ni) {
+    const int col = colw + ni * 16;
+    if (col >= N) continue;
+    const float bv = bias ? bias[col] : 0.f;
+#pragma unroll
+    for (int mi = 0; mi < 2; ++mi) {
+      const int rbase = m0 + wave * 32 + mi * 16 + ((lane >> 4) << 2);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = rbase + r;
+        if (row >= M) continue;
+        float v = act_fwd<ACT>(acc[mi][ni][r] + bv);
+        if constexpr (sizeof(OUT_T) == 2)
+          C[(long)row * N + col] = (OUT_T)f2bf(v);
+        else
+          C[(long)row * N + col] = (OUT_T)v;
+      }
+    }
+  }
+}
+
 // split-K finisher: out = act(ws + bias), bf16 or fp32 out
 template <typename OUT_T, int ACT>
 __global__ void bias_act_cast_k(const float* __restrict__ ws,
@@ -534,6 +667,30 @@ static inline int cdiv(int a, int b) { return (a + b - 1) / b; }
 template <typename LA, typename LB, typename OUT_T>
 void launch_gemm_nt(const LA& la, const LB& lb, OUT_T* C, const float* bias,
                     int M, int N, int K, int act, hipStream_t stream) {
+  if (N <= 32) {
+    const int tn = cdiv(N, 32);
+    dim3 grid(cdiv(M, BM) * tn), block(256);
+#define GO32(A)                                                              \
+    hipLaunchKernelGGL((gemm_ntn_k<LA, LB, OUT_T, A, 32>), grid, block,      \
+                       (NT_LDS_A + 32 * BK * 2) * 2, stream, la, lb, C,      \
+                       bias, M, N, K, tn)
+    ACT_SWITCH(act, GO32);
+#undef GO32
+    HIP_CHECK_LAST();
+    return;
+  }
+  if (N <= 64) {
+    const int tn = cdiv(N, 64);
+    dim3 grid(cdiv(M, BM) * tn), block(256);
+#define GO64(A)                                                              \
+    hipLaunchKernelGGL((gemm_ntn_k<LA, LB, OUT_T, A, 64>), grid, block,      \
+                       (NT_LDS_A + 64 * BK * 2) * 2, stream, la, lb, C,      \
+                       bias, M, N, K, tn)
+    ACT_SWITCH(act, GO64);
+#undef GO64
+    HIP_CHECK_LAST();
+    return;
+  }
   const int tn = cdiv(N, BN);
   dim3 grid(cdiv(M, BM) * tn), block(256);
 #define GO(A)                                                                \
