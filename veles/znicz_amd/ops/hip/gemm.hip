@@ -659,6 +659,130 @@ __global__ __launch_bounds__(256, 2) void gemm_tn2_k(
 }
 
 // ---------------------------------------------------------------------------
+// narrow-M TN2 core (M ≤ 64: wgrad of 32/64-kernel conv layers, where
+// M = n_kernels and the 128-row tile wastes 2-4× of the MFMA work).
+// BMT∈{32,64} M-rows per tile, 4 waves stacked 1M×4N (each wave: all
+// BMT rows × 32 n-cols).  The A LDS image keeps the TNP row pitch —
+// tr_b16 requires pitch ≡ 8 dwords (mod 64) — so only compute shrinks,
+// not staging legality.  Split-K atomics as in gemm_tn2_k.
+// ---------------------------------------------------------------------------
+
+template <typename LA, typename LB, int BMT>
+__global__ __launch_bounds__(256, 2) void gemm_tn2n_k(
+    LA la, LB lb, float* __restrict__ C, int M, int N, int K, float beta,
+    int tiles_n) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int tile = xcd_tile(blockIdx.x, gridDim.x);
+  const int tm = tile / tiles_n, tn = tile % tiles_n;
+  const int m0 = tm * BMT, n0 = tn * BN;
+  const int nk = (K + TBK - 1) / TBK;
+  const int sp = gridDim.y, sid = blockIdx.y;
+  const int t_lo = (int)(((long)nk * sid) / sp);
+  const int t_hi = (int)(((long)nk * (sid + 1)) / sp);
+  const int moA = m0 >> 3, moB = n0 >> 3;
+
+  auto stage2 = [&](int buf, int kt) {
+    char* As = smem + buf * TN2_LDS_BUF;
+    char* Bs = As + TN2_LDS_OP;
+    const int k0t = kt * TBK;
+    {
+      // A: BMT/8 octets × TBK k-rows
+      const int oa = tid & (BMT / 8 - 1);
+      const int ka = tid / (BMT / 8);
+      if (ka < TBK) {
+        unsigned short v[8];
+        la.load16(k0t + ka, moA + oa, v);
+        *(short8*)(As + (ka * TNP + oa * 8) * 2) = *(short8*)v;
+      }
+    }
+    const int oct = tid & 15;
+    const int kk0 = tid >> 4;
+#pragma unroll
+    for (int pass = 0; pass < TBK / 16; ++pass) {
+      const int kk = pass * 16 + kk0;
+      unsigned short v[8];
+      lb.load16(k0t + kk, moB + oct, v);
+      *(short8*)(Bs + (kk * TNP + oct * 8) * 2) = *(short8*)v;
+    }
+  };
+
+  f32x4 acc[BMT / 16][2] = {};
+  if (t_lo >= t_hi) return;
+  stage2(0, t_lo);
+  __syncthreads();
+  int buf = 0;
+  for (int t = t_lo; t < t_hi; ++t) {
+    if (t + 1 < t_hi) stage2(buf ^ 1, t + 1);
+    const unsigned short* As =
+        (const unsigned short*)(smem + buf * TN2_LDS_BUF);
+    const unsigned short* Bs = As + TBK * TNP;
+    typedef __attribute__((ext_vector_type(4))) short v4s;
+    typedef __attribute__((address_space(3))) v4s* lds_v4s_ptr;
+    const int gl = lane & 15;
+    const int a4 = gl >> 2;
+    const int c4 = gl & 3;
+    const int ko8 = (lane >> 4) * 8;
+#pragma unroll
+    for (int ks = 0; ks < TBK / 32; ++ks) {
+      const int kb = ks * 32 + ko8;
+      const int rowe = (kb + a4) * TNP + c4 * 4;
+      short8 af[BMT / 16], bf[2];
+#pragma unroll
+      for (int i = 0; i < BMT / 16; ++i) {
+        const unsigned short* ap = As + rowe + i * 16;
+        v4s lo = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+            (lds_v4s_ptr)ap);
+        v4s hi = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+            (lds_v4s_ptr)(ap + 4 * TNP));
+        *(v4s*)&af[i] = lo;
+        *((v4s*)&af[i] + 1) = hi;
+      }
+#pragma unroll
+      for (int j = 0; j < 2; ++j) {
+        const unsigned short* bp = Bs + rowe + (wave * 32 + j * 16);
+        v4s lo = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+            (lds_v4s_ptr)bp);
+        v4s hi = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+            (lds_v4s_ptr)(bp + 4 * TNP));
+        *(v4s*)&bf[j] = lo;
+        *((v4s*)&bf[j] + 1) = hi;
+      }
+#pragma unroll
+      for (int mi = 0; mi < BMT / 16; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+    }
+    __syncthreads();
+    buf ^= 1;
+  }
+
+  const int colw = n0 + wave * 32 + (lane & 15);
+#pragma unroll
+  for (int ni = 0; ni < 2; ++ni) {
+    const int col = colw + ni * 16;
+    if (col >= N) continue;
+#pragma unroll
+    for (int mi = 0; mi < BMT / 16; ++mi) {
+      const int rbase = m0 + mi * 16 + ((lane >> 4) << 2);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = rbase + r;
+        if (row >= M) continue;
+        float* p = C + (long)row * N + col;
+        if (sp > 1)
+          unsafeAtomicAdd(p, acc[mi][ni][r]);
+        else
+          *p = (beta == 0.f ? 0.f : beta * *p) + acc[mi][ni][r];
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // launch helpers (called from bindings.cpp)
 // ---------------------------------------------------------------------------
 
@@ -729,7 +853,8 @@ template <typename LA, typename LB>
 void launch_gemm_tn2(const LA& la, const LB& lb, float* C, int M, int N,
                      int K, float beta, hipStream_t stream) {
   const int tn = cdiv(N, BN);
-  const int tiles = cdiv(M, BM) * tn;
+  const int bmt = M <= 32 ? 32 : (M <= 64 ? 64 : BM);
+  const int tiles = cdiv(M, bmt) * tn;
   const int nk = cdiv(K, TBK);
   const int splits = max(1, min(nk, 1024 / tiles));
   if (splits > 1) {
@@ -744,8 +869,17 @@ void launch_gemm_tn2(const LA& la, const LB& lb, float* C, int M, int N,
     }
   }
   dim3 grid(tiles, splits), block(256);
-  hipLaunchKernelGGL((gemm_tn2_k<LA, LB>), grid, block, TN2_LDS_TOTAL,
-                     stream, la, lb, C, M, N, K, beta, tn);
+  if (bmt == 32)
+    hipLaunchKernelGGL((gemm_tn2n_k<LA, LB, 32>), grid, block,
+                       TN2_LDS_TOTAL, stream, la, lb, C, M, N, K, beta,
+                       tn);
+  else if (bmt == 64)
+    hipLaunchKernelGGL((gemm_tn2n_k<LA, LB, 64>), grid, block,
+                       TN2_LDS_TOTAL, stream, la, lb, C, M, N, K, beta,
+                       tn);
+  else
+    hipLaunchKernelGGL((gemm_tn2_k<LA, LB>), grid, block, TN2_LDS_TOTAL,
+                       stream, la, lb, C, M, N, K, beta, tn);
   HIP_CHECK_LAST();
 }
 
