@@ -126,7 +126,9 @@ def test_graph_stepper_recaptures_on_lr_change():
     assert g1 is not None
     for gd in wf.gds:
         gd.learning_rate *= 0.5
-    st.step()                      # signature mismatch -> re-capture
+    st.step()                      # signature mismatch -> eager rebuild
+    assert st._graph is None
+    st.step()                      # -> re-capture with new params
     assert st._graph is not None and st._graph is not g1
 
 

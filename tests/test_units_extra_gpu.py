@@ -345,3 +345,53 @@ def test_pooling_modes_gpu():
     numpy.testing.assert_allclose(out.reshape(-1), picked, atol=0.02)
     # positive-biased sampling: mean of picks > mean of inputs
     assert picked.mean() > data.mean() + 0.1
+
+
+def test_sgd_multi_matches_sgd_step():
+    """One-launch bucket SGD must match per-tensor sgd_step exactly
+    (same math, same element order)."""
+    import torch
+    from veles.znicz_amd import ops
+
+    class P(object):
+        def __init__(self, n, with_bf16):
+            self.devmem = torch.randn(n, device="cuda",
+                                      dtype=torch.float32)
+            self.bf16 = self.devmem.bfloat16() if with_bf16 else None
+
+    torch.manual_seed(3)
+    sizes = [5000, 32, 70000, 64]
+    params = [P(n, i % 2 == 0) for i, n in enumerate(sizes)]
+    grads = [torch.randn(n, device="cuda") for n in sizes]
+    vels = [torch.randn(n, device="cuda") for n in sizes]
+    hp = [(0.01 * (i + 1), 1e-4, 0.0, 0.9, 1.0 / 256) for i in
+          range(len(sizes))]
+    # reference: per-tensor kernels on copies
+    refs = []
+    for p, g, v, (lr, dec, l1, mom, sc) in zip(params, grads, vels, hp):
+        w2 = p.devmem.clone()
+        v2 = v.clone()
+        b2 = None if p.bf16 is None else p.bf16.clone()
+
+        class Q(object):
+            pass
+        q = Q()
+        q.devmem, q.bf16 = w2, b2
+        ops.sgd_step(q, g, v2, lr, dec, l1, mom, sc)
+        refs.append((w2, v2, b2))
+    # multi: one launch via the raw binding
+    wb = [p.bf16 for p in params]
+    blob, tbl = ops.require().sgd_multi_prepare(
+        [p.devmem for p in params], grads, vels, wb,
+        [h[0] for h in hp], [h[1] for h in hp], [h[2] for h in hp],
+        [h[3] for h in hp], [h[4] for h in hp])
+    ops.sgd_multi(blob.cuda(), tbl.cuda())
+    torch.cuda.synchronize()
+    for p, v, (w2, v2, b2) in zip(params, vels, refs):
+        numpy.testing.assert_allclose(p.devmem.cpu().numpy(),
+                                      w2.cpu().numpy(), rtol=1e-6)
+        numpy.testing.assert_allclose(v.cpu().numpy(), v2.cpu().numpy(),
+                                      rtol=1e-6)
+        if b2 is not None:
+            numpy.testing.assert_allclose(
+                p.bf16.float().cpu().numpy(), b2.float().cpu().numpy())

@@ -80,8 +80,12 @@ class GraphStepper(object):
             return
         self._train_steps += 1
         if self._graph is not None and self._sig != self._signature():
-            self._graph = None            # hyper-params changed: re-capture
-        if self._graph is not None:
+            # hyper-params changed: drop the graph and run ONE eager
+            # step first so downstream caches (e.g. the reducer's SGD
+            # descriptor blobs) rebuild outside of stream capture
+            self._graph = None
+            self._eager()
+        elif self._graph is not None:
             self._graph.replay()
             self._post_replay()
         elif self._train_steps > self.warmup:

@@ -151,6 +151,48 @@ def sgd_step(param_arr, grad, vel, lr, decay, l1_vs_l2, moment, scale):
                        float(moment), float(scale))
 
 
+def sgd_multi_prepare(units, world):
+    """Build the one-launch-per-bucket SGD descriptor blobs for the GD
+    units of a gradient bucket.  Returns (desc_blob_dev, b2t_dev);
+    reusable across steps while hyper-parameters are unchanged (the
+    kernel reads only device memory — hipGraph-replay safe)."""
+    import torch
+    w, g, vel, wb = [], [], [], []
+    lr, decay, l1, moment, scale = [], [], [], [], []
+
+    def push(param_arr, grad, v, lr_, decay_, l1_, moment_, batch):
+        w.append(param_arr.devmem.view(-1))
+        g.append(grad.view(-1))
+        vel.append(v.view(-1))
+        b = getattr(param_arr, "bf16", None)
+        wb.append(None if b is None else b.view(-1))
+        lr.append(float(lr_))
+        decay.append(float(decay_))
+        l1.append(float(l1_))
+        moment.append(float(moment_))
+        scale.append(1.0 / (batch * world))
+
+    for u, batch in units:
+        push(u.weights, u.gradient_weights.devmem,
+             u.accumulated_gradient_weights.devmem, u.learning_rate,
+             u.weights_decay, u.l1_vs_l2, u.gradient_moment, batch)
+        if u.include_bias and u.bias is not None and u.bias:
+            push(u.bias, u.gradient_bias.devmem,
+                 u.accumulated_gradient_bias.devmem, u.learning_rate_bias,
+                 u.weights_decay_bias, u.l1_vs_l2_bias,
+                 u.gradient_moment_bias, batch)
+    blob, tbl = require().sgd_multi_prepare(w, g, vel, wb, lr, decay, l1,
+                                            moment, scale)
+    dev = w[0].device
+    return blob.to(dev), tbl.to(dev)
+
+
+def sgd_multi(desc_blob_dev, b2t_dev):
+    """Apply the fused SGD update to every tensor of a prepared bucket
+    in ONE kernel launch."""
+    require().sgd_multi(desc_blob_dev, b2t_dev)
+
+
 def attach_conv_copies(weights_arr):
     """Conv weights [K,KY,KX,C]: bf16 GEMM image [K, KY·KX·C] and the
     dgrad image [C,KY,KX,K] (= W permuted, no flip — the dgrad loader

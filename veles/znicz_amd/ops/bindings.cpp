@@ -185,6 +185,60 @@ void sgd_step(torch::Tensor w, torch::Tensor g, torch::Tensor vel,
   }
 }
 
+// -- multi-tensor SGD (one launch per gradient bucket) ----------------------
+#include "hip/sgd_multi.h"
+void sgd_multi_launch(const void*, const int*, int, hipStream_t);
+
+// Build the descriptor blob + block->tensor table on the host (CPU
+// tensors; the caller uploads them once and reuses across steps —
+// hipGraph replays then read only device memory).
+std::vector<torch::Tensor> sgd_multi_prepare(
+    std::vector<torch::Tensor> w, std::vector<torch::Tensor> g,
+    std::vector<torch::Tensor> vel,
+    std::vector<c10::optional<torch::Tensor>> wb,
+    std::vector<double> lr, std::vector<double> decay,
+    std::vector<double> l1, std::vector<double> moment,
+    std::vector<double> scale) {
+  const size_t nt = w.size();
+  TORCH_CHECK(g.size() == nt && vel.size() == nt && wb.size() == nt &&
+              lr.size() == nt, "sgd_multi_prepare: length mismatch");
+  std::vector<SgdDesc> descs(nt);
+  std::vector<int> b2t;
+  long long blk = 0;
+  for (size_t i = 0; i < nt; ++i) {
+    SgdDesc& d = descs[i];
+    d.w = w[i].data_ptr<float>();
+    d.g = g[i].data_ptr<float>();
+    d.vel = vel[i].data_ptr<float>();
+    d.wb = wb[i].has_value() ? ubf_mut(*wb[i]) : nullptr;
+    d.n = w[i].numel();
+    d.blk0 = blk;
+    d.lr = (float)lr[i];
+    d.decay = (float)decay[i];
+    d.l1 = (float)l1[i];
+    d.moment = (float)moment[i];
+    d.scale = (float)scale[i];
+    const long long nb = (d.n + SGD_MULTI_SPAN - 1) / SGD_MULTI_SPAN;
+    for (long long b = 0; b < nb; ++b) b2t.push_back((int)i);
+    blk += nb;
+  }
+  auto blob = torch::empty({(long)(nt * sizeof(SgdDesc))},
+                           torch::dtype(torch::kUInt8));
+  memcpy(blob.data_ptr<uint8_t>(), descs.data(), nt * sizeof(SgdDesc));
+  auto tbl = torch::empty({(long)b2t.size()},
+                          torch::dtype(torch::kInt32));
+  memcpy(tbl.data_ptr<int>(), b2t.data(), b2t.size() * sizeof(int));
+  return {blob, tbl};
+}
+
+void sgd_multi(torch::Tensor desc_blob_dev, torch::Tensor b2t_dev) {
+  TORCH_CHECK(desc_blob_dev.is_cuda() && b2t_dev.is_cuda(),
+              "sgd_multi: device tensors required");
+  sgd_multi_launch(desc_blob_dev.data_ptr<uint8_t>(),
+                   b2t_dev.data_ptr<int>(), (int)b2t_dev.numel(),
+                   cur_stream());
+}
+
 void col_sum(torch::Tensor a, torch::Tensor out, bool accumulate) {
   col_sum_launch(ubf(a), out.data_ptr<float>(), a.size(0), a.size(1),
                  accumulate, cur_stream());
@@ -335,6 +389,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("act_fwd", &act_fwd);
   m.def("act_bwd", &act_bwd);
   m.def("sgd_step", &sgd_step);
+  m.def("sgd_multi_prepare", &sgd_multi_prepare);
+  m.def("sgd_multi", &sgd_multi);
   m.def("col_sum", &col_sum);
   m.def("transpose2d", &transpose2d);
   m.def("softmax_fwd", &softmax_fwd);

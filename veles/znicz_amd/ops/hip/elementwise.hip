@@ -72,23 +72,52 @@ void act_bwd_launch(unsigned short* err, const unsigned short* y, long n,
 // -- fused SGD: master fp32 update + bf16 working-copy refresh --------------
 //  g = grad*scale + decay*(l1*sign(w) + (1-l1)*w)
 //  v = moment*v + lr*g ;  w -= v ;  bf16[i] = w
+#include "sgd_multi.h"
+
+__device__ __forceinline__ void sgd_elem(
+    float* __restrict__ w, const float* __restrict__ g,
+    float* __restrict__ vel, unsigned short* __restrict__ wb, long p,
+    float lr, float decay, float l1, float moment, float scale) {
+  float wi = w[p];
+  float l1l2 = l1 * (wi > 0.f ? 1.f : (wi < 0.f ? -1.f : 0.f)) +
+               (1.f - l1) * wi;
+  float gi = g[p] * scale + decay * l1l2;
+  float v = moment * vel[p] + lr * gi;
+  if (moment != 0.f) vel[p] = v;
+  wi -= v;
+  w[p] = wi;
+  if (wb) wb[p] = f2bf(wi);
+}
+
 __global__ void sgd_k(float* __restrict__ w, const float* __restrict__ g,
                       float* __restrict__ vel,
                       unsigned short* __restrict__ wb, long n, float lr,
                       float decay, float l1, float moment, float scale) {
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   long stride = (long)gridDim.x * blockDim.x;
-  for (long p = i; p < n; p += stride) {
-    float wi = w[p];
-    float l1l2 = l1 * (wi > 0.f ? 1.f : (wi < 0.f ? -1.f : 0.f)) +
-                 (1.f - l1) * wi;
-    float gi = g[p] * scale + decay * l1l2;
-    float v = moment * vel[p] + lr * gi;
-    if (moment != 0.f) vel[p] = v;
-    wi -= v;
-    w[p] = wi;
-    if (wb) wb[p] = f2bf(wi);
+  for (long p = i; p < n; p += stride)
+    sgd_elem(w, g, vel, wb, p, lr, decay, l1, moment, scale);
+}
+
+// one launch for a whole gradient bucket: block -> tensor via b2t
+__global__ void sgd_multi_k(const SgdDesc* __restrict__ ds,
+                            const int* __restrict__ b2t) {
+  const SgdDesc d = ds[b2t[blockIdx.x]];
+  const long base = ((long)blockIdx.x - d.blk0) * SGD_MULTI_SPAN;
+#pragma unroll
+  for (int i = 0; i < SGD_MULTI_SPAN / 256; ++i) {
+    const long p = base + i * 256 + threadIdx.x;
+    if (p < d.n)
+      sgd_elem(d.w, d.g, d.vel, d.wb, p, d.lr, d.decay, d.l1, d.moment,
+               d.scale);
   }
+}
+
+void sgd_multi_launch(const void* descs, const int* b2t, int nblocks,
+                      hipStream_t s) {
+  hipLaunchKernelGGL(sgd_multi_k, dim3(nblocks), dim3(256), 0, s,
+                     (const SgdDesc*)descs, b2t);
+  HIP_CHECK_LAST();
 }
 
 void sgd_launch(float* w, const float* g, float* vel, unsigned short* wb,

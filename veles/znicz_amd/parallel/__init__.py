@@ -142,6 +142,9 @@ class GradientReducer(object):
         self._bucket_fill = 0
         self._tail_event = torch.cuda.Event()
         self._have_tail = False
+        # multi-tensor SGD descriptor cache: bucket composition ->
+        # (hyper-param signature, desc_blob_dev, b2t_dev)
+        self._sgd_cache = {}
 
     @property
     def world(self):
@@ -175,19 +178,25 @@ class GradientReducer(object):
                             u.gradient_bias.devmem is not None:
                         tensors.append(u.gradient_bias.devmem.view(-1))
                 coalesced_all_reduce(self.ctx, tensors)
+            # ONE fused launch for the whole bucket's SGD updates (the
+            # per-layer sgd_k launches were a ~5 µs/launch latency tail)
+            key = tuple(id(u) for u, _ in self._pending_apply)
+            sig = tuple(
+                (float(u.learning_rate), float(u.weights_decay),
+                 float(u.l1_vs_l2), float(u.gradient_moment),
+                 float(u.learning_rate_bias), float(u.weights_decay_bias),
+                 float(u.l1_vs_l2_bias), float(u.gradient_moment_bias),
+                 int(batch))
+                for u, batch in self._pending_apply)
+            cached = self._sgd_cache.get(key)
+            if cached is None or cached[0] != sig:
+                blob, tbl = ops.sgd_multi_prepare(self._pending_apply,
+                                                  self.world)
+                self._sgd_cache[key] = cached = (sig, blob, tbl)
+            ops.sgd_multi(cached[1], cached[2])
             for u, batch in self._pending_apply:
-                scale = 1.0 / (batch * self.world)
-                ops.sgd_step(u.weights, u.gradient_weights.devmem,
-                             u.accumulated_gradient_weights.devmem,
-                             u.learning_rate, u.weights_decay, u.l1_vs_l2,
-                             u.gradient_moment, scale)
                 u.weights.touch_device()
                 if u.include_bias and u.bias is not None and u.bias:
-                    ops.sgd_step(u.bias, u.gradient_bias.devmem,
-                                 u.accumulated_gradient_bias.devmem,
-                                 u.learning_rate_bias,
-                                 u.weights_decay_bias, u.l1_vs_l2_bias,
-                                 u.gradient_moment_bias, scale)
                     u.bias.touch_device()
                 u.post_update()
             self._tail_event.record(self.comm_stream)
