@@ -27,6 +27,8 @@ void act_bwd_launch(unsigned short*, const unsigned short*, long, int,
                     hipStream_t);
 void sgd_launch(float*, const float*, float*, unsigned short*, long, float,
                 float, float, float, float, hipStream_t);
+#include "hip/sgd_multi.h"
+void sgd_multi_launch(const void*, const int*, int, hipStream_t);
 void col_sum_launch(const unsigned short*, float*, int, int, bool,
                     hipStream_t);
 void transpose2d_launch(const unsigned short*, unsigned short*, int, int,
@@ -186,9 +188,6 @@ void sgd_step(torch::Tensor w, torch::Tensor g, torch::Tensor vel,
 }
 
 // -- multi-tensor SGD (one launch per gradient bucket) ----------------------
-#include "hip/sgd_multi.h"
-void sgd_multi_launch(const void*, const int*, int, hipStream_t);
-
 // Build the descriptor blob + block->tensor table on the host (CPU
 // tensors; the caller uploads them once and reuses across steps —
 // hipGraph replays then read only device memory).
