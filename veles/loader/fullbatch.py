@@ -55,9 +55,17 @@ class FullBatchLoader(Loader):
     def _dev_indices(self, idx, device):
         """Upload indices through persistent pinned + device buffers —
         no per-step allocation, async H2D (and the device buffer has a
-        stable address, which a captured hipGraph gather could read)."""
+        stable address, which a captured hipGraph gather could read).
+
+        The event guard closes a host/device race: without it the host
+        could overwrite the pinned staging buffer on step t+1 while the
+        deep-queued async H2D of step t is still reading it.  One upload
+        per minibatch — the MSE subclass's second call reuses it."""
         import torch
         size = len(idx)
+        key = (self.epoch_number, self.minibatch_offset, size)
+        if getattr(self, "_idx_key", None) == key:
+            return self._idx_dev[:size]
         if getattr(self, "_idx_pin", None) is None or \
                 self._idx_pin.shape[0] < self.max_minibatch_size:
             self._idx_pin = torch.empty(
@@ -65,10 +73,17 @@ class FullBatchLoader(Loader):
                 pin_memory=torch.cuda.is_available())
             self._idx_dev = torch.empty(
                 self.max_minibatch_size, dtype=torch.int64, device=device)
+            self._idx_evt = torch.cuda.Event()
+            self._idx_evt_live = False
+        if self._idx_evt_live:
+            self._idx_evt.synchronize()   # previous upload fully read
         self._idx_pin[:size].copy_(
             torch.from_numpy(idx.astype(numpy.int64)))
         self._idx_dev[:size].copy_(self._idx_pin[:size],
                                    non_blocking=True)
+        self._idx_evt.record()
+        self._idx_evt_live = True
+        self._idx_key = key
         return self._idx_dev[:size]
 
     def fill_minibatch(self):
