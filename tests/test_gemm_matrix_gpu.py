@@ -108,7 +108,9 @@ def test_conv_fwd_dgrad_vs_torch(N, H, W, C, K, KH, SY, PT):
 
     gw = torch.zeros(K, KH * KW * C, device="cuda",
                      dtype=torch.float32)
-    ops.conv_wgrad(eo.view(-1, K), x.view(N, -1), gw, desc, 0.0)
+    gb = torch.full((K,), 7.0, device="cuda", dtype=torch.float32)
+    fused = ops.conv_wgrad(eo.view(-1, K), x.view(N, -1), gw, desc, 0.0,
+                           bias_out=gb)
     ref_gw = torch.nn.grad.conv2d_weight(
         xf, (K, C, KH, KW), eo.float().permute(0, 3, 1, 2), stride=SY,
         padding=PT)
@@ -116,3 +118,10 @@ def test_conv_fwd_dgrad_vs_torch(N, H, W, C, K, KH, SY, PT):
         gw.view(K, KH, KW, C).permute(0, 3, 1, 2).cpu().numpy(),
         ref_gw.cpu().numpy(), rtol=3e-2,
         atol=0.3 + 0.001 * N * OH * OW)
+    # fused bias gradient = column sums of err_output (beta=0 path)
+    assert fused == ((K % 8 == 0) and (C % 8 == 0))
+    if fused:
+        ref_gb = eo.float().view(-1, K).sum(0)
+        numpy.testing.assert_allclose(
+            gb.cpu().numpy(), ref_gb.cpu().numpy(), rtol=2e-2,
+            atol=0.1 + 0.0005 * N * OH * OW)

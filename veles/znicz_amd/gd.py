@@ -95,14 +95,22 @@ class GradientDescent(GradientDescentBase):
             self.err_input.touch_device()
         x = ops.as2d_bf16(self.input)
         beta = 1.0 if self.accumulate_gradient else 0.0
+        want_bias = self.include_bias and self.bias is not None and \
+            self.bias
+        fused = False
         if self.weights_transposed:   # gw[in, out] = xᵀ · eo
             ops.gemm_tn(x, eo, self.gradient_weights.devmem, beta=beta)
         else:                         # gw[out, in] = eoᵀ · x
-            ops.gemm_tn(eo, x, self.gradient_weights.devmem, beta=beta)
+            # A = eo, so the TN2 route co-produces the bias gradient
+            fused = ops.gemm_tn(
+                eo, x, self.gradient_weights.devmem, beta=beta,
+                bias_out=self.gradient_bias.devmem if want_bias
+                else None)
         self.gradient_weights.touch_device()
-        if self.include_bias and self.bias is not None and self.bias:
-            ops.col_sum(eo, self.gradient_bias.devmem,
-                        accumulate=self.accumulate_gradient)
+        if want_bias:
+            if not fused:
+                ops.col_sum(eo, self.gradient_bias.devmem,
+                            accumulate=self.accumulate_gradient)
             self.gradient_bias.touch_device()
         self._hip_apply(batch)
 

@@ -119,14 +119,19 @@ class GradientDescentConv(GradientDescentBase):
             ops.conv_dgrad(eo, self.weights.bf16_dgrad,
                            self.err_input.devmem, desc)
             self.err_input.touch_device()
-        ops.conv_wgrad(eo, self.input.devmem,
-                       self.gradient_weights.devmem, desc,
-                       beta=1.0 if self.accumulate_gradient else 0.0)
+        want_bias = self.include_bias and self.bias is not None and \
+            self.bias
+        bias_out = self.gradient_bias.devmem if want_bias else None
+        fused = ops.conv_wgrad(eo, self.input.devmem,
+                               self.gradient_weights.devmem, desc,
+                               beta=1.0 if self.accumulate_gradient
+                               else 0.0, bias_out=bias_out)
         self.gradient_weights.touch_device()
-        if self.include_bias and self.bias is not None and self.bias:
-            K = fwd.n_kernels
-            ops.col_sum(eo.view(-1, K), self.gradient_bias.devmem,
-                        accumulate=self.accumulate_gradient)
+        if want_bias:
+            if not fused:   # non-vectorized TN route: separate col_sum
+                K = fwd.n_kernels
+                ops.col_sum(eo.view(-1, K), self.gradient_bias.devmem,
+                            accumulate=self.accumulate_gradient)
             self.gradient_bias.touch_device()
         self._hip_apply(batch)
 

@@ -116,9 +116,14 @@ def gemm_nt(a, b, out, bias=None, activation=0):
     require().gemm_nt(a, b, out, bias, int(activation))
 
 
-def gemm_tn(a, b, out, beta=0.0):
-    """out[M,N] (fp32) = a[K,M]ᵀ · b[K,N] (+beta*out).  a,b bf16."""
-    require().gemm_tn(a, b, out, float(beta))
+def gemm_tn(a, b, out, beta=0.0, bias_out=None):
+    """out[M,N] (fp32) = a[K,M]ᵀ · b[K,N] (+beta*out).  a,b bf16.
+
+    With ``bias_out`` (fp32[M]) the kernel co-produces the column sums
+    of ``a`` — the bias gradient when a is err_output — for free during
+    operand staging.  Returns True iff it did (the vectorized TN2 route
+    only); on False the caller must col_sum separately."""
+    return require().gemm_tn(a, b, out, float(beta), bias_out)
 
 
 def act_bwd_inplace(err, output, activation):
@@ -248,8 +253,11 @@ def conv_dgrad(err_out, w_dgrad, err_in, conv_desc):
     require().conv_dgrad(err_out, w_dgrad, err_in, conv_desc)
 
 
-def conv_wgrad(err_out, x, grad_w, conv_desc, beta=0.0):
-    require().conv_wgrad(err_out, x, grad_w, conv_desc, float(beta))
+def conv_wgrad(err_out, x, grad_w, conv_desc, beta=0.0, bias_out=None):
+    """Returns True iff bias_out (fp32[K]) was co-produced (see
+    gemm_tn)."""
+    return require().conv_wgrad(err_out, x, grad_w, conv_desc,
+                                float(beta), bias_out)
 
 
 def pool_fwd(x, out, offs, desc, mode):

@@ -11,16 +11,16 @@
 void gemm_nt_dense(const unsigned short*, const unsigned short*, void*,
                    bool, const float*, const unsigned short*, int, int,
                    int, int, hipStream_t);
-void gemm_tn_dense(const unsigned short*, const unsigned short*, float*,
-                   int, int, int, float, hipStream_t);
+bool gemm_tn_dense(const unsigned short*, const unsigned short*, float*,
+                   int, int, int, float, hipStream_t, float*);
 void conv_fwd_gemm(const unsigned short*, const unsigned short*, void*,
                    bool, const float*, const unsigned short*,
                    const ConvDesc&, int, hipStream_t);
 void conv_dgrad_gemm(const unsigned short*, const unsigned short*,
                      unsigned short*, const unsigned short*,
                      const ConvDesc&, hipStream_t);
-void conv_wgrad_gemm(const unsigned short*, const unsigned short*, float*,
-                     const ConvDesc&, float, hipStream_t);
+bool conv_wgrad_gemm(const unsigned short*, const unsigned short*, float*,
+                     const ConvDesc&, float, hipStream_t, float*);
 // from elementwise.hip
 void act_fwd_launch(unsigned short*, long, int, hipStream_t);
 void act_bwd_launch(unsigned short*, const unsigned short*, long, int,
@@ -122,15 +122,19 @@ void gemm_nt(torch::Tensor a, torch::Tensor b, torch::Tensor out,
                 N, K, (int)act, cur_stream());
 }
 
-void gemm_tn(torch::Tensor a, torch::Tensor b, torch::Tensor out,
-             double beta) {
+bool gemm_tn(torch::Tensor a, torch::Tensor b, torch::Tensor out,
+             double beta, c10::optional<torch::Tensor> bias_out) {
   // out[M,N] = aᵀ·b with a[K,M], b[K,N]
   const int K = a.size(0), M = a.size(1), N = b.size(1);
   TORCH_CHECK(b.size(0) == K, "gemm_tn: K mismatch");
   TORCH_CHECK(out.scalar_type() == torch::kFloat32, "gemm_tn: fp32 out");
   TORCH_CHECK(out.numel() == (int64_t)M * N, "gemm_tn: out shape");
-  gemm_tn_dense(ubf(a), ubf(b), out.data_ptr<float>(), M, N, K,
-                (float)beta, cur_stream());
+  float* bp = bias_out.has_value() ? bias_out->data_ptr<float>()
+                                   : nullptr;
+  // true: the kernel also produced bias_out = column sums of a (the
+  // bias gradient); false: caller must col_sum separately
+  return gemm_tn_dense(ubf(a), ubf(b), out.data_ptr<float>(), M, N, K,
+                       (float)beta, cur_stream(), bp);
 }
 
 void conv_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor out,
@@ -153,11 +157,14 @@ void conv_dgrad(torch::Tensor eo, torch::Tensor wdg, torch::Tensor ei,
                   d, cur_stream());
 }
 
-void conv_wgrad(torch::Tensor eo, torch::Tensor x, torch::Tensor gw,
-                std::vector<int64_t> desc, double beta) {
+bool conv_wgrad(torch::Tensor eo, torch::Tensor x, torch::Tensor gw,
+                std::vector<int64_t> desc, double beta,
+                c10::optional<torch::Tensor> bias_out) {
   ConvDesc d = make_conv_desc(desc);
-  conv_wgrad_gemm(ubf(eo), ubf(x), gw.data_ptr<float>(), d, (float)beta,
-                  cur_stream());
+  float* bp = bias_out.has_value() ? bias_out->data_ptr<float>()
+                                   : nullptr;
+  return conv_wgrad_gemm(ubf(eo), ubf(x), gw.data_ptr<float>(), d,
+                         (float)beta, cur_stream(), bp);
 }
 
 void act_fwd(torch::Tensor x, int64_t act) {

@@ -143,6 +143,22 @@ void zero_f32_launch(float* p, long n, hipStream_t s) {
   HIP_CHECK_LAST();
 }
 
+// zero two disjoint buffers in one launch (split-K C workspace + the
+// fused bias-gradient accumulator)
+__global__ void zero2_f32_k(float* __restrict__ p, long n,
+                            float* __restrict__ q, long m) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long t = i; t < n; t += stride) p[t] = 0.f;
+  for (long t = i; t < m; t += stride) q[t] = 0.f;
+}
+
+void zero2_f32_launch(float* p, long n, float* q, long m, hipStream_t s) {
+  int b = (int)min((n + 255) / 256, (long)2048);
+  hipLaunchKernelGGL(zero2_f32_k, dim3(b), dim3(256), 0, s, p, n, q, m);
+  HIP_CHECK_LAST();
+}
+
 // -- column sum (bias gradient): out[N] = Σ_m A[M][N], bf16 -> fp32 --------
 // Thread owns 8 consecutive cols (one b128 load per row — wave reads
 // 1 KiB contiguous); row-slab blocks reduce through LDS, then one fp32

@@ -550,7 +550,7 @@ struct Im2colTNv {
 template <typename LA, typename LB>
 __global__ __launch_bounds__(256, 2) void gemm_tn2_k(
     LA la, LB lb, float* __restrict__ C, int M, int N, int K, float beta,
-    int tiles_n) {
+    int tiles_n, float* __restrict__ bias_out) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const int tid = threadIdx.x;
   const int wave = tid >> 6, lane = tid & 63;
@@ -564,6 +564,10 @@ __global__ __launch_bounds__(256, 2) void gemm_tn2_k(
   const int t_hi = (int)(((long)nk * (sid + 1)) / sp);
   const int moA = m0 >> 3, moB = n0 >> 3;
 
+  // fused bias gradient (see gemm_tn2n_k): column sums of the staged A
+  float bsum[8] = {};
+  const bool do_bias = bias_out != nullptr && tn == 0;
+
   auto stage2 = [&](int buf, int kt) {
     char* As = smem + buf * TN2_LDS_BUF;
     char* Bs = As + TN2_LDS_OP;
@@ -576,6 +580,9 @@ __global__ __launch_bounds__(256, 2) void gemm_tn2_k(
       unsigned short v[8];
       la.load16(k0t + kk, moA + oct, v);
       *(short8*)(As + (kk * TNP + oct * 8) * 2) = *(short8*)v;
+      if (do_bias)
+#pragma unroll
+        for (int j = 0; j < 8; ++j) bsum[j] += bf2f(v[j]);
       lb.load16(k0t + kk, moB + oct, v);
       *(short8*)(Bs + (kk * TNP + oct * 8) * 2) = *(short8*)v;
     }
@@ -636,6 +643,26 @@ __global__ __launch_bounds__(256, 2) void gemm_tn2_k(
     buf ^= 1;
   }
 
+  if (do_bias) {
+    float* red = (float*)smem;
+    const int oct = tid & 15;
+    const int kk0 = tid >> 4;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      red[kk0 * BM + oct * 8 + j] = bsum[j];
+    __syncthreads();
+    if (tid < BM && m0 + tid < M) {
+      float s = 0.f;
+#pragma unroll 4
+      for (int r = 0; r < 16; ++r) s += red[r * BM + tid];
+      if (sp > 1)
+        unsafeAtomicAdd(bias_out + m0 + tid, s);
+      else
+        bias_out[m0 + tid] =
+            (beta == 0.f ? 0.f : beta * bias_out[m0 + tid]) + s;
+    }
+  }
+
   const int colw = n0 + wc * 64 + (lane & 15);
 #pragma unroll
   for (int ni = 0; ni < 4; ++ni) {
@@ -670,7 +697,7 @@ __global__ __launch_bounds__(256, 2) void gemm_tn2_k(
 template <typename LA, typename LB, int BMT>
 __global__ __launch_bounds__(256, 2) void gemm_tn2n_k(
     LA la, LB lb, float* __restrict__ C, int M, int N, int K, float beta,
-    int tiles_n) {
+    int tiles_n, float* __restrict__ bias_out) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const int tid = threadIdx.x;
   const int wave = tid >> 6, lane = tid & 63;
@@ -682,6 +709,13 @@ __global__ __launch_bounds__(256, 2) void gemm_tn2n_k(
   const int t_lo = (int)(((long)nk * sid) / sp);
   const int t_hi = (int)(((long)nk * (sid + 1)) / sp);
   const int moA = m0 >> 3, moB = n0 >> 3;
+
+  // fused bias gradient: A IS the (activation-folded) err_output, so
+  // each thread co-accumulates the column sums of the A elements it
+  // stages — replaces the separate col_sum kernel pair for this layer.
+  // Only tn==0 blocks contribute (A is identical across n-tiles).
+  float bsum[8] = {};
+  const bool do_bias = bias_out != nullptr && tn == 0;
 
   auto stage2 = [&](int buf, int kt) {
     char* As = smem + buf * TN2_LDS_BUF;
@@ -695,6 +729,9 @@ __global__ __launch_bounds__(256, 2) void gemm_tn2n_k(
         unsigned short v[8];
         la.load16(k0t + ka, moA + oa, v);
         *(short8*)(As + (ka * TNP + oa * 8) * 2) = *(short8*)v;
+        if (do_bias)
+#pragma unroll
+          for (int j = 0; j < 8; ++j) bsum[j] += bf2f(v[j]);
       }
     }
     const int oct = tid & 15;
@@ -758,6 +795,30 @@ __global__ __launch_bounds__(256, 2) void gemm_tn2n_k(
     }
     __syncthreads();
     buf ^= 1;
+  }
+
+  if (do_bias) {
+    // cross-thread reduce of the staged-A column sums (LDS is free
+    // after the loop's trailing barrier), one atomic per m-column
+    float* red = (float*)smem;
+    const int oa = tid & (BMT / 8 - 1);
+    const int ka = tid / (BMT / 8);
+    if (ka < TBK) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        red[ka * BMT + oa * 8 + j] = bsum[j];
+    }
+    __syncthreads();
+    if (tid < BMT && m0 + tid < M) {
+      float s = 0.f;
+#pragma unroll 4
+      for (int r = 0; r < TBK; ++r) s += red[r * BMT + tid];
+      if (sp > 1)
+        unsafeAtomicAdd(bias_out + m0 + tid, s);
+      else
+        bias_out[m0 + tid] =
+            (beta == 0.f ? 0.f : beta * bias_out[m0 + tid]) + s;
+    }
   }
 
   const int colw = n0 + wave * 32 + (lane & 15);
@@ -851,7 +912,8 @@ void launch_gemm_tn(const LA& la, const LB& lb, float* C, int M, int N,
 
 template <typename LA, typename LB>
 void launch_gemm_tn2(const LA& la, const LB& lb, float* C, int M, int N,
-                     int K, float beta, hipStream_t stream) {
+                     int K, float beta, hipStream_t stream,
+                     float* bias_out = nullptr) {
   const int tn = cdiv(N, BN);
   const int bmt = M <= 32 ? 32 : (M <= 64 ? 64 : BM);
   const int tiles = cdiv(M, bmt) * tn;
@@ -860,26 +922,33 @@ void launch_gemm_tn2(const LA& la, const LB& lb, float* C, int M, int N,
   if (splits > 1) {
     if (beta == 0.f) {
       void zero_f32_launch(float*, long, hipStream_t);
-      zero_f32_launch(C, (long)M * N, stream);
+      void zero2_f32_launch(float*, long, float*, long, hipStream_t);
+      if (bias_out)
+        zero2_f32_launch(C, (long)M * N, bias_out, M, stream);
+      else
+        zero_f32_launch(C, (long)M * N, stream);
     } else if (beta != 1.f) {
       long n = (long)M * N;
       int b = (int)min((n + 255) / 256, (long)2048);
       hipLaunchKernelGGL(scale_f32_k, dim3(b), dim3(256), 0, stream, C, n,
                          beta);
+      if (bias_out)
+        hipLaunchKernelGGL(scale_f32_k, dim3(1), dim3(256), 0, stream,
+                           bias_out, (long)M, beta);
     }
   }
   dim3 grid(tiles, splits), block(256);
   if (bmt == 32)
     hipLaunchKernelGGL((gemm_tn2n_k<LA, LB, 32>), grid, block,
                        TN2_LDS_TOTAL, stream, la, lb, C, M, N, K, beta,
-                       tn);
+                       tn, bias_out);
   else if (bmt == 64)
     hipLaunchKernelGGL((gemm_tn2n_k<LA, LB, 64>), grid, block,
                        TN2_LDS_TOTAL, stream, la, lb, C, M, N, K, beta,
-                       tn);
+                       tn, bias_out);
   else
     hipLaunchKernelGGL((gemm_tn2_k<LA, LB>), grid, block, TN2_LDS_TOTAL,
-                       stream, la, lb, C, M, N, K, beta, tn);
+                       stream, la, lb, C, M, N, K, beta, tn, bias_out);
   HIP_CHECK_LAST();
 }
 
@@ -935,16 +1004,17 @@ void gemm_nt_dense(const unsigned short* A, const unsigned short* B,
     launch_gemm_nt(la, lb, (unsigned short*)C, bias, M, N, K, act, s);
 }
 
-void gemm_tn_dense(const unsigned short* A, const unsigned short* B,
+bool gemm_tn_dense(const unsigned short* A, const unsigned short* B,
                    float* C, int M, int N, int K, float beta,
-                   hipStream_t s) {
+                   hipStream_t s, float* bias_out) {
   if ((M & 7) == 0 && (N & 7) == 0) {
     DenseTNv la{A, M, K, M}, lb{B, N, K, N};
-    launch_gemm_tn2(la, lb, C, M, N, K, beta, s);
-    return;
+    launch_gemm_tn2(la, lb, C, M, N, K, beta, s, bias_out);
+    return bias_out != nullptr;
   }
   DenseTN la{A, M, K, M}, lb{B, N, K, N};
   launch_gemm_tn(la, lb, C, M, N, K, beta, s);
+  return false;   // caller computes the bias gradient separately
 }
 
 void conv_fwd_gemm8(const unsigned short*, const unsigned short*, void*,
@@ -999,19 +1069,20 @@ void conv_dgrad_gemm(const unsigned short* eo, const unsigned short* wdg,
   launch_gemm_nt(la, lb, ei, (const float*)nullptr, M, N, K, ACT_LINEAR, s);
 }
 
-void conv_wgrad_gemm(const unsigned short* eo, const unsigned short* x,
+bool conv_wgrad_gemm(const unsigned short* eo, const unsigned short* x,
                      float* gw, const ConvDesc& d, float beta,
-                     hipStream_t s) {
+                     hipStream_t s, float* bias_out) {
   const int Kdim = d.N * d.OH * d.OW;
   const int M = d.K;                   // n_kernels
   const int N = d.KH * d.KW * d.C;
   if ((M & 7) == 0 && (d.C & 7) == 0) {
     DenseTNv la{eo, d.K, Kdim, d.K};
     Im2colTNv lb{x, d, Kdim, N};
-    launch_gemm_tn2(la, lb, gw, M, N, Kdim, beta, s);
-    return;
+    launch_gemm_tn2(la, lb, gw, M, N, Kdim, beta, s, bias_out);
+    return bias_out != nullptr;
   }
   DenseTN la{eo, d.K, Kdim, d.K};
   Im2colTN lb{x, d, Kdim, N};
   launch_gemm_tn(la, lb, gw, M, N, Kdim, beta, s);
+  return false;   // caller computes the bias gradient separately
 }
