@@ -121,6 +121,15 @@ class FullBatchLoader(Loader):
             self.minibatch_data.devmem[size:] = 0
             self.minibatch_labels.mem[size:] = -1
 
+    def __getstate__(self):
+        # the pinned staging buffer / CUDA event are device-session
+        # state, not snapshot state (recreated lazily on next fill)
+        state = super(FullBatchLoader, self).__getstate__()
+        for k in ("_idx_pin", "_idx_dev", "_idx_evt", "_idx_evt_live",
+                  "_idx_key"):
+            state.pop(k, None)
+        return state
+
 
 class FullBatchLoaderMSE(FullBatchLoader):
     """Adds regression targets (for MSE evaluators / autoencoders)."""

@@ -60,6 +60,13 @@ class DropoutForward(Forward):
                 (1,), int(self.counter), dtype=torch.int64,
                 device=device.torch_device)
 
+    def __getstate__(self):
+        # the device-side PRNG counter is session state; snapshots carry
+        # the host counter (initialize() re-seeds the device copy)
+        state = super(DropoutForward, self).__getstate__()
+        state.pop("_ctr_dev", None)
+        return state
+
     def numpy_run(self):
         self.input.map_read()
         self.output.map_invalidate()

@@ -550,7 +550,7 @@ struct Im2colTNv {
 template <typename LA, typename LB>
 __global__ __launch_bounds__(256, 2) void gemm_tn2_k(
     LA la, LB lb, float* __restrict__ C, int M, int N, int K, float beta,
-    int tiles_n, float* __restrict__ bias_out) {
+    int tiles_n) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const int tid = threadIdx.x;
   const int wave = tid >> 6, lane = tid & 63;
@@ -564,10 +564,6 @@ __global__ __launch_bounds__(256, 2) void gemm_tn2_k(
   const int t_hi = (int)(((long)nk * (sid + 1)) / sp);
   const int moA = m0 >> 3, moB = n0 >> 3;
 
-  // fused bias gradient (see gemm_tn2n_k): column sums of the staged A
-  float bsum[8] = {};
-  const bool do_bias = bias_out != nullptr && tn == 0;
-
   auto stage2 = [&](int buf, int kt) {
     char* As = smem + buf * TN2_LDS_BUF;
     char* Bs = As + TN2_LDS_OP;
@@ -580,9 +576,6 @@ __global__ __launch_bounds__(256, 2) void gemm_tn2_k(
       unsigned short v[8];
       la.load16(k0t + kk, moA + oct, v);
       *(short8*)(As + (kk * TNP + oct * 8) * 2) = *(short8*)v;
-      if (do_bias)
-#pragma unroll
-        for (int j = 0; j < 8; ++j) bsum[j] += bf2f(v[j]);
       lb.load16(k0t + kk, moB + oct, v);
       *(short8*)(Bs + (kk * TNP + oct * 8) * 2) = *(short8*)v;
     }
@@ -641,26 +634,6 @@ __global__ __launch_bounds__(256, 2) void gemm_tn2_k(
     }
     __syncthreads();
     buf ^= 1;
-  }
-
-  if (do_bias) {
-    float* red = (float*)smem;
-    const int oct = tid & 15;
-    const int kk0 = tid >> 4;
-#pragma unroll
-    for (int j = 0; j < 8; ++j)
-      red[kk0 * BM + oct * 8 + j] = bsum[j];
-    __syncthreads();
-    if (tid < BM && m0 + tid < M) {
-      float s = 0.f;
-#pragma unroll 4
-      for (int r = 0; r < 16; ++r) s += red[r * BM + tid];
-      if (sp > 1)
-        unsafeAtomicAdd(bias_out + m0 + tid, s);
-      else
-        bias_out[m0 + tid] =
-            (beta == 0.f ? 0.f : beta * bias_out[m0 + tid]) + s;
-    }
   }
 
   const int colw = n0 + wc * 64 + (lane & 15);
@@ -911,11 +884,15 @@ void launch_gemm_tn(const LA& la, const LB& lb, float* C, int M, int N,
 }
 
 template <typename LA, typename LB>
-void launch_gemm_tn2(const LA& la, const LB& lb, float* C, int M, int N,
+bool launch_gemm_tn2(const LA& la, const LB& lb, float* C, int M, int N,
                      int K, float beta, hipStream_t stream,
                      float* bias_out = nullptr) {
   const int tn = cdiv(N, BN);
   const int bmt = M <= 32 ? 32 : (M <= 64 ? 64 : BM);
+  // bias fusion is narrow-tile only: in the 128-row kernel the extra
+  // staging registers cost more on the big fc/conv wgrads than the
+  // separate col_sum kernel does
+  if (bmt == BM) bias_out = nullptr;
   const int tiles = cdiv(M, bmt) * tn;
   const int nk = cdiv(K, TBK);
   const int splits = max(1, min(nk, 1024 / tiles));
@@ -948,8 +925,9 @@ void launch_gemm_tn2(const LA& la, const LB& lb, float* C, int M, int N,
                        tn, bias_out);
   else
     hipLaunchKernelGGL((gemm_tn2_k<LA, LB>), grid, block, TN2_LDS_TOTAL,
-                       stream, la, lb, C, M, N, K, beta, tn, bias_out);
+                       stream, la, lb, C, M, N, K, beta, tn);
   HIP_CHECK_LAST();
+  return bias_out != nullptr;
 }
 
 // explicit entry points used by bindings.cpp --------------------------------
@@ -1009,8 +987,7 @@ bool gemm_tn_dense(const unsigned short* A, const unsigned short* B,
                    hipStream_t s, float* bias_out) {
   if ((M & 7) == 0 && (N & 7) == 0) {
     DenseTNv la{A, M, K, M}, lb{B, N, K, N};
-    launch_gemm_tn2(la, lb, C, M, N, K, beta, s, bias_out);
-    return bias_out != nullptr;
+    return launch_gemm_tn2(la, lb, C, M, N, K, beta, s, bias_out);
   }
   DenseTN la{A, M, K, M}, lb{B, N, K, N};
   launch_gemm_tn(la, lb, C, M, N, K, beta, s);
@@ -1078,8 +1055,7 @@ bool conv_wgrad_gemm(const unsigned short* eo, const unsigned short* x,
   if ((M & 7) == 0 && (d.C & 7) == 0) {
     DenseTNv la{eo, d.K, Kdim, d.K};
     Im2colTNv lb{x, d, Kdim, N};
-    launch_gemm_tn2(la, lb, gw, M, N, Kdim, beta, s, bias_out);
-    return bias_out != nullptr;
+    return launch_gemm_tn2(la, lb, gw, M, N, Kdim, beta, s, bias_out);
   }
   DenseTN la{eo, d.K, Kdim, d.K};
   Im2colTN lb{x, d, Kdim, N};
