@@ -352,9 +352,12 @@ def test_cifar_cnn_trains_on_gpu():
     from veles.loader.base import TRAIN
     prng.get().seed(4321)
     prng.get("loader").seed(8765)
+    # 4 epochs: at 3 the trajectory sits within a few points of the
+    # threshold and run-to-run fp32-atomic reduction order (split-K
+    # wgrad, fused bias sums) can push it past 60%
     root.cifar.update({"loader": {"minibatch_size": 128, "n_train": 512,
                                   "n_valid": 128},
-                       "decision": {"max_epochs": 3}})
+                       "decision": {"max_epochs": 4}})
     from veles.znicz_amd.models.cifar import CifarWorkflow
     wf = CifarWorkflow(DummyLauncher())
     wf.initialize(device=ROCmDevice(0))
