@@ -395,3 +395,16 @@ def test_sgd_multi_matches_sgd_step():
         if b2 is not None:
             numpy.testing.assert_allclose(
                 p.bf16.float().cpu().numpy(), b2.float().cpu().numpy())
+
+
+def test_permute_kc_matches_torch():
+    import torch
+    from veles.znicz_amd import ops
+    for (K, P, C) in [(96, 121, 8), (384, 9, 256), (32, 25, 8),
+                      (33, 9, 20)]:
+        src = torch.randn(K, P, C, device="cuda").bfloat16().contiguous()
+        dst = torch.empty(C, P, K, device="cuda", dtype=torch.bfloat16)
+        ops.require().permute_kc(src.view(-1), dst.view(-1), K, P, C)
+        want = src.permute(2, 1, 0).contiguous()
+        torch.cuda.synchronize()
+        assert torch.equal(dst, want), (K, P, C)

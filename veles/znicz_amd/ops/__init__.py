@@ -213,9 +213,11 @@ def attach_conv_copies(weights_arr):
 
 
 def refresh_conv_copies(weights_arr):
-    w = weights_arr.devmem          # fp32 master, already updated
-    weights_arr.bf16_dgrad.copy_(
-        weights_arr.bf16.view(w.shape).permute(3, 1, 2, 0))
+    # [K,KY,KX,C] bf16 working copy -> [C,KY,KX,K] dgrad image, one
+    # LDS-tiled HIP transpose (torch's strided copy_ was ~1 TB/s)
+    K, KY, KX, C = weights_arr.devmem.shape
+    require().permute_kc(weights_arr.bf16.view(-1),
+                         weights_arr.bf16_dgrad.view(-1), K, KY * KX, C)
 
 
 def refresh_fc_copies(weights_arr):

@@ -33,6 +33,8 @@ void col_sum_launch(const unsigned short*, float*, int, int, bool,
                     hipStream_t);
 void transpose2d_launch(const unsigned short*, unsigned short*, int, int,
                         hipStream_t);
+void permute_kc_launch(const unsigned short*, unsigned short*, int, int,
+                       int, hipStream_t);
 void gather_off_launch(const unsigned short*, const int*, unsigned short*,
                        long, hipStream_t);
 void scatter_off_launch(const unsigned short*, const int*, unsigned short*,
@@ -245,6 +247,14 @@ void sgd_multi(torch::Tensor desc_blob_dev, torch::Tensor b2t_dev) {
                    cur_stream());
 }
 
+void permute_kc(torch::Tensor src, torch::Tensor dst, int64_t K,
+                int64_t P, int64_t C) {
+  TORCH_CHECK(src.numel() == K * P * C && dst.numel() == src.numel(),
+              "permute_kc shape");
+  permute_kc_launch(ubf(src), ubf_mut(dst), (int)K, (int)P, (int)C,
+                    cur_stream());
+}
+
 void col_sum(torch::Tensor a, torch::Tensor out, bool accumulate) {
   col_sum_launch(ubf(a), out.data_ptr<float>(), a.size(0), a.size(1),
                  accumulate, cur_stream());
@@ -397,6 +407,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgd_step", &sgd_step);
   m.def("sgd_multi_prepare", &sgd_multi_prepare);
   m.def("sgd_multi", &sgd_multi);
+  m.def("permute_kc", &permute_kc);
   m.def("col_sum", &col_sum);
   m.def("transpose2d", &transpose2d);
   m.def("softmax_fwd", &softmax_fwd);

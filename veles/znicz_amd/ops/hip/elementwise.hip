@@ -143,6 +143,40 @@ void zero_f32_launch(float* p, long n, hipStream_t s) {
   HIP_CHECK_LAST();
 }
 
+// conv dgrad-image refresh: [K, p, C] -> [C, p, K] (p = KH·KW plane
+// index).  LDS-tiled batched transpose — replaces a torch strided
+// copy_ whose 2-B scattered reads ran at ~1 TB/s.
+#define PTT 32
+__global__ void permute_kc_k(const unsigned short* __restrict__ src,
+                             unsigned short* __restrict__ dst, int K,
+                             int P, int C) {
+  __shared__ unsigned short t[PTT][PTT + 1];
+  const int p = blockIdx.z;
+  const int k0 = blockIdx.x * PTT, c0 = blockIdx.y * PTT;
+  const int tx = threadIdx.x & 31, ty = threadIdx.x >> 5;   // 32 x 8
+#pragma unroll
+  for (int dy = 0; dy < PTT; dy += 8) {
+    const int k = k0 + ty + dy, c = c0 + tx;
+    if (k < K && c < C)
+      t[ty + dy][tx] = src[((long)k * P + p) * C + c];
+  }
+  __syncthreads();
+#pragma unroll
+  for (int dy = 0; dy < PTT; dy += 8) {
+    const int c = c0 + ty + dy, k = k0 + tx;
+    if (c < C && k < K)
+      dst[((long)c * P + p) * K + k] = t[tx][ty + dy];
+  }
+}
+
+void permute_kc_launch(const unsigned short* src, unsigned short* dst,
+                       int K, int P, int C, hipStream_t s) {
+  dim3 grid((K + PTT - 1) / PTT, (C + PTT - 1) / PTT, P);
+  hipLaunchKernelGGL(permute_kc_k, grid, dim3(256), 0, s, src, dst, K, P,
+                     C);
+  HIP_CHECK_LAST();
+}
+
 // zero two disjoint buffers in one launch (split-K C workspace + the
 // fused bias-gradient accumulator)
 __global__ void zero2_f32_k(float* __restrict__ p, long n,
