@@ -64,11 +64,11 @@ class LRNormalizerForward(Forward):
 
     def hip_run(self):
         from veles.znicz_amd import ops
-        ops.lrn_fwd(self.input.devmem, self.output.devmem,
-                    self.denominators.devmem, self.alpha, self.beta,
-                    self.k, self.n)
+        # no denom store: the backward kernel recomputes it from its
+        # own x halo — saves the fp32 round-trip (memory-wait bound)
+        ops.lrn_fwd(self.input.devmem, self.output.devmem, None,
+                    self.alpha, self.beta, self.k, self.n)
         self.output.touch_device()
-        self.denominators.touch_device()
 
 
 class LRNormalizerBackward(GradientDescentBase):
@@ -125,6 +125,6 @@ class LRNormalizerBackward(GradientDescentBase):
     def hip_run(self):
         from veles.znicz_amd import ops
         ops.lrn_bwd(self.err_output.devmem, self.input.devmem, None,
-                    self.denominators.devmem, self.err_input.devmem,
+                    None, self.err_input.devmem,
                     self.alpha, self.beta, self.k, self.n)
         self.err_input.touch_device()

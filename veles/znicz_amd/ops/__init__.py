@@ -271,6 +271,8 @@ def pool_bwd(err_out, err_in, offs, desc, mode):
 
 
 def lrn_fwd(x, out, denom, alpha, beta, k, n):
+    """denom (fp32, optional): pass None on the training path — the
+    backward kernel recomputes the denominators from x."""
     require().lrn_fwd(x, out, denom, float(alpha), float(beta), float(k),
                       int(n))
 

@@ -69,7 +69,7 @@ void lrn_fwd_launch(const unsigned short*, unsigned short*, float*, long,
                     int, float, float, float, int, hipStream_t);
 void lrn_bwd_launch(const unsigned short*, const unsigned short*,
                     const float*, float*, unsigned short*, long, int,
-                    float, float, int, hipStream_t);
+                    float, float, float, int, hipStream_t);
 
 namespace {
 
@@ -305,23 +305,28 @@ void pool_bwd(torch::Tensor ey, torch::Tensor ex, torch::Tensor offs,
                   cur_stream());
 }
 
-void lrn_fwd(torch::Tensor x, torch::Tensor y, torch::Tensor denom,
-             double alpha, double beta, double k, int64_t win) {
+void lrn_fwd(torch::Tensor x, torch::Tensor y,
+             c10::optional<torch::Tensor> denom, double alpha,
+             double beta, double k, int64_t win) {
   const int C = x.size(-1);
-  lrn_fwd_launch(ubf(x), ubf_mut(y), denom.data_ptr<float>(),
-                 x.numel() / C, C, (float)alpha, (float)beta, (float)k,
-                 (int)win, cur_stream());
+  // denom is optional: the backward recomputes it from x, so the hot
+  // path skips the fp32 round-trip entirely
+  float* dp = denom.has_value() ? denom->data_ptr<float>() : nullptr;
+  lrn_fwd_launch(ubf(x), ubf_mut(y), dp, x.numel() / C, C, (float)alpha,
+                 (float)beta, (float)k, (int)win, cur_stream());
 }
 
 void lrn_bwd(torch::Tensor ey, torch::Tensor x,
-             c10::optional<torch::Tensor> y, torch::Tensor denom,
-             torch::Tensor ex, double alpha, double beta, double k,
-             int64_t win) {
+             c10::optional<torch::Tensor> y,
+             c10::optional<torch::Tensor> denom, torch::Tensor ex,
+             double alpha, double beta, double k, int64_t win) {
   const int C = x.size(-1);
-  (void)y; (void)k;
-  lrn_bwd_launch(ubf(ey), ubf(x), denom.data_ptr<float>(), nullptr,
-                 ubf_mut(ex), x.numel() / C, C, (float)alpha,
-                 (float)beta, (int)win, cur_stream());
+  (void)y;
+  const float* dp = denom.has_value() ? denom->data_ptr<float>()
+                                      : nullptr;
+  lrn_bwd_launch(ubf(ey), ubf(x), dp, nullptr, ubf_mut(ex),
+                 x.numel() / C, C, (float)alpha, (float)beta, (float)k,
+                 (int)win, cur_stream());
 }
 
 void dropout_fwd(torch::Tensor x, torch::Tensor y, torch::Tensor mask,

@@ -355,22 +355,29 @@ __global__ void lrn_fwd_k(const unsigned short* __restrict__ x,
                           float* __restrict__ denom, unsigned rows, int C,
                           int lgo, float alpha, float beta, float kk,
                           bool b34) {
-  const int O2 = 1 << lgo;               // pow2 octet slots per row
+  // OPT adjacent octets per thread-slot: the ±1-octet halo is amortized
+  // over 4 output octets ((OPT+2)/OPT load ratio instead of 3×), and
+  // denom is OPTIONAL — the backward recomputes it from its own x halo,
+  // which removes the fp32 round-trip entirely (PMC: both kernels were
+  // 65% memory-wait).
+  constexpr int OPT = 4;
+  const int O2 = 1 << lgo;               // pow2 slot count per row
   const int o0 = threadIdx.x & (O2 - 1);
   const int rin = threadIdx.x >> lgo;
   const int rpb = blockDim.x >> lgo;
-  constexpr int half = WIN / 2;          // compile-time: v[] stays in regs
+  constexpr int half = WIN / 2;
   const int noct = (C + 7) >> 3;
+  const int nslot = (noct + OPT - 1) / OPT;
   for (unsigned row = blockIdx.x * rpb + rin; row < rows;
        row += gridDim.x * rpb) {
     const unsigned short* xr = x + (long)row * C;
-    float* dr = denom + (long)row * C;
+    float* dr = denom ? denom + (long)row * C : nullptr;
     unsigned short* yr = y + (long)row * C;
-    for (int oc = o0; oc < noct; oc += O2) {
-      const int cb = oc * 8;
-      float v[24];
+    for (int sl = o0; sl < nslot; sl += O2) {
+      const int cb = sl * OPT * 8;
+      float v[(OPT + 2) * 8];
 #pragma unroll
-      for (int h = 0; h < 3; ++h) {
+      for (int h = 0; h < OPT + 2; ++h) {
         const int c8 = cb + (h - 1) * 8;
         if (c8 >= 0 && c8 + 8 <= C) {
           unsigned short t8[8];
@@ -385,71 +392,84 @@ __global__ void lrn_fwd_k(const unsigned short* __restrict__ x,
           }
         }
       }
-      unsigned short o8[8];
-      float d8[8];
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        float s = 0.f;
+      for (int t = 0; t < OPT; ++t) {
+        const int co = cb + t * 8;
+        if (co >= C) break;
+        unsigned short o8[8];
+        float d8[8];
 #pragma unroll
-        for (int w = -half; w <= half; ++w) {
-          float t = v[8 + j + w];
-          s += t * t;
-        }
-        float d = kk + alpha * s;
-        d8[j] = d;
-        o8[j] = f2bf(v[8 + j] * pow_nbeta(d, beta, b34));
-      }
-      if (cb + 8 <= C) {
-        *(short8*)(yr + cb) = *(short8*)o8;
-        *(f32x4*)(dr + cb) = *(f32x4*)d8;
-        *(f32x4*)(dr + cb + 4) = *(f32x4*)(d8 + 4);
-      } else {
+        for (int j = 0; j < 8; ++j) {
+          float s = 0.f;
 #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          if (cb + j < C) {
-            yr[cb + j] = o8[j];
-            dr[cb + j] = d8[j];
+          for (int w = -half; w <= half; ++w) {
+            float tt = v[(t + 1) * 8 + j + w];
+            s += tt * tt;
           }
+          float d = kk + alpha * s;
+          d8[j] = d;
+          o8[j] = f2bf(v[(t + 1) * 8 + j] * pow_nbeta(d, beta, b34));
+        }
+        if (co + 8 <= C) {
+          *(short8*)(yr + co) = *(short8*)o8;
+          if (dr) {
+            *(f32x4*)(dr + co) = *(f32x4*)d8;
+            *(f32x4*)(dr + co + 4) = *(f32x4*)(d8 + 4);
+          }
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            if (co + j < C) {
+              yr[co + j] = o8[j];
+              if (dr) dr[co + j] = d8[j];
+            }
+        }
       }
     }
   }
 }
 
-//   fused backward:  dx_c = dy_c·D_c^{-β} − 2αβ·x_c · Σ_{|j−c|≤half} t_j
-// with t_j = dy_j·x_j·D_j^{-β-1} recomputed in registers for the octet
-// plus its halo — no fp32 workspace round-trip (a separate pre-pass
-// cost a full extra read+write of the tensor).
 template <int WIN>
 __global__ void lrn_bwd_k(const unsigned short* __restrict__ ey,
                           const unsigned short* __restrict__ x,
                           const float* __restrict__ denom,
                           unsigned short* __restrict__ ex, unsigned rows,
                           int C, int lgo, float alpha, float beta,
-                          bool b34) {
+                          float kk, bool b34) {
+  // denom == nullptr: recompute d_j = kk + α·Σ x² in registers — the
+  // ±(2·half)-wide x span needed fits inside the same ±1-octet halo
+  // for WIN ≤ 9, trading ~60 extra fma per octet for a full fp32
+  // tensor read (the kernel is memory-wait bound, the VALU is idle).
+  constexpr int OPT = 4;
   const int O2 = 1 << lgo;
   const int o0 = threadIdx.x & (O2 - 1);
   const int rin = threadIdx.x >> lgo;
   const int rpb = blockDim.x >> lgo;
   constexpr int half = WIN / 2;
+  static_assert(2 * half <= 8, "halo octet must cover d's x-span");
   const int noct = (C + 7) >> 3;
+  const int nslot = (noct + OPT - 1) / OPT;
   for (unsigned row = blockIdx.x * rpb + rin; row < rows;
        row += gridDim.x * rpb) {
     const unsigned short* er = ey + (long)row * C;
     const unsigned short* xr = x + (long)row * C;
-    const float* dr = denom + (long)row * C;
+    const float* dr = denom ? denom + (long)row * C : nullptr;
     unsigned short* or_ = ex + (long)row * C;
-    for (int oc = o0; oc < noct; oc += O2) {
-      const int cb = oc * 8;
-      float ev[24], xv[24], dv[24];
+    for (int sl = o0; sl < nslot; sl += O2) {
+      const int cb = sl * OPT * 8;
+      constexpr int NV = (OPT + 2) * 8;
+      float ev[NV], xv[NV], dv[NV];
 #pragma unroll
-      for (int h = 0; h < 3; ++h) {
+      for (int h = 0; h < OPT + 2; ++h) {
         const int c8 = cb + (h - 1) * 8;
         if (c8 >= 0 && c8 + 8 <= C) {
           unsigned short e8[8], x8[8];
           *(short8*)e8 = *(const short8*)(er + c8);
           *(short8*)x8 = *(const short8*)(xr + c8);
-          *(f32x4*)(dv + h * 8) = *(const f32x4*)(dr + c8);
-          *(f32x4*)(dv + h * 8 + 4) = *(const f32x4*)(dr + c8 + 4);
+          if (dr) {
+            *(f32x4*)(dv + h * 8) = *(const f32x4*)(dr + c8);
+            *(f32x4*)(dv + h * 8 + 4) = *(const f32x4*)(dr + c8 + 4);
+          }
 #pragma unroll
           for (int j = 0; j < 8; ++j) {
             ev[h * 8 + j] = bf2f(e8[j]);
@@ -462,33 +482,53 @@ __global__ void lrn_bwd_k(const unsigned short* __restrict__ ey,
             bool ok = c >= 0 && c < C;
             ev[h * 8 + j] = ok ? bf2f(er[c]) : 0.f;
             xv[h * 8 + j] = ok ? bf2f(xr[c]) : 0.f;
-            dv[h * 8 + j] = ok ? dr[c] : 1.f;
+            if (dr) dv[h * 8 + j] = ok ? dr[c] : 1.f;
           }
         }
       }
-      float tv[8 + 2 * half];
+      if (!dr) {
+        // recompute d over every position whose t_j a window touches:
+        // [cb-half, cb+OPT·8+half) ⊂ [cb-8+half, cb+(OPT+1)·8-half)
 #pragma unroll
-      for (int j = 0; j < 8 + 2 * half; ++j) {
-        const int q = 8 - half + j;          // window [cb-half, cb+7+half]
-        tv[j] = ev[q] * xv[q] * pow_nbeta(dv[q], beta, b34) *
-                __builtin_amdgcn_rcpf(dv[q]);
+        for (int q = 8 - half; q < NV - 8 + half; ++q) {
+          float s = 0.f;
+#pragma unroll
+          for (int w = -half; w <= half; ++w) {
+            float tt = (q + w >= 0 && q + w < NV) ? xv[q + w] : 0.f;
+            s += tt * tt;
+          }
+          dv[q] = kk + alpha * s;
+        }
       }
-      unsigned short o8[8];
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        float acc = 0.f;
+      for (int t = 0; t < OPT; ++t) {
+        const int co = cb + t * 8;
+        if (co >= C) break;
+        const int b = (t + 1) * 8;           // center base in v[]
+        float tv[8 + 2 * half];
 #pragma unroll
-        for (int w = 0; w < WIN; ++w) acc += tv[j + w];
-        float v = ev[8 + j] * pow_nbeta(dv[8 + j], beta, b34) -
-                  2.f * alpha * beta * xv[8 + j] * acc;
-        o8[j] = f2bf(v);
-      }
-      if (cb + 8 <= C) {
-        *(short8*)(or_ + cb) = *(short8*)o8;
-      } else {
+        for (int j = 0; j < 8 + 2 * half; ++j) {
+          const int q = b - half + j;        // window [co-half, co+7+half]
+          tv[j] = ev[q] * xv[q] * pow_nbeta(dv[q], beta, b34) *
+                  __builtin_amdgcn_rcpf(dv[q]);
+        }
+        unsigned short o8[8];
 #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          if (cb + j < C) or_[cb + j] = o8[j];
+        for (int j = 0; j < 8; ++j) {
+          float acc = 0.f;
+#pragma unroll
+          for (int w = 0; w < WIN; ++w) acc += tv[j + w];
+          float vv = ev[b + j] * pow_nbeta(dv[b + j], beta, b34) -
+                     2.f * alpha * beta * xv[b + j] * acc;
+          o8[j] = f2bf(vv);
+        }
+        if (co + 8 <= C) {
+          *(short8*)(or_ + co) = *(short8*)o8;
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            if (co + j < C) or_[co + j] = o8[j];
+        }
       }
     }
   }
@@ -506,9 +546,9 @@ __global__ void lrn_bwd_k(const unsigned short* __restrict__ ey,
   }
 
 static inline int lrn_lgo(int C) {
-  int noct = (C + 7) >> 3;
+  const int nslot = (((C + 7) >> 3) + 3) / 4;   // OPT=4 octets per slot
   int lg = 0;
-  while ((1 << lg) < noct && lg < 5) ++lg;
+  while ((1 << lg) < nslot && lg < 5) ++lg;
   return lg;
 }
 
@@ -530,8 +570,8 @@ void lrn_fwd_launch(const unsigned short* x, unsigned short* y,
 
 void lrn_bwd_launch(const unsigned short* ey, const unsigned short* x,
                     const float* denom, float* ws, unsigned short* ex,
-                    long rows, int C, float alpha, float beta, int win,
-                    hipStream_t s) {
+                    long rows, int C, float alpha, float beta, float kk,
+                    int win, hipStream_t s) {
   (void)ws;                       // fused kernel needs no workspace
   const bool b34 = beta == 0.75f;
   const int lgo = lrn_lgo(C);
@@ -539,7 +579,7 @@ void lrn_bwd_launch(const unsigned short* ey, const unsigned short* x,
   dim3 grid((unsigned)min((rows + rpb - 1) / rpb, (long)131072));
 #define GO(W)                                                               \
   hipLaunchKernelGGL(lrn_bwd_k<W>, grid, dim3(256), 0, s, ey, x, denom,     \
-                     ex, (unsigned)rows, C, lgo, alpha, beta, b34)
+                     ex, (unsigned)rows, C, lgo, alpha, beta, kk, b34)
   LRN_WIN_SWITCH(win, GO);
 #undef GO
   HIP_CHECK_LAST();
