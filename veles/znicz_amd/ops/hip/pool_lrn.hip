@@ -356,11 +356,14 @@ __global__ void lrn_fwd_k(const unsigned short* __restrict__ x,
                           int lgo, float alpha, float beta, float kk,
                           bool b34) {
   // OPT adjacent octets per thread-slot: the ±1-octet halo is amortized
-  // over 4 output octets ((OPT+2)/OPT load ratio instead of 3×), and
+  // over OPT output octets ((OPT+2)/OPT load ratio instead of 3×;
+  // OPT=2 — OPT=4 measured SLOWER: it cut the thread count to ~1 wave
+  // per SIMD slot and the memory-wait-bound kernel lost the
+  // concurrency that hides latency), and
   // denom is OPTIONAL — the backward recomputes it from its own x halo,
   // which removes the fp32 round-trip entirely (PMC: both kernels were
   // 65% memory-wait).
-  constexpr int OPT = 4;
+  constexpr int OPT = 2;
   const int O2 = 1 << lgo;               // pow2 slot count per row
   const int o0 = threadIdx.x & (O2 - 1);
   const int rin = threadIdx.x >> lgo;
@@ -440,7 +443,7 @@ __global__ void lrn_bwd_k(const unsigned short* __restrict__ ey,
   // ±(2·half)-wide x span needed fits inside the same ±1-octet halo
   // for WIN ≤ 9, trading ~60 extra fma per octet for a full fp32
   // tensor read (the kernel is memory-wait bound, the VALU is idle).
-  constexpr int OPT = 4;
+  constexpr int OPT = 2;
   const int O2 = 1 << lgo;
   const int o0 = threadIdx.x & (O2 - 1);
   const int rin = threadIdx.x >> lgo;
@@ -546,7 +549,7 @@ __global__ void lrn_bwd_k(const unsigned short* __restrict__ ey,
   }
 
 static inline int lrn_lgo(int C) {
-  const int nslot = (((C + 7) >> 3) + 3) / 4;   // OPT=4 octets per slot
+  const int nslot = (((C + 7) >> 3) + 1) / 2;   // OPT=2 octets per slot
   int lg = 0;
   while ((1 << lg) < nslot && lg < 5) ++lg;
   return lg;
