@@ -80,27 +80,34 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_k(
       const int k8e = k8 ^ rl ^ (chunk & 1);   // swizzled source octet
       const int kc0 = (k8base + k8e) * 8;
       const bool straddle = (kc0 < K) & (kc0 + 8 > K);
-      if (!straddle) {
-        const void* sa = la.addr(min(m0 + row, M - 1), k8base + k8e);
+      // loaders may refuse the 16-B DMA for a chunk (addr() == nullptr:
+      // C%8 im2col kh-row crossings / misalignment) — per-element path
+      const void* sa =
+          straddle ? nullptr : la.addr(min(m0 + row, M - 1), k8base + k8e);
+      const void* sb =
+          straddle ? nullptr : lb.addr(min(n0 + row, N - 1), k8base + k8e);
+      if (sa) {
         __builtin_amdgcn_global_load_lds(
             (const __attribute__((address_space(1))) void*)sa,
             (__attribute__((address_space(3))) void*)(As + chunk * 1024),
             16, 0, 0);
-        const void* sb = lb.addr(min(n0 + row, N - 1), k8base + k8e);
+      } else {
+        unsigned short va[8];
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          va[j] = la.elem(min(m0 + row, M - 1), kc0 + j);
+        *(short8*)(As + chunk * 1024 + lane * 16) = *(short8*)va;
+      }
+      if (sb) {
         __builtin_amdgcn_global_load_lds(
             (const __attribute__((address_space(1))) void*)sb,
             (__attribute__((address_space(3))) void*)(Bs + chunk * 1024),
             16, 0, 0);
       } else {
-        // rare slow path: per-element guarded loads + LDS write at the
-        // same slot the DMA would have filled (lane*16 within chunk)
-        unsigned short va[8], vb[8];
+        unsigned short vb[8];
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          va[j] = la.elem(min(m0 + row, M - 1), kc0 + j);
+        for (int j = 0; j < 8; ++j)
           vb[j] = lb.elem(min(n0 + row, N - 1), kc0 + j);
-        }
-        *(short8*)(As + chunk * 1024 + lane * 16) = *(short8*)va;
         *(short8*)(Bs + chunk * 1024 + lane * 16) = *(short8*)vb;
       }
     }
@@ -210,8 +217,10 @@ __global__ __launch_bounds__(256, 2) void gemm_ntn_k(
       const int k8e = k8 ^ rl ^ (chunk & 1);
       const int kc0 = (k8base + k8e) * 8;
       const bool straddle = (kc0 < K) & (kc0 + 8 > K);
-      if (!straddle) {
-        const void* sa = la.addr(min(m0 + row, M - 1), k8base + k8e);
+      const void* sa = straddle ? nullptr
+                                : la.addr(min(m0 + row, M - 1),
+                                          k8base + k8e);
+      if (sa) {
         __builtin_amdgcn_global_load_lds(
             (const __attribute__((address_space(1))) void*)sa,
             (__attribute__((address_space(3))) void*)(As + chunk * 1024),
@@ -231,8 +240,10 @@ __global__ __launch_bounds__(256, 2) void gemm_ntn_k(
       const int k8e = k8 ^ rl ^ (chunk & 1);
       const int kc0 = (k8base + k8e) * 8;
       const bool straddle = (kc0 < K) & (kc0 + 8 > K);
-      if (!straddle) {
-        const void* sb = lb.addr(min(n0 + row, N - 1), k8base + k8e);
+      const void* sb = straddle ? nullptr
+                                : lb.addr(min(n0 + row, N - 1),
+                                          k8base + k8e);
+      if (sb) {
         __builtin_amdgcn_global_load_lds(
             (const __attribute__((address_space(1))) void*)sb,
             (__attribute__((address_space(3))) void*)(Bs + chunk * 1024),
@@ -536,6 +547,40 @@ struct Im2colTNv {
     int oh = t2 % d.OH, n = t2 / d.OH;
     int ih = oh * d.SY - d.PT + kh;
     int iw = ow * d.SX - d.PL + kw;
+    if (d.C & 7) {
+      // C%8 (e.g. RGB->4): the 8-m span covers several contiguous
+      // (kw,c) pairs of one kh row; fall to per-element loads when it
+      // crosses the row, padding is in play, or the source is not
+      // 16-B aligned (mirrors Im2colNT::addr)
+      const int rowlen = d.KW * d.C;
+      long off = (((long)n * d.H + ih) * d.W + iw) * d.C + c;
+      const bool fast = ((m % rowlen) + 8 <= rowlen) &&
+                        !(d.PT | d.PL) && !(off & 3) && (k < Klim) &&
+                        (m + 8 <= Mlim) && (kh < d.KH) &&
+                        ((unsigned)ih < (unsigned)d.H);
+      if (fast) {
+        // two 8-B loads: off is only guaranteed 4-elem aligned
+        *(unsigned long long*)v =
+            *(const unsigned long long*)(x + off);
+        *((unsigned long long*)v + 1) =
+            *((const unsigned long long*)(x + off) + 1);
+        return;
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int mj = m + j;
+        int cj = mj % d.C, tj = mj / d.C;
+        int kwj = tj % d.KW, khj = tj / d.KW;
+        int ihj = oh * d.SY - d.PT + khj;
+        int iwj = ow * d.SX - d.PL + kwj;
+        const bool okj = (k < Klim) & (mj < Mlim) & (khj < d.KH) &
+                         ((unsigned)ihj < (unsigned)d.H) &
+                         ((unsigned)iwj < (unsigned)d.W);
+        v[j] = okj ? x[(((long)n * d.H + ihj) * d.W + iwj) * d.C + cj]
+                   : (unsigned short)0;
+      }
+      return;
+    }
     const bool ok = (k < Klim) & (m < Mlim) & (kh < d.KH) &
                     ((unsigned)ih < (unsigned)d.H) &
                     ((unsigned)iw < (unsigned)d.W);
@@ -1052,7 +1097,7 @@ bool conv_wgrad_gemm(const unsigned short* eo, const unsigned short* x,
   const int Kdim = d.N * d.OH * d.OW;
   const int M = d.K;                   // n_kernels
   const int N = d.KH * d.KW * d.C;
-  if ((M & 7) == 0 && (d.C & 7) == 0) {
+  if ((M & 7) == 0 && (d.C & 3) == 0) {
     DenseTNv la{eo, d.K, Kdim, d.K};
     Im2colTNv lb{x, d, Kdim, N};
     return launch_gemm_tn2(la, lb, gw, M, N, Kdim, beta, s, bias_out);

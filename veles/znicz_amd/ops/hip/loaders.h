@@ -22,7 +22,13 @@ struct DenseNT {
 };
 
 // implicit im2col over NHWC input: row=(n,oh,ow), kdim=(kh,kw,c) c-inner.
-// Requires C % 8 == 0 so each 16-B chunk stays inside one pixel's C-run.
+// C % 8 == 0: every 16-B chunk stays inside one pixel's C-run.
+// C % 8 != 0 (e.g. RGB padded only to 4): a chunk spans several (kw,c)
+// pairs, which ARE contiguous in NHWC within one kh row when there is
+// no horizontal padding; addr() returns nullptr for the chunks where
+// the fast DMA is illegal (kh-row crossing, padding in play, or a
+// misaligned 16-B source) and the staging loop falls back to the
+// per-element path for just those chunks.
 struct Im2colNT {
   const unsigned short* x;
   const unsigned short* zero;
@@ -36,6 +42,18 @@ struct Im2colNT {
     int oh = t % d.OH, n = t / d.OH;
     int ih = oh * d.SY - d.PT + kh;
     int iw = ow * d.SX - d.PL + kw;
+    if (d.C & 7) {
+      const int rowlen = d.KW * d.C;
+      if ((k % rowlen) + 8 > rowlen || (d.PT | d.PL))
+        return nullptr;                  // kh-cross or padded: slow path
+      if (kh >= d.KH || (unsigned)ih >= (unsigned)d.H)
+        return zero;
+      long off = (((long)n * d.H + ih) * d.W + iw) * d.C + c;
+      // 8-B source alignment suffices for the 16-B glds (probed:
+      // tools/probe_glds_align.hip); 4-elem-multiple offsets are
+      // guaranteed by C%4==0
+      return (off & 3) ? nullptr : (const void*)(x + off);
+    }
     if (kh >= d.KH || (unsigned)ih >= (unsigned)d.H ||
         (unsigned)iw >= (unsigned)d.W)
       return zero;
