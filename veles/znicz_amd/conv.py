@@ -64,9 +64,12 @@ class Conv(Forward):
         if self.input is None or not self.input:
             return False
         n, h, w, c = self.in_shape()
-        if self.on_gpu and c % 8 != 0:
+        if self.on_gpu and c % 4 != 0:
+            # C%4: the im2col staging packs contiguous (kw,c) pairs into
+            # its 16-B chunks (8-B-aligned glds is legal — probed), so
+            # RGB only needs padding to 4, not 8
             raise ValueError(
-                "HIP conv path needs C %% 8 == 0 (got C=%d); pad the "
+                "HIP conv path needs C %% 4 == 0 (got C=%d); pad the "
                 "input channels (the loaders do this for RGB)" % c)
         oh, ow = self.out_spatial()
         kshape = (self.n_kernels, self.ky, self.kx, c)
