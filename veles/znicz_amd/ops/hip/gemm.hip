@@ -80,35 +80,62 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_k(
       const int k8e = k8 ^ rl ^ (chunk & 1);   // swizzled source octet
       const int kc0 = (k8base + k8e) * 8;
       const bool straddle = (kc0 < K) & (kc0 + 8 > K);
-      // loaders may refuse the 16-B DMA for a chunk (addr() == nullptr:
-      // C%8 im2col kh-row crossings / misalignment) — per-element path
-      const void* sa =
-          straddle ? nullptr : la.addr(min(m0 + row, M - 1), k8base + k8e);
-      const void* sb =
-          straddle ? nullptr : lb.addr(min(n0 + row, N - 1), k8base + k8e);
-      if (sa) {
-        __builtin_amdgcn_global_load_lds(
-            (const __attribute__((address_space(1))) void*)sa,
-            (__attribute__((address_space(3))) void*)(As + chunk * 1024),
-            16, 0, 0);
-      } else {
-        unsigned short va[8];
+      if constexpr (!LA::MAY_SLOW && !LB::MAY_SLOW) {
+        // fast loaders: the only per-element case is the K%8 tail
+        if (!straddle) {
+          const void* sa = la.addr(min(m0 + row, M - 1), k8base + k8e);
+          __builtin_amdgcn_global_load_lds(
+              (const __attribute__((address_space(1))) void*)sa,
+              (__attribute__((address_space(3))) void*)(As + chunk * 1024),
+              16, 0, 0);
+          const void* sb = lb.addr(min(n0 + row, N - 1), k8base + k8e);
+          __builtin_amdgcn_global_load_lds(
+              (const __attribute__((address_space(1))) void*)sb,
+              (__attribute__((address_space(3))) void*)(Bs + chunk * 1024),
+              16, 0, 0);
+        } else {
+          unsigned short va[8], vb[8];
 #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          va[j] = la.elem(min(m0 + row, M - 1), kc0 + j);
-        *(short8*)(As + chunk * 1024 + lane * 16) = *(short8*)va;
-      }
-      if (sb) {
-        __builtin_amdgcn_global_load_lds(
-            (const __attribute__((address_space(1))) void*)sb,
-            (__attribute__((address_space(3))) void*)(Bs + chunk * 1024),
-            16, 0, 0);
+          for (int j = 0; j < 8; ++j) {
+            va[j] = la.elem(min(m0 + row, M - 1), kc0 + j);
+            vb[j] = lb.elem(min(n0 + row, N - 1), kc0 + j);
+          }
+          *(short8*)(As + chunk * 1024 + lane * 16) = *(short8*)va;
+          *(short8*)(Bs + chunk * 1024 + lane * 16) = *(short8*)vb;
+        }
       } else {
-        unsigned short vb[8];
+        // a packed loader may refuse individual chunks (addr()==nullptr:
+        // kh-row crossing / padding / misalignment) — per-operand path
+        const void* sa = straddle ? nullptr
+                                  : la.addr(min(m0 + row, M - 1),
+                                            k8base + k8e);
+        const void* sb = straddle ? nullptr
+                                  : lb.addr(min(n0 + row, N - 1),
+                                            k8base + k8e);
+        if (sa) {
+          __builtin_amdgcn_global_load_lds(
+              (const __attribute__((address_space(1))) void*)sa,
+              (__attribute__((address_space(3))) void*)(As + chunk * 1024),
+              16, 0, 0);
+        } else {
+          unsigned short va[8];
 #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          vb[j] = lb.elem(min(n0 + row, N - 1), kc0 + j);
-        *(short8*)(Bs + chunk * 1024 + lane * 16) = *(short8*)vb;
+          for (int j = 0; j < 8; ++j)
+            va[j] = la.elem(min(m0 + row, M - 1), kc0 + j);
+          *(short8*)(As + chunk * 1024 + lane * 16) = *(short8*)va;
+        }
+        if (sb) {
+          __builtin_amdgcn_global_load_lds(
+              (const __attribute__((address_space(1))) void*)sb,
+              (__attribute__((address_space(3))) void*)(Bs + chunk * 1024),
+              16, 0, 0);
+        } else {
+          unsigned short vb[8];
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            vb[j] = lb.elem(min(n0 + row, N - 1), kc0 + j);
+          *(short8*)(Bs + chunk * 1024 + lane * 16) = *(short8*)vb;
+        }
       }
     }
   };
@@ -217,10 +244,12 @@ __global__ __launch_bounds__(256, 2) void gemm_ntn_k(
       const int k8e = k8 ^ rl ^ (chunk & 1);
       const int kc0 = (k8base + k8e) * 8;
       const bool straddle = (kc0 < K) & (kc0 + 8 > K);
-      const void* sa = straddle ? nullptr
-                                : la.addr(min(m0 + row, M - 1),
-                                          k8base + k8e);
-      if (sa) {
+      const void* sa = (straddle || LA::MAY_SLOW)
+                           ? (straddle ? nullptr
+                                       : la.addr(min(m0 + row, M - 1),
+                                                 k8base + k8e))
+                           : la.addr(min(m0 + row, M - 1), k8base + k8e);
+      if (!LA::MAY_SLOW ? !straddle : (sa != nullptr)) {
         __builtin_amdgcn_global_load_lds(
             (const __attribute__((address_space(1))) void*)sa,
             (__attribute__((address_space(3))) void*)(As + chunk * 1024),
@@ -240,10 +269,12 @@ __global__ __launch_bounds__(256, 2) void gemm_ntn_k(
       const int k8e = k8 ^ rl ^ (chunk & 1);
       const int kc0 = (k8base + k8e) * 8;
       const bool straddle = (kc0 < K) & (kc0 + 8 > K);
-      const void* sb = straddle ? nullptr
-                                : lb.addr(min(n0 + row, N - 1),
-                                          k8base + k8e);
-      if (sb) {
+      const void* sb = (straddle || LB::MAY_SLOW)
+                           ? (straddle ? nullptr
+                                       : lb.addr(min(n0 + row, N - 1),
+                                                 k8base + k8e))
+                           : lb.addr(min(n0 + row, N - 1), k8base + k8e);
+      if (!LB::MAY_SLOW ? !straddle : (sb != nullptr)) {
         __builtin_amdgcn_global_load_lds(
             (const __attribute__((address_space(1))) void*)sb,
             (__attribute__((address_space(3))) void*)(Bs + chunk * 1024),
@@ -547,40 +578,6 @@ struct Im2colTNv {
     int oh = t2 % d.OH, n = t2 / d.OH;
     int ih = oh * d.SY - d.PT + kh;
     int iw = ow * d.SX - d.PL + kw;
-    if (d.C & 7) {
-      // C%8 (e.g. RGB->4): the 8-m span covers several contiguous
-      // (kw,c) pairs of one kh row; fall to per-element loads when it
-      // crosses the row, padding is in play, or the source is not
-      // 16-B aligned (mirrors Im2colNT::addr)
-      const int rowlen = d.KW * d.C;
-      long off = (((long)n * d.H + ih) * d.W + iw) * d.C + c;
-      const bool fast = ((m % rowlen) + 8 <= rowlen) &&
-                        !(d.PT | d.PL) && !(off & 3) && (k < Klim) &&
-                        (m + 8 <= Mlim) && (kh < d.KH) &&
-                        ((unsigned)ih < (unsigned)d.H);
-      if (fast) {
-        // two 8-B loads: off is only guaranteed 4-elem aligned
-        *(unsigned long long*)v =
-            *(const unsigned long long*)(x + off);
-        *((unsigned long long*)v + 1) =
-            *((const unsigned long long*)(x + off) + 1);
-        return;
-      }
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const int mj = m + j;
-        int cj = mj % d.C, tj = mj / d.C;
-        int kwj = tj % d.KW, khj = tj / d.KW;
-        int ihj = oh * d.SY - d.PT + khj;
-        int iwj = ow * d.SX - d.PL + kwj;
-        const bool okj = (k < Klim) & (mj < Mlim) & (khj < d.KH) &
-                         ((unsigned)ihj < (unsigned)d.H) &
-                         ((unsigned)iwj < (unsigned)d.W);
-        v[j] = okj ? x[(((long)n * d.H + ihj) * d.W + iwj) * d.C + cj]
-                   : (unsigned short)0;
-      }
-      return;
-    }
     const bool ok = (k < Klim) & (m < Mlim) & (kh < d.KH) &
                     ((unsigned)ih < (unsigned)d.H) &
                     ((unsigned)iw < (unsigned)d.W);
@@ -589,6 +586,50 @@ struct Im2colTNv {
     *(short8*)v = *(const short8*)p;
 #pragma unroll
     for (int j = 0; j < 8; ++j) v[j] = ok ? v[j] : (unsigned short)0;
+  }
+};
+
+// packed C%4 variant (RGB padded to 4): a SEPARATE type so the
+// fallback machinery costs the C%8 wgrad instantiations nothing
+struct Im2colTNvP {
+  const unsigned short* x;
+  ConvDesc d;
+  int Klim, Mlim;
+  __device__ void load16(int k, int mo, unsigned short* v) const {
+    const int m = mo * 8;
+    int c = m % d.C, t = m / d.C;
+    int kh = t / d.KW;
+    int ow = k % d.OW;
+    int t2 = k / d.OW;
+    int oh = t2 % d.OH, n = t2 / d.OH;
+    int ih = oh * d.SY - d.PT + kh;
+    int iw = ow * d.SX - d.PL + (m % (d.KW * d.C)) / d.C;
+    const int rowlen = d.KW * d.C;
+    long off = (((long)n * d.H + ih) * d.W + iw) * d.C + c;
+    const bool fast = ((m % rowlen) + 8 <= rowlen) &&
+                      !(d.PT | d.PL) && !(off & 3) && (k < Klim) &&
+                      (m + 8 <= Mlim) && (kh < d.KH) &&
+                      ((unsigned)ih < (unsigned)d.H);
+    if (fast) {
+      // two 8-B loads: off is only guaranteed 4-elem aligned
+      *(unsigned long long*)v = *(const unsigned long long*)(x + off);
+      *((unsigned long long*)v + 1) =
+          *((const unsigned long long*)(x + off) + 1);
+      return;
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int mj = m + j;
+      int cj = mj % d.C, tj = mj / d.C;
+      int kwj = tj % d.KW, khj = tj / d.KW;
+      int ihj = oh * d.SY - d.PT + khj;
+      int iwj = ow * d.SX - d.PL + kwj;
+      const bool okj = (k < Klim) & (mj < Mlim) & (khj < d.KH) &
+                       ((unsigned)ihj < (unsigned)d.H) &
+                       ((unsigned)iwj < (unsigned)d.W);
+      v[j] = okj ? x[(((long)n * d.H + ihj) * d.W + iwj) * d.C + cj]
+                 : (unsigned short)0;
+    }
   }
 };
 
@@ -1061,6 +1102,16 @@ void conv_fwd_gemm(const unsigned short* x, const unsigned short* w,
     conv_fwd_gemm8(x, w, out, out_f32, bias, zero, d, act, s);
     return;
   }
+  if (d.C & 7) {
+    Im2colNTP la{x, zero, d};
+    DenseNT lb{w, zero, K, K};
+    if (out_f32)
+      launch_gemm_nt(la, lb, (float*)out, bias, M, d.K, K, act, s);
+    else
+      launch_gemm_nt(la, lb, (unsigned short*)out, bias, M, d.K, K, act,
+                     s);
+    return;
+  }
   Im2colNT la{x, zero, d};
   DenseNT lb{w, zero, K, K};
   if (out_f32)
@@ -1097,9 +1148,14 @@ bool conv_wgrad_gemm(const unsigned short* eo, const unsigned short* x,
   const int Kdim = d.N * d.OH * d.OW;
   const int M = d.K;                   // n_kernels
   const int N = d.KH * d.KW * d.C;
-  if ((M & 7) == 0 && (d.C & 3) == 0) {
+  if ((M & 7) == 0 && (d.C & 7) == 0) {
     DenseTNv la{eo, d.K, Kdim, d.K};
     Im2colTNv lb{x, d, Kdim, N};
+    return launch_gemm_tn2(la, lb, gw, M, N, Kdim, beta, s, bias_out);
+  }
+  if ((M & 7) == 0 && (d.C & 3) == 0) {
+    DenseTNv la{eo, d.K, Kdim, d.K};
+    Im2colTNvP lb{x, d, Kdim, N};
     return launch_gemm_tn2(la, lb, gw, M, N, Kdim, beta, s, bias_out);
   }
   DenseTN la{eo, d.K, Kdim, d.K};

@@ -6,6 +6,7 @@
 
 // dense row-major [rows][ld], K-innermost; k past Klim reads the zero page
 struct DenseNT {
+  static constexpr bool MAY_SLOW = false;   // addr() never nullptr
   const unsigned short* base;
   const unsigned short* zero;
   long ld;
@@ -22,14 +23,10 @@ struct DenseNT {
 };
 
 // implicit im2col over NHWC input: row=(n,oh,ow), kdim=(kh,kw,c) c-inner.
-// C % 8 == 0: every 16-B chunk stays inside one pixel's C-run.
-// C % 8 != 0 (e.g. RGB padded only to 4): a chunk spans several (kw,c)
-// pairs, which ARE contiguous in NHWC within one kh row when there is
-// no horizontal padding; addr() returns nullptr for the chunks where
-// the fast DMA is illegal (kh-row crossing, padding in play, or a
-// misaligned 16-B source) and the staging loop falls back to the
-// per-element path for just those chunks.
+// Requires C % 8 == 0 so each 16-B chunk stays inside one pixel's C-run
+// (Im2colNTP below handles C%4).
 struct Im2colNT {
+  static constexpr bool MAY_SLOW = false;
   const unsigned short* x;
   const unsigned short* zero;
   ConvDesc d;
@@ -42,18 +39,6 @@ struct Im2colNT {
     int oh = t % d.OH, n = t / d.OH;
     int ih = oh * d.SY - d.PT + kh;
     int iw = ow * d.SX - d.PL + kw;
-    if (d.C & 7) {
-      const int rowlen = d.KW * d.C;
-      if ((k % rowlen) + 8 > rowlen || (d.PT | d.PL))
-        return nullptr;                  // kh-cross or padded: slow path
-      if (kh >= d.KH || (unsigned)ih >= (unsigned)d.H)
-        return zero;
-      long off = (((long)n * d.H + ih) * d.W + iw) * d.C + c;
-      // 8-B source alignment suffices for the 16-B glds (probed:
-      // tools/probe_glds_align.hip); 4-elem-multiple offsets are
-      // guaranteed by C%4==0
-      return (off & 3) ? nullptr : (const void*)(x + off);
-    }
     if (kh >= d.KH || (unsigned)ih >= (unsigned)d.H ||
         (unsigned)iw >= (unsigned)d.W)
       return zero;
@@ -75,9 +60,54 @@ struct Im2colNT {
   }
 };
 
+// Packed im2col for C % 4 == 0 (RGB padded to 4, not 8): a 16-B chunk
+// spans several contiguous (kw,c) pairs of one kh row.  addr() returns
+// nullptr for chunks where the straight DMA is illegal — kh-row
+// crossing, horizontal padding in play, or a non-8-B-aligned source
+// (8-B-aligned glds-16 is legal on gfx950: tools/probe_glds_align.hip)
+// — and the staging loop per-element-loads just those.  A separate
+// TYPE so the fallback machinery costs the C%8 kernels nothing.
+struct Im2colNTP {
+  static constexpr bool MAY_SLOW = true;
+  const unsigned short* x;
+  const unsigned short* zero;
+  ConvDesc d;
+  __device__ const void* addr(int row, int k8) const {
+    int k = k8 * 8;
+    int c = k % d.C, t = k / d.C;
+    int kh = t / d.KW;
+    int ow = row % d.OW;
+    t = row / d.OW;
+    int oh = t % d.OH, n = t / d.OH;
+    int ih = oh * d.SY - d.PT + kh;
+    int iw = ow * d.SX - d.PL + (k % (d.KW * d.C)) / d.C;
+    const int rowlen = d.KW * d.C;
+    if ((k % rowlen) + 8 > rowlen || (d.PT | d.PL))
+      return nullptr;                  // kh-cross or padded: slow path
+    if (kh >= d.KH || (unsigned)ih >= (unsigned)d.H)
+      return zero;
+    long off = (((long)n * d.H + ih) * d.W + iw) * d.C + c;
+    return (off & 3) ? nullptr : (const void*)(x + off);
+  }
+  __device__ unsigned short elem(int row, int k) const {
+    int c = k % d.C, t = k / d.C;
+    int kw = t % d.KW, kh = t / d.KW;
+    int ow = row % d.OW;
+    t = row / d.OW;
+    int oh = t % d.OH, n = t / d.OH;
+    int ih = oh * d.SY - d.PT + kh;
+    int iw = ow * d.SX - d.PL + kw;
+    if (kh >= d.KH || (unsigned)ih >= (unsigned)d.H ||
+        (unsigned)iw >= (unsigned)d.W)
+      return 0;
+    return x[(((long)n * d.H + ih) * d.W + iw) * d.C + c];
+  }
+};
+
 // conv dgrad gather over NHWC err_output: row=(n,ih,iw) of the FORWARD
 // input, kdim=(kh,kw,co) co-inner; strided positions read the zero page.
 struct DgradNT {
+  static constexpr bool MAY_SLOW = false;
   const unsigned short* eo;
   const unsigned short* zero;
   ConvDesc d;
@@ -115,6 +145,7 @@ struct DgradNT {
 // stride-1 dgrad specialization: SY == SX == 1 removes the per-address
 // stride divisibility checks and divisions of the generic DgradNT.
 struct DgradNT1 {
+  static constexpr bool MAY_SLOW = false;
   const unsigned short* eo;
   const unsigned short* zero;
   ConvDesc d;
