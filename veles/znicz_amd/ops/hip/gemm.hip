@@ -119,9 +119,13 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_k(
               16, 0, 0);
         } else {
           unsigned short va[8];
+          if constexpr (LA::MAY_SLOW)
+            la.slowload(min(m0 + row, M - 1), kc0, va);
+          else {
 #pragma unroll
-          for (int j = 0; j < 8; ++j)
-            va[j] = la.elem(min(m0 + row, M - 1), kc0 + j);
+            for (int j = 0; j < 8; ++j)
+              va[j] = la.elem(min(m0 + row, M - 1), kc0 + j);
+          }
           *(short8*)(As + chunk * 1024 + lane * 16) = *(short8*)va;
         }
         if (sb) {
@@ -131,9 +135,13 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_k(
               16, 0, 0);
         } else {
           unsigned short vb[8];
+          if constexpr (LB::MAY_SLOW)
+            lb.slowload(min(n0 + row, N - 1), kc0, vb);
+          else {
 #pragma unroll
-          for (int j = 0; j < 8; ++j)
-            vb[j] = lb.elem(min(n0 + row, N - 1), kc0 + j);
+            for (int j = 0; j < 8; ++j)
+              vb[j] = lb.elem(min(n0 + row, N - 1), kc0 + j);
+          }
           *(short8*)(Bs + chunk * 1024 + lane * 16) = *(short8*)vb;
         }
       }
@@ -615,6 +623,21 @@ struct Im2colTNvP {
       *(unsigned long long*)v = *(const unsigned long long*)(x + off);
       *((unsigned long long*)v + 1) =
           *((const unsigned long long*)(x + off) + 1);
+      return;
+    }
+    const int pos = m % rowlen;
+    if (!(d.PT | d.PL) && rowlen - pos == 4 && !(d.C & 3) &&
+        (k < Klim) && (m + 8 <= Mlim)) {
+      // clean kh crossing: two contiguous 8-B-aligned runs (4+4)
+      unsigned long long r1 = 0, r2 = 0;
+      if (kh < d.KH && (unsigned)ih < (unsigned)d.H)
+        r1 = *(const unsigned long long*)(x + off);
+      if (kh + 1 < d.KH && (unsigned)(ih + 1) < (unsigned)d.H)
+        r2 = *(const unsigned long long*)(
+            x + (((long)n * d.H + ih + 1) * d.W + ow * d.SX - d.PL) *
+                    d.C);
+      *(unsigned long long*)v = r1;
+      *((unsigned long long*)v + 1) = r2;
       return;
     }
 #pragma unroll

@@ -102,6 +102,35 @@ struct Im2colNTP {
       return 0;
     return x[(((long)n * d.H + ih) * d.W + iw) * d.C + c];
   }
+
+  // refused-chunk path: a kh-crossing octet is exactly two contiguous
+  // 8-B-aligned runs (C=4: 4+4 elements) — two vector loads instead of
+  // eight guarded scalars (per-element vmcnt serialization measured
+  // ~10-30× a fast chunk).  Padded convs take the scalar path.
+  __device__ void slowload(int row, int kc0, unsigned short* v) const {
+    const int rowlen = d.KW * d.C;
+    const int pos = kc0 % rowlen;
+    if ((d.PT | d.PL) || rowlen - pos != 4 || (d.C & 3)) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) v[j] = elem(row, kc0 + j);
+      return;
+    }
+    const int kh = kc0 / rowlen;         // run 1: last 4 elems of row kh
+    const int ow = row % d.OW;
+    const int t = row / d.OW;
+    const int oh = t % d.OH, n = t / d.OH;
+    const int ih = oh * d.SY + kh;       // PT == 0 here
+    const int iw = ow * d.SX + (pos / d.C);
+    unsigned long long r1 = 0, r2 = 0;
+    if (kh < d.KH && ih < d.H)
+      r1 = *(const unsigned long long*)(
+          x + (((long)n * d.H + ih) * d.W + iw) * d.C + (pos % d.C));
+    if (kh + 1 < d.KH && ih + 1 < d.H)
+      r2 = *(const unsigned long long*)(
+          x + (((long)n * d.H + ih + 1) * d.W + ow * d.SX) * d.C);
+    *(unsigned long long*)v = r1;
+    *((unsigned long long*)v + 1) = r2;
+  }
 };
 
 // conv dgrad gather over NHWC err_output: row=(n,ih,iw) of the FORWARD
