@@ -4,10 +4,11 @@ Parity target: the reference's AlexNet-class imagenet sample
 (znicz/samples/ imagenet workflow): conv/LRN/pool stack + 2×(FC 4096 +
 dropout) + softmax 1000, trained with the StandardWorkflow graph.
 
-MI355X specifics: NHWC activations with the RGB input padded 3→4
-channels (the C=4 im2col staging packs contiguous (kw,c) pairs into the
-16-B DMA chunks — 8/3× fewer conv1 FLOPs than padding to 8), strict-ReLU
-activations, bf16
+MI355X specifics: NHWC activations with the RGB input padded 3→8
+channels (C=4 packed staging exists and is tested, but measured ~3×
+lower GEMM efficiency on conv1 — profiles/perf_r1_session2.md — so the
+2.67× FLOP saving nets a loss; C%8 keeps every 16-B DMA chunk
+pixel-aligned), strict-ReLU activations, bf16
 compute / fp32 accumulate.  Synthetic 227×227 data (BASELINE: no
 network, random-init weights).
 """
@@ -18,7 +19,7 @@ from veles.znicz_amd.standard_workflow import StandardWorkflow
 
 root.alexnet.update({
     "loader": {"minibatch_size": 256, "n_train": 1024, "n_valid": 0,
-               "sample_shape": (227, 227, 4), "n_classes": 1000,
+               "sample_shape": (227, 227, 8), "n_classes": 1000,
                "on_device": True, "device_dtype": "bf16"},
     "decision": {"max_epochs": 1, "fail_iterations": 1000000},
     "snapshotter": {"enabled": False},
