@@ -141,3 +141,78 @@ def test_video_ae():
     assert wf.is_finished
     hist = wf.decision.mse_history
     assert hist[-1] < 0.85 * hist[0], hist
+
+
+@pytest.mark.timeout(600)
+def test_mnist_simple():
+    """Reference znicz/samples/MnistSimple — minimal 1-hidden MLP."""
+    prng.seed_all(11)
+    from veles.znicz_amd.models.mnist_simple import MnistSimpleWorkflow
+    wf = MnistSimpleWorkflow(DummyLauncher())
+    wf.initialize(device=NumpyDevice())
+    wf.run()
+    assert wf.is_finished
+    assert wf.decision.epoch_n_err_pt[TRAIN] < 40.0, \
+        wf.decision.epoch_n_err_pt
+
+
+@pytest.mark.timeout(600)
+def test_yale_faces():
+    """Reference znicz/samples/YaleFaces — illumination-invariant face
+    id; the conv net must beat chance (90%) clearly despite the
+    per-sample lighting scale."""
+    prng.seed_all(12)
+    from veles.znicz_amd.models.yale_faces import YaleFacesWorkflow
+    from veles.config import root
+    root.yale_faces.decision.update({"max_epochs": 8})
+    root.yale_faces.loader.update({"n_train": 200, "n_valid": 40})
+    wf = YaleFacesWorkflow(DummyLauncher())
+    wf.initialize(device=NumpyDevice())
+    wf.run()
+    assert wf.is_finished
+    assert wf.decision.epoch_n_err_pt[TRAIN] < 50.0, \
+        wf.decision.epoch_n_err_pt
+
+
+@pytest.mark.timeout(600)
+def test_spam_kohonen():
+    """Reference znicz/samples/SpamKohonen — the SOM must separate the
+    two word-distribution topics: most neurons end up class-pure."""
+    import numpy
+    prng.seed_all(13)
+    from veles.znicz_amd.models.spam_kohonen import SpamKohonenWorkflow
+    wf = SpamKohonenWorkflow(DummyLauncher())
+    wf.initialize(device=NumpyDevice())
+    wf.run()
+    assert wf.is_finished
+    # winner purity: map each sample to its BMU, check dominant-class
+    # fraction per neuron
+    ld, fwd = wf.loader, wf.forward
+    data = ld.original_data.mem
+    labels = ld.original_labels.mem
+    fwd.weights.map_read()
+    w = fwd.weights.mem                      # [n_neurons, features]
+    d2 = ((data[:, None, :] - w[None, :, :]) ** 2).sum(axis=2)
+    bmu = d2.argmin(axis=1)
+    pure = total = 0
+    for n in set(bmu):
+        cls = labels[bmu == n]
+        pure += max((cls == 0).sum(), (cls == 1).sum())
+        total += len(cls)
+    assert pure / total > 0.85, pure / total
+
+
+@pytest.mark.timeout(600)
+def test_imagenet_ae():
+    """Reference znicz/samples/ImagenetAE — conv AE on natural-image
+    statistics; reconstruction must improve."""
+    prng.seed_all(14)
+    from veles.znicz_amd.models.imagenet_ae import ImagenetAEWorkflow
+    from veles.config import root
+    root.imagenet_ae.decision.update({"max_epochs": 4})
+    wf = ImagenetAEWorkflow(DummyLauncher())
+    wf.initialize(device=NumpyDevice())
+    wf.run()
+    assert wf.is_finished
+    h = wf.decision.mse_history
+    assert h[-1] < 0.9 * h[0], h

@@ -36,12 +36,15 @@ root.mnist_ae.update({
 class MnistAEWorkflow(NNWorkflow):
     """loader → conv → maxpool → depool → deconv → EvaluatorMSE."""
 
-    def __init__(self, workflow=None, **kwargs):
+    def __init__(self, workflow=None, cfg=None, loader_factory=None,
+                 **kwargs):
         super(MnistAEWorkflow, self).__init__(workflow, **kwargs)
-        cfg = root.mnist_ae
+        cfg = cfg or root.mnist_ae
+        loader_factory = loader_factory or (
+            lambda wf, **kw: SyntheticAELoader(wf, **kw))
         self.repeater.link_from(self.start_point)
 
-        self.loader = SyntheticAELoader(self, **cfg.loader.to_dict())
+        self.loader = loader_factory(self, **cfg.loader.to_dict())
         self.loader.link_from(self.repeater)
 
         k, kx = cfg.n_kernels, cfg.kx
