@@ -408,3 +408,20 @@ def test_permute_kc_matches_torch():
         want = src.permute(2, 1, 0).contiguous()
         torch.cuda.synchronize()
         assert torch.equal(dst, want), (K, P, C)
+
+
+def test_yale_faces_trains_on_gpu():
+    """New-sample smoke on the HIP path (conv stack + illumination
+    nuisance)."""
+    from veles import prng
+    from veles.znicz_amd.models.yale_faces import YaleFacesWorkflow
+    from veles.loader.base import TRAIN
+    prng.seed_all(31)
+    root.yale_faces.decision.update({"max_epochs": 6})
+    root.yale_faces.loader.update({"n_train": 200, "n_valid": 40})
+    wf = YaleFacesWorkflow(DummyLauncher())
+    wf.initialize(device=ROCmDevice(0))
+    wf.run()
+    assert wf.is_finished
+    assert wf.decision.epoch_n_err_pt[TRAIN] < 60.0, \
+        wf.decision.epoch_n_err_pt
