@@ -99,15 +99,46 @@ __global__ void sgd_k(float* __restrict__ w, const float* __restrict__ g,
     sgd_elem(w, g, vel, wb, p, lr, decay, l1, moment, scale);
 }
 
-// one launch for a whole gradient bucket: block -> tensor via b2t
+// one launch for a whole gradient bucket: block -> tensor via b2t.
+// Each thread owns 8 CONSECUTIVE elements (2× float4 per stream) — the
+// strided-single-float version measured 86% memory-wait at 4.6 TB/s.
 __global__ void sgd_multi_k(const SgdDesc* __restrict__ ds,
                             const int* __restrict__ b2t) {
   const SgdDesc d = ds[b2t[blockIdx.x]];
   const long base = ((long)blockIdx.x - d.blk0) * SGD_MULTI_SPAN;
+  const long p0 = base + (long)threadIdx.x * 8;
+  if (p0 + 8 <= d.n) {
+    f32x4 w[2], g[2], vel[2];
+    w[0] = *(const f32x4*)(d.w + p0);
+    w[1] = *(const f32x4*)(d.w + p0 + 4);
+    g[0] = *(const f32x4*)(d.g + p0);
+    g[1] = *(const f32x4*)(d.g + p0 + 4);
+    vel[0] = *(const f32x4*)(d.vel + p0);
+    vel[1] = *(const f32x4*)(d.vel + p0 + 4);
+    unsigned short wb8[8];
 #pragma unroll
-  for (int i = 0; i < SGD_MULTI_SPAN / 256; ++i) {
-    const long p = base + i * 256 + threadIdx.x;
-    if (p < d.n)
+    for (int h = 0; h < 2; ++h)
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        float wi = w[h][j];
+        float l1l2 = d.l1 * (wi > 0.f ? 1.f : (wi < 0.f ? -1.f : 0.f)) +
+                     (1.f - d.l1) * wi;
+        float gi = g[h][j] * d.scale + d.decay * l1l2;
+        float v = d.moment * vel[h][j] + d.lr * gi;
+        vel[h][j] = v;
+        wi -= v;
+        w[h][j] = wi;
+        wb8[h * 4 + j] = f2bf(wi);
+      }
+    *(f32x4*)(d.w + p0) = w[0];
+    *(f32x4*)(d.w + p0 + 4) = w[1];
+    if (d.moment != 0.f) {
+      *(f32x4*)(d.vel + p0) = vel[0];
+      *(f32x4*)(d.vel + p0 + 4) = vel[1];
+    }
+    if (d.wb) *(short8*)(d.wb + p0) = *(short8*)wb8;
+  } else {
+    for (long p = p0; p < d.n; ++p)
       sgd_elem(d.w, d.g, d.vel, d.wb, p, d.lr, d.decay, d.l1, d.moment,
                d.scale);
   }
