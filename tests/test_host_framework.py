@@ -47,6 +47,10 @@ def test_cli_config_file(tmp_path):
          "veles/znicz_amd/models/wine.py", str(cfg), "-b", "numpy"],
         cwd=REPO, capture_output=True, text=True, timeout=600)
     assert out.returncode == 0, out.stderr[-2000:]
+    # the CLI must not run silently: epoch accounting reaches the user
+    # (a missing setup_logging() once swallowed every self.info line)
+    assert "epoch 1" in out.stderr or "epoch 1" in out.stdout, \
+        (out.stdout[-500:], out.stderr[-500:])
 
 
 def test_genetics_ga():

@@ -130,6 +130,11 @@ class Main(object):
 
 def main(argv=None):
     args = parse_args(argv if argv is not None else sys.argv[1:])
+    # without this the units' self.info() epoch lines go to an
+    # unconfigured logger and the CLI runs in total silence
+    import logging
+    from veles.logger import setup_logging
+    setup_logging(logging.DEBUG if args.verbose else logging.INFO)
     if args.gpus > 1 and "WORLD_SIZE" not in os.environ:
         spawn_ranks(args)                      # does not return
     if args.random_seed is not None:
