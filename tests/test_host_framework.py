@@ -236,3 +236,23 @@ def test_graph_stepper_cpu_fallback():
         st.step()
     w1 = wf.forwards[0].weights.mem
     assert not numpy.allclose(w0, w1)   # training progressed
+
+
+def test_launcher_rejects_cpu_multirank():
+    """Multi-rank without the GPU reducer would silently train divergent
+    replicas — boot must refuse instead."""
+    from veles.config import root
+    from veles.launcher import Launcher
+    from veles import prng
+    prng.seed_all(9)
+    from veles.znicz_amd.models.wine import WineWorkflow
+    root.wine.decision.update({"max_epochs": 1})
+    lch = Launcher(backend="numpy")
+    wf = WineWorkflow(lch)
+    assert lch.workflow is wf
+
+    class _Stub(object):
+        rank, world = 0, 2
+    lch.dist = _Stub()
+    with pytest.raises(RuntimeError, match="single-process"):
+        lch.boot()

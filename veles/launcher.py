@@ -85,6 +85,19 @@ class Launcher(Logger):
             if wf.loader is not None:
                 wf.loader.shard(self.dist.rank, self.dist.world)
         wf.initialize(device=device, **kwargs)
+        if self.dist is not None:
+            # WITHOUT the reducer each rank applies only its LOCAL
+            # gradients and the replicas silently diverge — gradient
+            # all-reduce is the reducer's job (GD units submit to it,
+            # the workflow's CommSync joins the comm stream each loop)
+            if getattr(device, "is_gpu", False):
+                from veles.znicz_amd.parallel import GradientReducer
+                wf.grad_reducer = GradientReducer(self.dist, device)
+            else:
+                raise RuntimeError(
+                    "multi-rank training needs the GPU gradient "
+                    "reducer (RCCL over xGMI); the numpy backend is "
+                    "single-process only — drop --gpus or use -b rocm")
         try:
             wf.run()
         except BaseException as e:

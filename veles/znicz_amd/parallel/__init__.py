@@ -45,7 +45,8 @@ class DistributedContext(object):
         world = int(os.environ.get("WORLD_SIZE", 1))
         local_rank = int(os.environ.get("LOCAL_RANK", rank))
         if backend is None:
-            backend = "nccl" if torch.cuda.is_available() else "gloo"
+            backend = os.environ.get("VELES_DIST_BACKEND") or (
+                "nccl" if torch.cuda.is_available() else "gloo")
         if not dist.is_initialized():
             os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
             os.environ.setdefault("MASTER_PORT", "29500")
