@@ -84,6 +84,8 @@ class ROCmDevice(Device):
             raise RuntimeError("ROCmDevice requested but no HIP GPU visible")
         if index is None:
             index = int(os.environ.get("LOCAL_RANK", 0))
+            # test lane: multiple ranks may share one GPU (gloo fallback)
+            index %= max(1, torch.cuda.device_count())
         self.index = index
         torch.cuda.set_device(index)
         self.torch_device = torch.device("cuda", index)
