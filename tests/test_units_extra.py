@@ -546,3 +546,29 @@ def test_accumulate_gradient_fc_and_conv():
                                   rtol=1e-5)
     numpy.testing.assert_allclose(gd.gradient_bias.mem, 2 * b1,
                                   rtol=1e-5)
+
+
+def test_decision_dp_percentage_denominator():
+    """Summed n_err across ranks must divide by the GLOBAL class length
+    (per-rank shard length × world), not the shard length."""
+    from veles.dummy import DummyWorkflow
+    from veles.memory import Array
+    from veles.znicz_amd.decision import DecisionGD
+
+    class _Ctx(object):
+        world = 2
+
+        def all_reduce_scalar(self, v):
+            return v * 2          # both ranks saw the same count
+
+    wf = DummyWorkflow()
+    wf.dist = _Ctx()
+    dec = DecisionGD(wf)
+    dec.class_lengths = [0, 0, 256]   # per-rank shard
+    dec.minibatch_n_err = Array(numpy.array([64, 0], numpy.int32))
+    dec.epoch_number = 0
+    dec.min_train_n_err = None
+    from veles.loader.base import TRAIN
+    dec.on_last_minibatch(TRAIN)
+    assert dec.epoch_n_err[TRAIN] == 128
+    assert abs(dec.epoch_n_err_pt[TRAIN] - 25.0) < 1e-6

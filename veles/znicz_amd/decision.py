@@ -70,6 +70,10 @@ class DecisionBase(Unit):
             return value
         return ctx.all_reduce_scalar(value)
 
+    def _world(self):
+        ctx = getattr(self.workflow, "dist", None)
+        return ctx.world if ctx is not None else 1
+
 
 class DecisionGD(DecisionBase):
     """Classification decision driven by evaluator n_err counts."""
@@ -89,7 +93,9 @@ class DecisionGD(DecisionBase):
         arr = self.minibatch_n_err
         arr.map_read()
         n_err = self._dist_sum(int(arr.mem[0]))
-        length = max(1, int(self.class_lengths[clazz]))
+        # class_lengths are PER-RANK after sharding; the summed error
+        # count is global, so the denominator scales by world
+        length = max(1, int(self.class_lengths[clazz]) * self._world())
         self.epoch_n_err[clazz] = n_err
         self.epoch_n_err_pt[clazz] = 100.0 * n_err / length
         # reset the device counter for the next set
@@ -146,7 +152,7 @@ class DecisionMSE(DecisionBase):
         arr = self.minibatch_metrics
         arr.map_read()
         total = self._dist_sum(float(arr.mem[0]))
-        length = max(1, int(self.class_lengths[clazz]))
+        length = max(1, int(self.class_lengths[clazz]) * self._world())
         mse = total / length
         self.epoch_metrics[clazz] = (mse, float(arr.mem[1]),
                                      float(arr.mem[2]))
