@@ -22,8 +22,8 @@ def parse_args():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int,
                    default=int(os.environ.get("WORLD_SIZE", 1)))
-    p.add_argument("--steps", type=int, default=30)
-    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--steps", type=int, default=100)
+    p.add_argument("--warmup", type=int, default=20)
     p.add_argument("--model", default="alexnet",
                    choices=["alexnet", "cifar", "vgg16"])
     p.add_argument("--batch", type=int, default=0,
@@ -103,12 +103,11 @@ def main():
         wf.dist = dist_ctx
     wf.initialize(device=device)
 
-    if dist_ctx is not None or True:
-        # bucketed all-reduce + comm-stream SGD (also used at world=1 so
-        # the update overlaps the rest of backward)
-        from veles.znicz_amd.parallel import GradientReducer
-        wf.grad_reducer = GradientReducer(dist_ctx or _NullCtx(world),
-                                          device)
+    # the reducer is ALWAYS attached: at world>1 it all-reduces gradient
+    # buckets over RCCL; at world=1 (_NullCtx) it still runs the
+    # comm-stream fused SGD so the update overlaps the rest of backward
+    from veles.znicz_amd.parallel import GradientReducer
+    wf.grad_reducer = GradientReducer(dist_ctx or _NullCtx(world), device)
 
     loader, fwds, ev, dec = wf.loader, wf.forwards, wf.evaluator, \
         wf.decision

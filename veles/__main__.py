@@ -97,6 +97,9 @@ class Main(object):
         if self.args.snapshot:
             self.workflow = SnapshotterToFile.import_(self.args.snapshot)
             self.workflow.workflow = self.launcher
+            # the workflow property setter does not register with the
+            # launcher; without this boot() finds launcher.workflow None
+            self.launcher.add_ref(self.workflow)
         else:
             self.workflow = workflow_class(self.launcher, **kwargs)
         return self.workflow, bool(self.args.snapshot)

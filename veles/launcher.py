@@ -120,6 +120,10 @@ class Launcher(Logger):
                 wf, prefix="emergency_r%d" % self.rank, compression="gz")
             snap.suffix = "abort"
             _os.makedirs(snap.directory, exist_ok=True)
+            # detach before exporting so the emergency snapshotter unit
+            # itself is not baked into the pickled workflow (it would be
+            # re-added on every resume/failure cycle)
+            wf.del_ref(snap)
             path = snap.export()
             self.warning("emergency snapshot after failure (%s): %s",
                          reason, path)

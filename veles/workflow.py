@@ -11,10 +11,24 @@ Gate semantics (``gate_block`` swallows, ``gate_skip`` forwards) and the
 Repeater-loop construct are preserved exactly.
 """
 
+import inspect
 from collections import deque
 
 from veles.units import Unit, TrivialUnit, Container
 from veles.mutable import Bool
+
+
+def _init_signature(unit):
+    """Signature of unit.initialize, or None if it takes **kwargs (or
+    cannot be inspected) — None means 'pass everything through'."""
+    try:
+        sig = inspect.signature(unit.initialize)
+    except (TypeError, ValueError):
+        return None
+    for p in sig.parameters.values():
+        if p.kind is inspect.Parameter.VAR_KEYWORD:
+            return None
+    return sig
 
 
 class StartPoint(TrivialUnit):
@@ -44,8 +58,12 @@ class Repeater(TrivialUnit):
         super(Repeater, self).__init__(workflow, **kwargs)
 
     def open_gate(self, src):
-        # A repeater fires whenever ANY input fires (loop-back semantics),
-        # matching the reference behavior for loop hubs.
+        # A repeater fires whenever ANY input fires (loop-back
+        # semantics), matching the reference behavior for loop hubs.
+        # Note: this resets EVERY input flag on each fire, which
+        # diverges from per-edge gate accounting when a Repeater has
+        # more than one non-loop input — by design: a Repeater is a
+        # loop hub, not a join point (use a plain Unit to join).
         if src in self.links_from:
             for u in self.links_from:
                 self.links_from[u] = False
@@ -119,12 +137,14 @@ class Workflow(Container):
             for u in pending:
                 kw = dict(kwargs)
                 kw["device"] = device
-                try:
-                    result = u.initialize(**kw)
-                except TypeError as e:
-                    if "unexpected keyword argument" not in str(e):
-                        raise
-                    result = u.initialize()
+                # Pass only the kwargs the unit's initialize() accepts
+                # (signature-inspected, so a unit's own genuine
+                # TypeError is never swallowed).
+                sig = _init_signature(u)
+                if sig is not None:
+                    kw = {k: v for k, v in kw.items()
+                          if k in sig.parameters}
+                result = u.initialize(**kw)
                 if result is False:
                     deferred.append(u)
             if not deferred:

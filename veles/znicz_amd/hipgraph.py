@@ -138,6 +138,13 @@ class GraphStepper(object):
             self._graph = g
             self._sig = self._signature()
             self._touched = tuple(touched)
+            # stream capture RECORDS the kernels without executing
+            # them — replay once immediately so the capture iteration's
+            # minibatch still trains (otherwise its weight update and
+            # n_err contribution are silently dropped, repeated on
+            # every re-capture after an lr change)
+            g.replay()
+            self._post_replay()
         except Exception as e:
             # capture is an optimization; any failure (a unit that syncs,
             # an unsupported collective) falls back to eager for good

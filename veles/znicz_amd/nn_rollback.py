@@ -40,9 +40,14 @@ class NNRollback(Unit):
         for arr, saved in self._best.values():
             arr.map_write()
             arr.mem[...] = saved
-            if hasattr(arr, "bf16") and arr.bf16 is not None:
+            # push the restored host copy to the fp32 master on device,
+            # THEN rebuild whichever bf16 working copies this array
+            # carries (fc: bf16+bf16_t, conv: bf16+bf16_dgrad) — without
+            # the unmap the kernels keep computing on the old weights
+            arr.unmap()
+            if getattr(arr, "bf16", None) is not None:
                 from veles.znicz_amd import ops
-                ops.refresh_working_copies(arr)
+                ops.refresh_all_copies(arr)
         for gd in self.gds:
             gd.learning_rate *= self.lr_cut_ratio
             gd.learning_rate_bias *= self.lr_cut_ratio

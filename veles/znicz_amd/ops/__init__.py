@@ -87,7 +87,22 @@ def refresh_working_copies(weights_arr):
     if getattr(weights_arr, "bf16", None) is not None:
         import torch
         weights_arr.bf16.copy_(w.to(torch.bfloat16))
-        weights_arr.bf16_t.copy_(w.t().to(torch.bfloat16))
+        if getattr(weights_arr, "bf16_t", None) is not None:
+            weights_arr.bf16_t.copy_(w.t().to(torch.bfloat16))
+
+
+def refresh_all_copies(arr):
+    """Rebuild every bf16 working copy an Array carries from its fp32
+    master ``devmem`` — fc arrays have ``bf16``+``bf16_t``, conv weight
+    arrays ``bf16``+``bf16_dgrad``, biases sometimes just ``bf16``.
+    Used after host-side weight edits (NNRollback, snapshot surgery)."""
+    if arr.devmem is None or getattr(arr, "bf16", None) is None:
+        return
+    arr.bf16.copy_(arr.devmem.reshape(arr.bf16.shape).to(arr.bf16.dtype))
+    if getattr(arr, "bf16_t", None) is not None:
+        refresh_fc_copies(arr)
+    if getattr(arr, "bf16_dgrad", None) is not None:
+        refresh_conv_copies(arr)
 
 
 def labels_on_device(labels_arr, device):

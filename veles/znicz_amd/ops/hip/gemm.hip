@@ -21,7 +21,9 @@
 //
 // Wave64 / LDS-tiled throughout; nothing here is a CUDA translation.
 #include <cstdlib>
+#include <mutex>
 #include <stdexcept>
+#include <unordered_map>
 #include "common.h"
 
 #define BM 128
@@ -1050,16 +1052,28 @@ void gemm_nt_dense(const unsigned short* A, const unsigned short* B,
   const int nk = cdiv(K, BK);
   const int splits = max(1, min(nk, 512 / tiles));
   if (splits > 1) {
-    // fp32 workspace + atomic partials, then one bias/act/cast pass
-    static float* ws = nullptr;
-    static size_t cap = 0;
+    // fp32 workspace + atomic partials, then one bias/act/cast pass.
+    // The workspace is PER-STREAM (map below): the comm-stream design
+    // allows two streams to hit FC split-K concurrently, and a shared
+    // buffer would race between their atomics and finisher passes.
+    static std::mutex ws_mu;
+    static std::unordered_map<hipStream_t, std::pair<float*, size_t>>
+        ws_map;
+    float* ws;
     size_t need = (size_t)cdiv(M, BM) * BM * cdiv(N, BN) * BN *
                   sizeof(float);
-    if (need > cap) {
-      if (ws) (void)hipFree(ws);
-      cap = need < (32u << 20) ? (32u << 20) : need;
-      if (hipMalloc(&ws, cap) != hipSuccess)
-        throw std::runtime_error("gemm_nt split workspace alloc");
+    {
+      std::lock_guard<std::mutex> lk(ws_mu);
+      auto& ent = ws_map[s];
+      if (need > ent.second) {
+        if (ent.first) (void)hipFree(ent.first);
+        ent.second = need < (32u << 20) ? (32u << 20) : need;
+        if (hipMalloc(&ent.first, ent.second) != hipSuccess) {
+          ent = {nullptr, 0};
+          throw std::runtime_error("gemm_nt split workspace alloc");
+        }
+      }
+      ws = ent.first;
     }
     void zero_f32_launch(float*, long, hipStream_t);
     zero_f32_launch(ws, (long)M * N, s);

@@ -146,6 +146,49 @@ class GradientReducer(object):
         # multi-tensor SGD descriptor cache: bucket composition ->
         # (hyper-param signature, desc_blob_dev, b2t_dev)
         self._sgd_cache = {}
+        # bucket composition -> persistent flat fp32 buffer whose views
+        # BACK the bucket's gradient tensors (zero-copy collectives)
+        self._flat = {}
+
+    @staticmethod
+    def _grad_tensors(u):
+        arrs = [u.gradient_weights]
+        if u.include_bias and u.gradient_bias.devmem is not None:
+            arrs.append(u.gradient_bias)
+        return arrs
+
+    def _flat_bucket(self, key):
+        """Persistent flat fp32 buffer backing this bucket's gradients.
+
+        First flush for a composition: allocate one flat buffer, copy
+        the current gradients in, then REPARENT each gradient Array's
+        ``devmem`` to a view of it — every later step's wgrad kernels
+        write straight into the flat buffer and the all-reduce needs no
+        cat/scatter copies (the round-1 version paid two full-bucket
+        device copies per flush).  Segments are 16-B aligned; the pad
+        gaps reduce zeros (harmless).
+        """
+        import torch
+        flat = self._flat.get(key)
+        if flat is not None:
+            return flat
+        segs, total = [], 0
+        for u, _ in self._pending_apply:
+            for arr in self._grad_tensors(u):
+                n = arr.devmem.numel()
+                segs.append((arr, total, n))
+                total += (n + 3) & ~3
+        flat = torch.zeros(total, dtype=torch.float32,
+                           device=self.device.torch_device)
+        for arr, off, n in segs:
+            view = flat[off:off + n].view(arr.devmem.shape)
+            view.copy_(arr.devmem)
+            arr.devmem = view
+        self._flat[key] = flat
+        # SGD descriptor blobs hold raw data pointers into the OLD
+        # gradient tensors — rebuild them against the flat views
+        self._sgd_cache.clear()
+        return flat
 
     @property
     def world(self):
@@ -170,18 +213,13 @@ class GradientReducer(object):
         import torch
         from veles.znicz_amd import ops
         with torch.cuda.stream(self.comm_stream):
+            key = tuple(id(u) for u, _ in self._pending_apply)
             if self.world > 1:
-                # coalesce the bucket's tensors into one collective
-                tensors = []
-                for u, _ in self._pending_apply:
-                    tensors.append(u.gradient_weights.devmem.view(-1))
-                    if u.include_bias and \
-                            u.gradient_bias.devmem is not None:
-                        tensors.append(u.gradient_bias.devmem.view(-1))
-                coalesced_all_reduce(self.ctx, tensors)
+                # ONE collective on the persistent flat buffer that
+                # backs this bucket's gradient tensors (no copies)
+                self.ctx.all_reduce_(self._flat_bucket(key))
             # ONE fused launch for the whole bucket's SGD updates (the
             # per-layer sgd_k launches were a ~5 µs/launch latency tail)
-            key = tuple(id(u) for u, _ in self._pending_apply)
             sig = tuple(
                 (float(u.learning_rate), float(u.weights_decay),
                  float(u.l1_vs_l2), float(u.gradient_moment),
