@@ -104,6 +104,7 @@ ConvDesc make_conv_desc(const std::vector<int64_t>& v) {
   ConvDesc d{(int)v[0], (int)v[1], (int)v[2], (int)v[3], (int)v[4],
              (int)v[5], (int)v[6], (int)v[7], (int)v[8], (int)v[9],
              (int)v[10], (int)v[11], (int)v[12]};
+  finish_conv_desc(d);
   return d;
 }
 

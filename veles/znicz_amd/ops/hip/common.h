@@ -9,6 +9,11 @@
 
 #define WAVE 64
 
+// device side of api.h's DivM (see there for the exactness bound)
+__device__ __forceinline__ unsigned divq(unsigned n, DivM m) {
+  return (unsigned)(((unsigned long long)n * m.mul) >> m.shift);
+}
+
 typedef __attribute__((ext_vector_type(8))) short short8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 typedef __attribute__((ext_vector_type(2))) float f32x2;

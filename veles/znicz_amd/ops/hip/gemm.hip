@@ -409,12 +409,15 @@ struct Im2colTN {
   ConvDesc d;
   int Klim, Mlim;
   __device__ void loadrow8(int k0, int m, unsigned short* v) const {
-    int c = m % d.C, t = m / d.C;
-    int kw = t % d.KW, kh = t / d.KW;
+    unsigned tm = divq((unsigned)m, d.mC);
+    int c = m - (int)(tm * d.C);
+    unsigned khu = divq(tm, d.mKW);
+    int kw = (int)(tm - khu * d.KW), kh = (int)khu;
     const bool mok = (m < Mlim) & (kh < d.KH);
-    int ow = k0 % d.OW;
-    t = k0 / d.OW;
-    int oh = t % d.OH, n = t / d.OH;
+    unsigned tk = divq((unsigned)k0, d.mOW);
+    int ow = k0 - (int)(tk * d.OW);
+    unsigned nu = divq(tk, d.mOH);
+    int oh = (int)(tk - nu * d.OH), n = (int)nu;
     const int iwk = kw - d.PL, ihk = kh - d.PT;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
@@ -581,11 +584,14 @@ struct Im2colTNv {
   int Klim, Mlim;
   __device__ void load16(int k, int mo, unsigned short* v) const {
     const int m = mo * 8;
-    int c = m % d.C, t = m / d.C;        // c stays in one pixel C-run
-    int kw = t % d.KW, kh = t / d.KW;
-    int ow = k % d.OW;
-    int t2 = k / d.OW;
-    int oh = t2 % d.OH, n = t2 / d.OH;
+    unsigned tm = divq((unsigned)m, d.mC);   // c stays in one C-run
+    int c = m - (int)(tm * d.C);
+    unsigned khu = divq(tm, d.mKW);
+    int kw = (int)(tm - khu * d.KW), kh = (int)khu;
+    unsigned tk = divq((unsigned)k, d.mOW);
+    int ow = k - (int)(tk * d.OW);
+    unsigned nu = divq(tk, d.mOH);
+    int oh = (int)(tk - nu * d.OH), n = (int)nu;
     int ih = oh * d.SY - d.PT + kh;
     int iw = ow * d.SX - d.PL + kw;
     const bool ok = (k < Klim) & (m < Mlim) & (kh < d.KH) &
@@ -607,13 +613,16 @@ struct Im2colTNvP {
   int Klim, Mlim;
   __device__ void load16(int k, int mo, unsigned short* v) const {
     const int m = mo * 8;
-    int c = m % d.C, t = m / d.C;
-    int kh = t / d.KW;
-    int ow = k % d.OW;
-    int t2 = k / d.OW;
-    int oh = t2 % d.OH, n = t2 / d.OH;
+    unsigned tm = divq((unsigned)m, d.mC);
+    int c = m - (int)(tm * d.C);
+    unsigned khu = divq(tm, d.mKW);
+    int kwm = (int)(tm - khu * d.KW), kh = (int)khu;
+    unsigned tk = divq((unsigned)k, d.mOW);
+    int ow = k - (int)(tk * d.OW);
+    unsigned nu = divq(tk, d.mOH);
+    int oh = (int)(tk - nu * d.OH), n = (int)nu;
     int ih = oh * d.SY - d.PT + kh;
-    int iw = ow * d.SX - d.PL + (m % (d.KW * d.C)) / d.C;
+    int iw = ow * d.SX - d.PL + kwm;
     const int rowlen = d.KW * d.C;
     long off = (((long)n * d.H + ih) * d.W + iw) * d.C + c;
     const bool fast = ((m % rowlen) + 8 <= rowlen) &&

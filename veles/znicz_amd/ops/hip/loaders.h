@@ -31,15 +31,18 @@ struct Im2colNT {
   const unsigned short* zero;
   ConvDesc d;
   __device__ const void* addr(int row, int k8) const {
-    int k = k8 * 8;
-    int c = k % d.C, t = k / d.C;
-    int kw = t % d.KW, kh = t / d.KW;
-    int ow = row % d.OW;
-    t = row / d.OW;
-    int oh = t % d.OH, n = t / d.OH;
-    int ih = oh * d.SY - d.PT + kh;
-    int iw = ow * d.SX - d.PL + kw;
-    if (kh >= d.KH || (unsigned)ih >= (unsigned)d.H ||
+    unsigned k = (unsigned)k8 * 8;
+    unsigned t = divq(k, d.mC);
+    unsigned c = k - t * d.C;
+    unsigned kh = divq(t, d.mKW);
+    unsigned kw = t - kh * d.KW;
+    unsigned t2 = divq((unsigned)row, d.mOW);
+    unsigned ow = (unsigned)row - t2 * d.OW;
+    unsigned n = divq(t2, d.mOH);
+    unsigned oh = t2 - n * d.OH;
+    int ih = (int)oh * d.SY - d.PT + (int)kh;
+    int iw = (int)ow * d.SX - d.PL + (int)kw;
+    if ((int)kh >= d.KH || (unsigned)ih >= (unsigned)d.H ||
         (unsigned)iw >= (unsigned)d.W)
       return zero;
     return x + (((long)n * d.H + ih) * d.W + iw) * d.C + c;
@@ -73,18 +76,22 @@ struct Im2colNTP {
   const unsigned short* zero;
   ConvDesc d;
   __device__ const void* addr(int row, int k8) const {
-    int k = k8 * 8;
-    int c = k % d.C, t = k / d.C;
-    int kh = t / d.KW;
-    int ow = row % d.OW;
-    t = row / d.OW;
-    int oh = t % d.OH, n = t / d.OH;
-    int ih = oh * d.SY - d.PT + kh;
-    int iw = ow * d.SX - d.PL + (k % (d.KW * d.C)) / d.C;
+    unsigned k = (unsigned)k8 * 8;
+    unsigned t = divq(k, d.mC);
+    unsigned c = k - t * d.C;
+    unsigned kh = divq(t, d.mKW);
+    unsigned kw = t - kh * d.KW;
+    unsigned t2 = divq((unsigned)row, d.mOW);
+    unsigned ow = (unsigned)row - t2 * d.OW;
+    unsigned n = divq(t2, d.mOH);
+    unsigned oh = t2 - n * d.OH;
+    int ih = (int)oh * d.SY - d.PT + (int)kh;
+    int iw = (int)ow * d.SX - d.PL + (int)kw;
     const int rowlen = d.KW * d.C;
-    if ((k % rowlen) + 8 > rowlen || (d.PT | d.PL))
+    const int kinrow = (int)(kw * d.C + c);    // k % (KW*C)
+    if (kinrow + 8 > rowlen || (d.PT | d.PL))
       return nullptr;                  // kh-cross or padded: slow path
-    if (kh >= d.KH || (unsigned)ih >= (unsigned)d.H)
+    if ((int)kh >= d.KH || (unsigned)ih >= (unsigned)d.H)
       return zero;
     long off = (((long)n * d.H + ih) * d.W + iw) * d.C + c;
     return (off & 3) ? nullptr : (const void*)(x + off);
@@ -141,18 +148,23 @@ struct DgradNT {
   const unsigned short* zero;
   ConvDesc d;
   __device__ const void* addr(int row, int k8) const {
-    int k = k8 * 8;
-    int co = k % d.K, t = k / d.K;
-    int kw = t % d.KW, kh = t / d.KW;
-    int iw = row % d.W;
-    t = row / d.W;
-    int ih = t % d.H, n = t / d.H;
-    int ohn = ih + d.PT - kh;
-    int own = iw + d.PL - kw;
-    if (kh >= d.KH || ohn < 0 || own < 0 || ohn % d.SY || own % d.SX)
+    unsigned k = (unsigned)k8 * 8;
+    unsigned t = divq(k, d.mK);
+    unsigned co = k - t * d.K;
+    unsigned kh = divq(t, d.mKW);
+    unsigned kw = t - kh * d.KW;
+    unsigned t2 = divq((unsigned)row, d.mW);
+    unsigned iw = (unsigned)row - t2 * d.W;
+    unsigned n = divq(t2, d.mH);
+    unsigned ih = t2 - n * d.H;
+    int ohn = (int)ih + d.PT - (int)kh;
+    int own = (int)iw + d.PL - (int)kw;
+    unsigned oh = divq((unsigned)ohn, d.mSY);
+    unsigned ow = divq((unsigned)own, d.mSX);
+    if ((int)kh >= d.KH || ohn < 0 || own < 0 ||
+        ohn != (int)(oh * d.SY) || own != (int)(ow * d.SX))
       return zero;
-    int oh = ohn / d.SY, ow = own / d.SX;
-    if (oh >= d.OH || ow >= d.OW) return zero;
+    if ((int)oh >= d.OH || (int)ow >= d.OW) return zero;
     return eo + (((long)n * d.OH + oh) * d.OW + ow) * d.K + co;
   }
   __device__ unsigned short elem(int row, int k) const {
@@ -179,15 +191,18 @@ struct DgradNT1 {
   const unsigned short* zero;
   ConvDesc d;
   __device__ const void* addr(int row, int k8) const {
-    int k = k8 * 8;
-    int co = k % d.K, t = k / d.K;
-    int kw = t % d.KW, kh = t / d.KW;
-    int iw = row % d.W;
-    t = row / d.W;
-    int ih = t % d.H, n = t / d.H;
-    int oh = ih + d.PT - kh;
-    int ow = iw + d.PL - kw;
-    if (kh >= d.KH || (unsigned)oh >= (unsigned)d.OH ||
+    unsigned k = (unsigned)k8 * 8;
+    unsigned t = divq(k, d.mK);
+    unsigned co = k - t * d.K;
+    unsigned kh = divq(t, d.mKW);
+    unsigned kw = t - kh * d.KW;
+    unsigned t2 = divq((unsigned)row, d.mW);
+    unsigned iw = (unsigned)row - t2 * d.W;
+    unsigned n = divq(t2, d.mH);
+    unsigned ih = t2 - n * d.H;
+    int oh = (int)ih + d.PT - (int)kh;
+    int ow = (int)iw + d.PL - (int)kw;
+    if ((int)kh >= d.KH || (unsigned)oh >= (unsigned)d.OH ||
         (unsigned)ow >= (unsigned)d.OW)
       return zero;
     return eo + (((long)n * d.OH + oh) * d.OW + ow) * d.K + co;
