@@ -153,11 +153,30 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_k(
   f32x4 acc[4][4] = {};
 
   if (t_lo >= nk) return;
+  // Counted-vmcnt pipeline (guide T3/T4): tile t+1's 8 glds are issued
+  // right after the barrier that frees its buffer, and the wait before
+  // computing tile t is `vmcnt(8)` — t's own loads retired (in-order),
+  // t+1's stay IN FLIGHT across the raw barrier.  __syncthreads() here
+  // would emit s_waitcnt vmcnt(0) (glds is a pending LDS write on the
+  // VM counter) and drain the prefetch every tile — the measured ~20%
+  // stall of the 2-barrier structure.  Straddle/slow chunks take
+  // ordinary loads whose count breaks the formula -> those shapes wait
+  // vmcnt(0) instead (still pipelined one tile deep).
+  const bool counted = !(LA::MAY_SLOW || LB::MAY_SLOW) && (K % 8 == 0);
   stage(0, t_lo);
-  __syncthreads();
   int buf = 0;
   for (int t = t_lo; t < nk; ++t) {
-    if (t + 1 < nk) stage(buf ^ 1, t + 1);
+    __builtin_amdgcn_s_barrier();      // all waves done reading buf^1
+    if (t + 1 < nk) {
+      stage(buf ^ 1, t + 1);
+      if (counted)
+        asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+      else
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();      // tile t's LDS complete everywhere
     const char* As = smem + buf * NT_LDS_BUF;
     const char* Bs = As + NT_LDS_A;
     const int fr = lane & 15;          // fragment row/col
@@ -184,7 +203,6 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_k(
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
     }
-    __syncthreads();
     buf ^= 1;
   }
 
@@ -301,11 +319,23 @@ __global__ __launch_bounds__(256, 2) void gemm_ntn_k(
 
   f32x4 acc[2][BNT / 16] = {};
 
+  // counted-vmcnt pipeline — same discipline as gemm_nt_k above
+  const bool counted = !(LA::MAY_SLOW || LB::MAY_SLOW) && (K % 8 == 0);
+  constexpr int NLOADS = 4 + BNT / 32;
   stage(0, 0);
-  __syncthreads();
   int buf = 0;
   for (int t = 0; t < nk; ++t) {
-    if (t + 1 < nk) stage(buf ^ 1, t + 1);
+    __builtin_amdgcn_s_barrier();
+    if (t + 1 < nk) {
+      stage(buf ^ 1, t + 1);
+      if (counted)
+        asm volatile("s_waitcnt vmcnt(%0)" :: "i"(NLOADS) : "memory");
+      else
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
     const char* As = smem + buf * LDS_BUF;
     const char* Bs = As + NT_LDS_A;
     const int fr = lane & 15;
@@ -332,7 +362,6 @@ __global__ __launch_bounds__(256, 2) void gemm_ntn_k(
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
     }
-    __syncthreads();
     buf ^= 1;
   }
 
@@ -1062,27 +1091,29 @@ void gemm_nt_dense(const unsigned short* A, const unsigned short* B,
   const int splits = max(1, min(nk, 512 / tiles));
   if (splits > 1) {
     // fp32 workspace + atomic partials, then one bias/act/cast pass.
-    // The workspace is PER-STREAM (map below): the comm-stream design
-    // allows two streams to hit FC split-K concurrently, and a shared
-    // buffer would race between their atomics and finisher passes.
+    // ONE process-global workspace: every FC split-K launch in this
+    // engine runs on the compute stream (or its hipGraph capture
+    // proxy), strictly serialized by the unit graph — the comm stream
+    // never calls gemm_nt_dense.  A per-stream map is NOT usable here:
+    // torch captures on a fresh internal stream, and a lazy hipMalloc
+    // during capture invalidates the whole hipGraph.  Warmup steps
+    // (always before capture) size the buffer for the model's shapes.
     static std::mutex ws_mu;
-    static std::unordered_map<hipStream_t, std::pair<float*, size_t>>
-        ws_map;
-    float* ws;
+    static float* ws = nullptr;
+    static size_t cap = 0;
     size_t need = (size_t)cdiv(M, BM) * BM * cdiv(N, BN) * BN *
                   sizeof(float);
     {
       std::lock_guard<std::mutex> lk(ws_mu);
-      auto& ent = ws_map[s];
-      if (need > ent.second) {
-        if (ent.first) (void)hipFree(ent.first);
-        ent.second = need < (32u << 20) ? (32u << 20) : need;
-        if (hipMalloc(&ent.first, ent.second) != hipSuccess) {
-          ent = {nullptr, 0};
+      if (need > cap) {
+        if (ws) (void)hipFree(ws);
+        cap = need < (32u << 20) ? (32u << 20) : need;
+        if (hipMalloc(&ws, cap) != hipSuccess) {
+          ws = nullptr;
+          cap = 0;
           throw std::runtime_error("gemm_nt split workspace alloc");
         }
       }
-      ws = ent.first;
     }
     void zero_f32_launch(float*, long, hipStream_t);
     zero_f32_launch(ws, (long)M * N, s);

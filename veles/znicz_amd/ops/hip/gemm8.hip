@@ -112,31 +112,34 @@ __global__ __launch_bounds__(O_THREADS, 1) void gemm_nt8_k(
   const int b_which = 2 + (wn >> 1);
   const int b_row0 = (wn & 1) * 64;
 
-  // ---- prologue: stage K-tile 0, drain, enter the pipeline --------------
+  // ---- prologue: stage K-tile 0, enter the pipeline ----------------------
   if (nk > 0) stage_ktile(0, 0);
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-  __builtin_amdgcn_s_barrier();
 
-  // Steady state per K-tile t (buffer t&1): phase q computes C-quadrant
-  // q against tile t AND stages half q of tile t+1 into the buffer just
-  // freed by tile t-1.  One counted wait per tile — vmcnt(2) after
-  // phase 0 issues its own 2 glds — so ≥2 prefetch loads stay in
-  // flight across every barrier (never vmcnt(0) in the loop).
+  // Counted-vmcnt pipeline (guide T3/T4, same discipline as the 128²
+  // core): barrier #1 frees tile t-1's buffer, ALL 8 of tile t+1's glds
+  // issue into it at once (maximum latency cover: the loads have the
+  // whole of tile t's compute — 4 quadrant phases — to land), then
+  // `vmcnt(8)` retires exactly tile t's loads while t+1's stay in
+  // flight across the raw barrier #2.  No barriers between quadrant
+  // phases: waves drift inside a tile (role diversity for s_setprio)
+  // and re-sync at the next tile's barrier; the compiler's own counted
+  // lgkmcnt waits pace ds_read→MFMA.
   int buf = 0;
   for (int t = 0; t < nk; ++t) {
+    __builtin_amdgcn_s_barrier();      // all waves done reading buf^1
+    if (t + 1 < nk) {
+      stage_ktile(buf ^ 1, t + 1);
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();      // tile t's LDS complete everywhere
     const char* Ab = smem + buf * O_KTILE + a_which * O_HALF;
     const char* Bb = smem + buf * O_KTILE + b_which * O_HALF;
-    const bool pf = t + 1 < nk;
 #pragma unroll
     for (int q = 0; q < 4; ++q) {
       const int qm = (q >> 1) * 4;            // fragment-row base 0|4
       const int qn = (q & 1) * 2;             // fragment-col base 0|2
-      if (pf) stage_half(q, buf ^ 1, t + 1);
-      if (q == 0)
-        asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
-      // all 12 fragment reads BEFORE the barrier: their LDS latency
-      // hides behind the waves' barrier skew (the template's phase
-      // order); the compiler's own lgkmcnt waits guard the MFMA uses
       short8 af0[4], af1[4], bf[2][2];
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks)
@@ -148,9 +151,6 @@ __global__ __launch_bounds__(O_THREADS, 1) void gemm_nt8_k(
         af0[i] = rd_frag(Ab, (qm + i) * 16, 0);
         af1[i] = rd_frag(Ab, (qm + i) * 16, 1);
       }
-      if (q == 3)   // retire reads before the next tile overwrites buf
-        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      __builtin_amdgcn_s_barrier();
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int i = 0; i < 4; ++i)
