@@ -595,12 +595,20 @@ __global__ void scale_f32_k(float* __restrict__ p, long n, float beta) {
 #define TN2_LDS_TOTAL (TN2_LDS_BUF * 2) // double-buffered
 
 struct DenseTNv {
+  static constexpr bool HAS_ADDR16 = true;
   const unsigned short* base;
   long ld;
   int Klim, Mlim;
+  // T14 split API: address+bounds now, raw load by the caller, the
+  // ok?v:0 select deferred to the LDS-write pass (a select at load
+  // time forces the vmcnt wait at ISSUE, defeating the async split)
+  __device__ const unsigned short* addr16(int k, int mo, bool& ok) const {
+    ok = (k < Klim) & (mo * 8 + 8 <= Mlim);
+    return ok ? base + (long)k * ld + mo * 8 : base;
+  }
   __device__ void load16(int k, int mo, unsigned short* v) const {
-    const bool ok = (k < Klim) & (mo * 8 + 8 <= Mlim);
-    const unsigned short* p = ok ? base + (long)k * ld + mo * 8 : base;
+    bool ok;
+    const unsigned short* p = addr16(k, mo, ok);
     *(short8*)v = *(const short8*)p;     // unconditional 16-B load
 #pragma unroll
     for (int j = 0; j < 8; ++j) v[j] = ok ? v[j] : (unsigned short)0;
@@ -608,10 +616,11 @@ struct DenseTNv {
 };
 
 struct Im2colTNv {
+  static constexpr bool HAS_ADDR16 = true;
   const unsigned short* x;
   ConvDesc d;
   int Klim, Mlim;
-  __device__ void load16(int k, int mo, unsigned short* v) const {
+  __device__ const unsigned short* addr16(int k, int mo, bool& ok) const {
     const int m = mo * 8;
     unsigned tm = divq((unsigned)m, d.mC);   // c stays in one C-run
     int c = m - (int)(tm * d.C);
@@ -623,11 +632,14 @@ struct Im2colTNv {
     int oh = (int)(tk - nu * d.OH), n = (int)nu;
     int ih = oh * d.SY - d.PT + kh;
     int iw = ow * d.SX - d.PL + kw;
-    const bool ok = (k < Klim) & (m < Mlim) & (kh < d.KH) &
-                    ((unsigned)ih < (unsigned)d.H) &
-                    ((unsigned)iw < (unsigned)d.W);
-    const unsigned short* p =
-        ok ? x + (((long)n * d.H + ih) * d.W + iw) * d.C + c : x;
+    ok = (k < Klim) & (m < Mlim) & (kh < d.KH) &
+         ((unsigned)ih < (unsigned)d.H) &
+         ((unsigned)iw < (unsigned)d.W);
+    return ok ? x + (((long)n * d.H + ih) * d.W + iw) * d.C + c : x;
+  }
+  __device__ void load16(int k, int mo, unsigned short* v) const {
+    bool ok;
+    const unsigned short* p = addr16(k, mo, ok);
     *(short8*)v = *(const short8*)p;
 #pragma unroll
     for (int j = 0; j < 8; ++j) v[j] = ok ? v[j] : (unsigned short)0;
@@ -637,6 +649,7 @@ struct Im2colTNv {
 // packed C%4 variant (RGB padded to 4): a SEPARATE type so the
 // fallback machinery costs the C%8 wgrad instantiations nothing
 struct Im2colTNvP {
+  static constexpr bool HAS_ADDR16 = false;  // refused-chunk machinery
   const unsigned short* x;
   ConvDesc d;
   int Klim, Mlim;
@@ -732,11 +745,60 @@ __global__ __launch_bounds__(256, 2) void gemm_tn2_k(
 
   f32x4 acc[4][4] = {};
   if (t_lo >= t_hi) return;
-  stage2(0, t_lo);
-  __syncthreads();
+
+  // T14 async-stage split (PMC round 2: waves were 48% parked in
+  // waits, MFMA 26% busy, with the synchronous load->ds_write stage):
+  // tile t+1's raw 16-B loads are ISSUED one full compute phase before
+  // their vmcnt wait; the bounds select + ds_write happen after the
+  // barrier that frees the target buffer, and the next tile's loads
+  // re-issue immediately so they span the following compute.
+  unsigned short vA[TBK / 16][8], vB[TBK / 16][8];
+  bool okA[TBK / 16], okB[TBK / 16];
+  auto load_regs = [&](int kt) {
+    if constexpr (LA::HAS_ADDR16 && LB::HAS_ADDR16) {
+      const int k0t = kt * TBK;
+      const int oct = tid & 15;
+      const int kk0 = tid >> 4;
+#pragma unroll
+      for (int pass = 0; pass < TBK / 16; ++pass) {
+        const int kk = pass * 16 + kk0;
+        *(short8*)vA[pass] =
+            *(const short8*)la.addr16(k0t + kk, moA + oct, okA[pass]);
+        *(short8*)vB[pass] =
+            *(const short8*)lb.addr16(k0t + kk, moB + oct, okB[pass]);
+      }
+    }
+  };
+  auto write_lds = [&](int buf_) {
+    char* As_ = smem + buf_ * TN2_LDS_BUF;
+    char* Bs_ = As_ + TN2_LDS_OP;
+    const int oct = tid & 15;
+    const int kk0 = tid >> 4;
+#pragma unroll
+    for (int pass = 0; pass < TBK / 16; ++pass) {
+      const int kk = pass * 16 + kk0;
+      short8 a = *(short8*)vA[pass], b = *(short8*)vB[pass];
+      *(short8*)(As_ + (kk * TNP + oct * 8) * 2) =
+          okA[pass] ? a : (short8)0;
+      *(short8*)(Bs_ + (kk * TNP + oct * 8) * 2) =
+          okB[pass] ? b : (short8)0;
+    }
+  };
+
   int buf = 0;
+  if constexpr (LA::HAS_ADDR16 && LB::HAS_ADDR16) {
+    load_regs(t_lo);
+    write_lds(0);
+    if (t_lo + 1 < t_hi) load_regs(t_lo + 1);
+    __syncthreads();
+  } else {
+    stage2(0, t_lo);
+    __syncthreads();
+  }
   for (int t = t_lo; t < t_hi; ++t) {
-    if (t + 1 < t_hi) stage2(buf ^ 1, t + 1);
+    if constexpr (!(LA::HAS_ADDR16 && LB::HAS_ADDR16)) {
+      if (t + 1 < t_hi) stage2(buf ^ 1, t + 1);
+    }
     const unsigned short* As =
         (const unsigned short*)(smem + buf * TN2_LDS_BUF);
     const unsigned short* Bs = As + TBK * TNP;
@@ -780,6 +842,16 @@ __global__ __launch_bounds__(256, 2) void gemm_tn2_k(
         for (int ni = 0; ni < 4; ++ni)
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+    }
+    if constexpr (LA::HAS_ADDR16 && LB::HAS_ADDR16) {
+      // write target buf^1 was freed by the PREVIOUS iteration's
+      // trailing barrier (tile t-1's reads all precede it), and
+      // compute(t) above reads buf — no conflict, so one barrier per
+      // tile: writes visible before the next iteration's reads.
+      if (t + 1 < t_hi) {
+        write_lds(buf ^ 1);            // tile t+1 from the in-flight regs
+        if (t + 2 < t_hi) load_regs(t + 2);  // re-issue immediately
+      }
     }
     __syncthreads();
     buf ^= 1;
@@ -869,11 +941,66 @@ __global__ __launch_bounds__(256, 2) void gemm_tn2n_k(
 
   f32x4 acc[BMT / 16][2] = {};
   if (t_lo >= t_hi) return;
-  stage2(0, t_lo);
-  __syncthreads();
+
+  // T14 async-stage split (see gemm_tn2_k): raw loads issue a full
+  // compute phase ahead; bounds select, bias co-accumulation and
+  // ds_write happen in the write pass after the barrier.
+  unsigned short nvA[8], nvB[TBK / 16][8];
+  bool nokA = false, nokB[TBK / 16];
+  const int oaA = tid & (BMT / 8 - 1);
+  const int kaA = tid / (BMT / 8);
+  auto load_regs = [&](int kt) {
+    if constexpr (LA::HAS_ADDR16 && LB::HAS_ADDR16) {
+      const int k0t = kt * TBK;
+      if (kaA < TBK)
+        *(short8*)nvA =
+            *(const short8*)la.addr16(k0t + kaA, moA + oaA, nokA);
+      const int oct = tid & 15;
+      const int kk0 = tid >> 4;
+#pragma unroll
+      for (int pass = 0; pass < TBK / 16; ++pass) {
+        const int kk = pass * 16 + kk0;
+        *(short8*)nvB[pass] =
+            *(const short8*)lb.addr16(k0t + kk, moB + oct, nokB[pass]);
+      }
+    }
+  };
+  auto write_lds = [&](int buf_) {
+    char* As_ = smem + buf_ * TN2_LDS_BUF;
+    char* Bs_ = As_ + TN2_LDS_OP;
+    if (kaA < TBK) {
+      short8 a = *(short8*)nvA;
+      a = nokA ? a : (short8)0;
+      *(short8*)(As_ + (kaA * TNP + oaA * 8) * 2) = a;
+      if (do_bias)
+#pragma unroll
+        for (int j = 0; j < 8; ++j) bsum[j] += bf2f(((unsigned short*)&a)[j]);
+    }
+    const int oct = tid & 15;
+    const int kk0 = tid >> 4;
+#pragma unroll
+    for (int pass = 0; pass < TBK / 16; ++pass) {
+      const int kk = pass * 16 + kk0;
+      short8 b = *(short8*)nvB[pass];
+      *(short8*)(Bs_ + (kk * TNP + oct * 8) * 2) =
+          nokB[pass] ? b : (short8)0;
+    }
+  };
+
   int buf = 0;
+  if constexpr (LA::HAS_ADDR16 && LB::HAS_ADDR16) {
+    load_regs(t_lo);
+    write_lds(0);
+    if (t_lo + 1 < t_hi) load_regs(t_lo + 1);
+    __syncthreads();
+  } else {
+    stage2(0, t_lo);
+    __syncthreads();
+  }
   for (int t = t_lo; t < t_hi; ++t) {
-    if (t + 1 < t_hi) stage2(buf ^ 1, t + 1);
+    if constexpr (!(LA::HAS_ADDR16 && LB::HAS_ADDR16)) {
+      if (t + 1 < t_hi) stage2(buf ^ 1, t + 1);
+    }
     const unsigned short* As =
         (const unsigned short*)(smem + buf * TN2_LDS_BUF);
     const unsigned short* Bs = As + TBK * TNP;
@@ -914,6 +1041,12 @@ __global__ __launch_bounds__(256, 2) void gemm_tn2n_k(
         for (int ni = 0; ni < 2; ++ni)
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+    }
+    if constexpr (LA::HAS_ADDR16 && LB::HAS_ADDR16) {
+      if (t + 1 < t_hi) {
+        write_lds(buf ^ 1);
+        if (t + 2 < t_hi) load_regs(t + 2);
+      }
     }
     __syncthreads();
     buf ^= 1;
