@@ -195,9 +195,11 @@ def main():
             "config": {"model": args.model,
                        "global_batch": batch * world,
                        "input": list(sample[1:]),
-                       # RGB padded to 8 channels for the NHWC HIP conv
-                       # DMA constraint — the kernels do MORE work than
-                       # the 3-channel baseline config requires
+                       # RGB padded to 8 channels for the NHWC conv DMA
+                       # grain — the kernels do MORE work than the
+                       # 3-channel config requires (a C=4 materialized-
+                       # im2col path exists and is GPU-tested, but
+                       # measured net-slower: profiles/perf_r2.md)
                        "channels_padded_from": 3,
                        "parallelism": "dp%d" % world},
         }))

@@ -25,6 +25,7 @@ REPS = 20
 # does:  (name, N, H, W, C, K, KH, KW, SY, SX, PT, PL)
 CONVS = [
     ("conv1", 256, 227, 227, 8, 96, 11, 11, 4, 4, 0, 0),
+    ("conv1c4", 256, 227, 227, 4, 96, 11, 11, 4, 4, 0, 0),
     ("conv2", 256, 27, 27, 96, 256, 5, 5, 1, 1, 2, 2),
     ("conv3", 256, 13, 13, 256, 384, 3, 3, 1, 1, 1, 1),
     ("conv4", 256, 13, 13, 384, 384, 3, 3, 1, 1, 1, 1),

@@ -69,12 +69,29 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_k(
   // addresses ≡ 0 (mod 64 dw) — an 8-way bank conflict measured at
   // 13-16% of NT wave cycles.  Swizzling WHICH k-octet each slot loads
   // spreads the 16 lanes of a fragment read over distinct banks.
+  // Per-thread ROW contexts are loop-invariant across K-tiles: the
+  // magic-division row decomposition runs once per kernel here, and
+  // per tile only the two k-octet decompositions + add/bounds remain
+  // (the staging address math used to dominate the VALU pipe).
+  const int rl = lane >> 3;            // row within 8-row chunk
+  const int k8 = lane & 7;             // 16-B slot within the 64-k row
+  typename LA::RowC rca[4];
+  typename LB::RowC rcb[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int row = (wave * 4 + i) * 8 + rl;
+    rca[i] = la.rowc(min(m0 + row, M - 1));
+    rcb[i] = lb.rowc(min(n0 + row, N - 1));
+  }
+
   auto stage = [&](int buf, int kt) {
     char* As = smem + buf * NT_LDS_BUF;
     char* Bs = As + NT_LDS_A;
     const int k8base = kt * (BK / 8);
-    const int rl = lane >> 3;          // row within 8-row chunk
-    const int k8 = lane & 7;           // 16-B slot within the 64-k row
+    const auto kca0 = la.kc(k8base + (k8 ^ rl));
+    const auto kca1 = la.kc(k8base + (k8 ^ rl ^ 1));
+    const auto kcb0 = lb.kc(k8base + (k8 ^ rl));
+    const auto kcb1 = lb.kc(k8base + (k8 ^ rl ^ 1));
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
       const int chunk = wave * 4 + i;  // 16 chunks x 8 rows = 128 rows
@@ -85,12 +102,12 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_k(
       if constexpr (!LA::MAY_SLOW && !LB::MAY_SLOW) {
         // fast loaders: the only per-element case is the K%8 tail
         if (!straddle) {
-          const void* sa = la.addr(min(m0 + row, M - 1), k8base + k8e);
+          const void* sa = la.addr_rk(rca[i], (i & 1) ? kca1 : kca0);
           __builtin_amdgcn_global_load_lds(
               (const __attribute__((address_space(1))) void*)sa,
               (__attribute__((address_space(3))) void*)(As + chunk * 1024),
               16, 0, 0);
-          const void* sb = lb.addr(min(n0 + row, N - 1), k8base + k8e);
+          const void* sb = lb.addr_rk(rcb[i], (i & 1) ? kcb1 : kcb0);
           __builtin_amdgcn_global_load_lds(
               (const __attribute__((address_space(1))) void*)sb,
               (__attribute__((address_space(3))) void*)(Bs + chunk * 1024),
@@ -259,12 +276,27 @@ __global__ __launch_bounds__(256, 2) void gemm_ntn_k(
 
   const int nk = (K + BK - 1) / BK;
 
+  const int rl = lane >> 3;
+  const int k8 = lane & 7;
+  // loop-invariant per-thread row contexts (see gemm_nt_k)
+  typename LA::RowC rca[4];
+  typename LB::RowC rcb[BNT / 32];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+    rca[i] = la.rowc(min(m0 + (wave * 4 + i) * 8 + rl, M - 1));
+#pragma unroll
+  for (int i = 0; i < BNT / 32; ++i)
+    rcb[i] = lb.rowc(
+        min(n0 + (wave * (BNT / 32) + i) * 8 + rl, N - 1));
+
   auto stage = [&](int buf, int kt) {
     char* As = smem + buf * LDS_BUF;
     char* Bs = As + NT_LDS_A;
     const int k8base = kt * (BK / 8);
-    const int rl = lane >> 3;
-    const int k8 = lane & 7;
+    const auto kca0 = la.kc(k8base + (k8 ^ rl));
+    const auto kca1 = la.kc(k8base + (k8 ^ rl ^ 1));
+    const auto kcb0 = lb.kc(k8base + (k8 ^ rl));
+    const auto kcb1 = lb.kc(k8base + (k8 ^ rl ^ 1));
 #pragma unroll
     for (int i = 0; i < 4; ++i) {       // A: 16 chunks × 8 rows
       const int chunk = wave * 4 + i;
@@ -272,11 +304,9 @@ __global__ __launch_bounds__(256, 2) void gemm_ntn_k(
       const int k8e = k8 ^ rl ^ (chunk & 1);
       const int kc0 = (k8base + k8e) * 8;
       const bool straddle = (kc0 < K) & (kc0 + 8 > K);
-      const void* sa = (straddle || LA::MAY_SLOW)
-                           ? (straddle ? nullptr
-                                       : la.addr(min(m0 + row, M - 1),
-                                                 k8base + k8e))
-                           : la.addr(min(m0 + row, M - 1), k8base + k8e);
+      const void* sa = straddle
+                           ? nullptr
+                           : la.addr_rk(rca[i], (chunk & 1) ? kca1 : kca0);
       if (!LA::MAY_SLOW ? !straddle : (sa != nullptr)) {
         __builtin_amdgcn_global_load_lds(
             (const __attribute__((address_space(1))) void*)sa,
@@ -297,11 +327,9 @@ __global__ __launch_bounds__(256, 2) void gemm_ntn_k(
       const int k8e = k8 ^ rl ^ (chunk & 1);
       const int kc0 = (k8base + k8e) * 8;
       const bool straddle = (kc0 < K) & (kc0 + 8 > K);
-      const void* sb = (straddle || LB::MAY_SLOW)
-                           ? (straddle ? nullptr
-                                       : lb.addr(min(n0 + row, N - 1),
-                                                 k8base + k8e))
-                           : lb.addr(min(n0 + row, N - 1), k8base + k8e);
+      const void* sb = straddle
+                           ? nullptr
+                           : lb.addr_rk(rcb[i], (chunk & 1) ? kcb1 : kcb0);
       if (!LB::MAY_SLOW ? !straddle : (sb != nullptr)) {
         __builtin_amdgcn_global_load_lds(
             (const __attribute__((address_space(1))) void*)sb,
@@ -1131,6 +1159,20 @@ void launch_gemm_nt(const LA& la, const LB& lb, OUT_T* C, const float* bias,
     HIP_CHECK_LAST();
     return;
   }
+  if (N <= 96) {
+    // AlexNet conv1 class (K=96 kernels): the 128-wide tile would
+    // waste 25% of the MFMA columns
+    const int tn = cdiv(N, 96);
+    dim3 grid(cdiv(M, BM) * tn), block(256);
+#define GO96(A)                                                              \
+    hipLaunchKernelGGL((gemm_ntn_k<LA, LB, OUT_T, A, 96>), grid, block,      \
+                       (NT_LDS_A + 96 * BK * 2) * 2, stream, la, lb, C,      \
+                       bias, M, N, K, tn)
+    ACT_SWITCH(act, GO96);
+#undef GO96
+    HIP_CHECK_LAST();
+    return;
+  }
   const int tn = cdiv(N, BN);
   dim3 grid(cdiv(M, BM) * tn), block(256);
 #define GO(A)                                                                \
@@ -1290,6 +1332,100 @@ bool gemm_tn_dense(const unsigned short* A, const unsigned short* B,
   return false;   // caller computes the bias gradient separately
 }
 
+// ---------------------------------------------------------------------------
+// materialized im2col for C % 8 != 0 convs (honest RGB→4 first layer):
+// the implicit packed loaders measured ~3× lower GEMM efficiency (mixed
+// 8-B source parity halves DMA width), so these convs write the unroll
+// ONCE as a [N·OH·OW, Kpad] bf16 image (Kpad = K rounded to 8, all
+// chunks 16-B aligned) and run plain dense GEMMs — the image is shared
+// by forward and wgrad within a step (pointer+geometry cache below).
+// ---------------------------------------------------------------------------
+// One thread per 8-B unit (4 bf16 = one C-quad of one (kh,kw) tap),
+// unit-linear over the OUTPUT so every wave's 64 stores are 512
+// contiguous bytes (the first cut mapped threads to rows — 8-B stores
+// at a Kpad-stride, fully uncoalesced, and cost more than it saved).
+__global__ void im2col_mat_k(const unsigned short* __restrict__ x,
+                             unsigned short* __restrict__ out,
+                             ConvDesc d, int Kpad, int nunits,
+                             DivM mUrow, DivM mKWc, DivM mc4) {
+  const unsigned u = blockIdx.x * blockDim.x + threadIdx.x;
+  if (u >= (unsigned)nunits) return;
+  // u -> (row, unit j within the padded row)
+  const unsigned r = divq(u, mUrow);
+  const unsigned j = u - r * (unsigned)(Kpad >> 2);
+  // j -> (kh, kw, c4); j >= K/4 is the alignment pad tail (zeros)
+  const unsigned kh = divq(j, mKWc);
+  const unsigned jr = j - kh * ((unsigned)d.KW * ((unsigned)d.C >> 2));
+  const unsigned kw = divq(jr, mc4);
+  const unsigned c4 = jr - kw * ((unsigned)d.C >> 2);
+  // row -> (n, oh, ow)
+  unsigned t2 = divq(r, d.mOW);
+  unsigned ow = r - t2 * d.OW;
+  unsigned n = divq(t2, d.mOH);
+  unsigned oh = t2 - n * d.OH;
+  const int ih = (int)oh * d.SY - d.PT + (int)kh;
+  const int iw = (int)ow * d.SX - d.PL + (int)kw;
+  const bool ok = (j < (unsigned)((d.KH * d.KW * d.C) >> 2)) &&
+                  (unsigned)ih < (unsigned)d.H &&
+                  (unsigned)iw < (unsigned)d.W;
+  unsigned long long v =
+      ok ? *(const unsigned long long*)(
+               x + ((((long)n * d.H + ih) * d.W + iw) * d.C) + c4 * 4)
+         : 0ull;
+  *(unsigned long long*)(out + (long)u * 4) = v;
+}
+
+static unsigned short* im2col_ws_buf(size_t bytes) {
+  static std::mutex mu;
+  static unsigned short* ws = nullptr;
+  static size_t cap = 0;
+  std::lock_guard<std::mutex> lk(mu);
+  if (bytes > cap) {
+    if (ws) (void)hipFree(ws);
+    cap = bytes;
+    if (hipMalloc(&ws, cap) != hipSuccess) {
+      ws = nullptr;
+      cap = 0;
+      throw std::runtime_error("im2col workspace alloc");
+    }
+  }
+  return ws;
+}
+
+// Reuse cache: forward ALWAYS materializes (fresh minibatch each step);
+// wgrad reuses the image iff the immediately-preceding materialize was
+// for the same input+geometry (true within one step unless another
+// C%8!=0 conv ran in between).  Under hipGraph capture the decision is
+// baked once and stays correct: the captured sequence re-runs the
+// forward's materialize node before the wgrad GEMM on every replay.
+static const void* g_i2c_key_x = nullptr;
+static long g_i2c_key_geom = 0;
+
+static unsigned short* materialize_im2col(const unsigned short* x,
+                                          const ConvDesc& d, int Kpad,
+                                          hipStream_t s, bool allow_reuse) {
+  const int rows = d.N * d.OH * d.OW;
+  unsigned short* ws = im2col_ws_buf((size_t)rows * Kpad * 2);
+  const long geom = ((long)d.N << 44) ^ ((long)d.H << 30) ^
+                    ((long)d.C << 24) ^ ((long)d.KH << 18) ^
+                    ((long)d.KW << 12) ^ ((long)d.SY << 8) ^ d.OW;
+  if (allow_reuse && g_i2c_key_x == x && g_i2c_key_geom == geom)
+    return ws;
+  const long nunits_l = (long)rows * (Kpad >> 2);
+  if (nunits_l >= (1l << 31))
+    throw std::runtime_error("im2col image too large for one pass");
+  const int nunits = (int)nunits_l;
+  dim3 grid((nunits + 255) / 256), block(256);
+  hipLaunchKernelGGL(im2col_mat_k, grid, block, 0, s, x, ws, d, Kpad,
+                     nunits, make_divm((unsigned)(Kpad >> 2)),
+                     make_divm((unsigned)(d.KW * (d.C >> 2))),
+                     make_divm((unsigned)(d.C >> 2)));
+  HIP_CHECK_LAST();
+  g_i2c_key_x = x;
+  g_i2c_key_geom = geom;
+  return ws;
+}
+
 void conv_fwd_gemm8(const unsigned short*, const unsigned short*, void*,
                     bool, const float*, const unsigned short*,
                     const ConvDesc&, int, hipStream_t);
@@ -1313,6 +1449,20 @@ void conv_fwd_gemm(const unsigned short* x, const unsigned short* w,
     return;
   }
   if (d.C & 7) {
+    if ((d.C & 3) == 0) {
+      // honest C=4 path: one-shot materialized im2col + dense GEMM
+      const int Kpad = (K + 7) & ~7;
+      unsigned short* col = materialize_im2col(x, d, Kpad, s,
+                                               /*allow_reuse=*/false);
+      DenseNT la{col, zero, Kpad, Kpad};
+      DenseNT lb{w, zero, K, K};        // k >= K reads the zero page
+      if (out_f32)
+        launch_gemm_nt(la, lb, (float*)out, bias, M, d.K, Kpad, act, s);
+      else
+        launch_gemm_nt(la, lb, (unsigned short*)out, bias, M, d.K, Kpad,
+                       act, s);
+      return;
+    }
     Im2colNTP la{x, zero, d};
     DenseNT lb{w, zero, K, K};
     if (out_f32)
@@ -1364,8 +1514,13 @@ bool conv_wgrad_gemm(const unsigned short* eo, const unsigned short* x,
     return launch_gemm_tn2(la, lb, gw, M, N, Kdim, beta, s, bias_out);
   }
   if ((M & 7) == 0 && (d.C & 3) == 0) {
+    // reuse (or rebuild) the forward's materialized im2col image and
+    // run a plain dense K-major GEMM — the image IS K-major for TN
+    const int Kpad = (N + 7) & ~7;
+    unsigned short* col = materialize_im2col(x, d, Kpad, s,
+                                             /*allow_reuse=*/true);
     DenseTNv la{eo, d.K, Kdim, d.K};
-    Im2colTNvP lb{x, d, Kdim, N};
+    DenseTNv lb{col, Kpad, Kdim, Kpad};  // pad cols are zero-filled
     return launch_gemm_tn2(la, lb, gw, M, N, Kdim, beta, s, bias_out);
   }
   DenseTN la{eo, d.K, Kdim, d.K};

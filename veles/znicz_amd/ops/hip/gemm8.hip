@@ -63,34 +63,51 @@ __global__ __launch_bounds__(O_THREADS, 1) void gemm_nt8_k(
   // ---- staging: one half-tile = 2 glds per 512 threads ------------------
   // physical byte p = lane-linear (tid*16 within the half); the LOGICAL
   // (row, k-chunk) fetched is the de-swizzle of p, so ds_read can use
-  // the swizzled address directly.
-  auto stage_half = [&](int which, int buf, int kt) {
-    // which: 0=A0 1=A1 2=B0 3=B1 (rows half*128)
-    char* base = smem + buf * O_KTILE + which * O_HALF;
-    const int halfrow = (which & 1) * 128;
-    const bool isA = which < 2;
+  // the swizzled address directly.  Per-thread row contexts (4 A + 4 B:
+  // 2 de-swizzled rows × 2 half-rows) are loop-invariant across K-tiles
+  // — the magic-div row decomposition runs once per kernel; per tile
+  // only 2 k-octet decompositions per operand remain.
+  unsigned p_i[2], kc_i[2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    p_i[i] = (unsigned)(i * O_THREADS + tid) * 16u;
+    const unsigned lg = swz(p_i[i]);
+    kc_i[i] = (lg & 127u) >> 4;
+  }
+  typename LA::RowC rca[2][2];       // [halfrow][i]
+  typename LB::RowC rcb[2][2];
+#pragma unroll
+  for (int h = 0; h < 2; ++h)
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
-      const unsigned p = (unsigned)(i * O_THREADS + tid) * 16u;
-      const unsigned lg = swz(p);
-      const int row = (int)(lg >> 7);          // 128-B logical rows
-      const int kc = (int)((lg & 127u) >> 4);  // 16-B chunk (8 k)
-      const int grow = (isA ? m0 : n0) + halfrow + row;
-      const int gk8 = kt * (O_BK / 8) + kc;
-      const void* src = isA
-          ? la.addr(min(grow, M - 1), gk8)
-          : lb.addr(min(grow, N - 1), gk8);
-      __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) void*)src,
-          (__attribute__((address_space(3))) void*)(base + p), 16, 0, 0);
+      const int row = (int)(swz(p_i[i]) >> 7) + h * 128;
+      rca[h][i] = la.rowc(min(m0 + row, M - 1));
+      rcb[h][i] = lb.rowc(min(n0 + row, N - 1));
     }
-  };
 
   auto stage_ktile = [&](int buf, int kt) {
-    stage_half(0, buf, kt);
-    stage_half(1, buf, kt);
-    stage_half(2, buf, kt);
-    stage_half(3, buf, kt);
+    typename LA::KC kca[2];
+    typename LB::KC kcb[2];
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      kca[i] = la.kc(kt * (O_BK / 8) + (int)kc_i[i]);
+      kcb[i] = lb.kc(kt * (O_BK / 8) + (int)kc_i[i]);
+    }
+#pragma unroll
+    for (int which = 0; which < 4; ++which) {
+      char* base = smem + buf * O_KTILE + which * O_HALF;
+      const int h = which & 1;
+      const bool isA = which < 2;
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        const void* src = isA ? la.addr_rk(rca[h][i], kca[i])
+                              : lb.addr_rk(rcb[h][i], kcb[i]);
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) void*)src,
+            (__attribute__((address_space(3))) void*)(base + p_i[i]), 16,
+            0, 0);
+      }
+    }
   };
 
   // ---- fragment reads (swizzled b128) ------------------------------------

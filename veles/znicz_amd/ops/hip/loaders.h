@@ -4,6 +4,13 @@
 #pragma once
 #include "common.h"
 
+// Two-level address API (all fast loaders): the ROW decomposition of a
+// staged chunk is loop-invariant across K-tiles, so the cores compute
+// `RowC rowc(row)` once per thread before the K-loop and per tile only
+// the K side runs: `KC kc(k8)` (shared by every chunk with that
+// k-octet) + `addr_rk(rowc, kc)` (adds + bounds only).  This moved the
+// magic-division work out of the staging hot path entirely.
+
 // dense row-major [rows][ld], K-innermost; k past Klim reads the zero page
 struct DenseNT {
   static constexpr bool MAY_SLOW = false;   // addr() never nullptr
@@ -19,6 +26,16 @@ struct DenseNT {
   // single-element access for the K%8 tail chunk
   __device__ unsigned short elem(int row, int k) const {
     return k < Klim ? base[(long)row * ld + k] : (unsigned short)0;
+  }
+  struct RowC { const unsigned short* rowbase; };
+  struct KC { int k; };
+  __device__ RowC rowc(int row) const {
+    return RowC{base + (long)row * ld};
+  }
+  __device__ KC kc(int k8) const { return KC{k8 * 8}; }
+  __device__ const void* addr_rk(const RowC& rc, const KC& c) const {
+    return c.k < Klim ? (const void*)(rc.rowbase + c.k)
+                      : (const void*)zero;
   }
 };
 
@@ -46,6 +63,31 @@ struct Im2colNT {
         (unsigned)iw >= (unsigned)d.W)
       return zero;
     return x + (((long)n * d.H + ih) * d.W + iw) * d.C + c;
+  }
+  struct RowC { long off0; int ih0, iw0; };
+  struct KC { int c, kw, kh; };
+  __device__ RowC rowc(int row) const {
+    unsigned t2 = divq((unsigned)row, d.mOW);
+    unsigned ow = (unsigned)row - t2 * d.OW;
+    unsigned n = divq(t2, d.mOH);
+    unsigned oh = t2 - n * d.OH;
+    int ih0 = (int)oh * d.SY - d.PT;
+    int iw0 = (int)ow * d.SX - d.PL;
+    return RowC{(((long)n * d.H + ih0) * d.W + iw0) * d.C, ih0, iw0};
+  }
+  __device__ KC kc(int k8) const {
+    unsigned k = (unsigned)k8 * 8;
+    unsigned t = divq(k, d.mC);
+    unsigned c = k - t * d.C;
+    unsigned kh = divq(t, d.mKW);
+    unsigned kw = t - kh * d.KW;
+    return KC{(int)c, (int)kw, (int)kh};
+  }
+  __device__ const void* addr_rk(const RowC& rc, const KC& c) const {
+    if (c.kh >= d.KH || (unsigned)(rc.ih0 + c.kh) >= (unsigned)d.H ||
+        (unsigned)(rc.iw0 + c.kw) >= (unsigned)d.W)
+      return zero;
+    return x + rc.off0 + ((long)c.kh * d.W + c.kw) * d.C + c.c;
   }
   __device__ unsigned short elem(int row, int k) const {
     // only hit when K%8 != 0 (conv K-dims are C-padded so never here)
@@ -110,6 +152,16 @@ struct Im2colNTP {
     return x[(((long)n * d.H + ih) * d.W + iw) * d.C + c];
   }
 
+  // passthrough two-level API (cores only take it on !MAY_SLOW paths,
+  // but the templates must compile for every loader)
+  struct RowC { int row; };
+  struct KC { int k8; };
+  __device__ RowC rowc(int row) const { return RowC{row}; }
+  __device__ KC kc(int k8) const { return KC{k8}; }
+  __device__ const void* addr_rk(const RowC& rc, const KC& c) const {
+    return addr(rc.row, c.k8);
+  }
+
   // refused-chunk path: a kh-crossing octet is exactly two contiguous
   // 8-B-aligned runs (C=4: 4+4 elements) — two vector loads instead of
   // eight guarded scalars (per-element vmcnt serialization measured
@@ -167,6 +219,34 @@ struct DgradNT {
     if ((int)oh >= d.OH || (int)ow >= d.OW) return zero;
     return eo + (((long)n * d.OH + oh) * d.OW + ow) * d.K + co;
   }
+  struct RowC { long nbase; int oh0, ow0; };
+  struct KC { int co, kw, kh; };
+  __device__ RowC rowc(int row) const {
+    unsigned t2 = divq((unsigned)row, d.mW);
+    unsigned iw = (unsigned)row - t2 * d.W;
+    unsigned n = divq(t2, d.mH);
+    unsigned ih = t2 - n * d.H;
+    return RowC{(long)n * d.OH * d.OW * d.K, (int)ih + d.PT,
+                (int)iw + d.PL};
+  }
+  __device__ KC kc(int k8) const {
+    unsigned k = (unsigned)k8 * 8;
+    unsigned t = divq(k, d.mK);
+    unsigned co = k - t * d.K;
+    unsigned kh = divq(t, d.mKW);
+    unsigned kw = t - kh * d.KW;
+    return KC{(int)co, (int)kw, (int)kh};
+  }
+  __device__ const void* addr_rk(const RowC& rc, const KC& c) const {
+    int ohn = rc.oh0 - c.kh, own = rc.ow0 - c.kw;
+    unsigned oh = divq((unsigned)ohn, d.mSY);
+    unsigned ow = divq((unsigned)own, d.mSX);
+    if (c.kh >= d.KH || ohn < 0 || own < 0 ||
+        ohn != (int)(oh * d.SY) || own != (int)(ow * d.SX) ||
+        (int)oh >= d.OH || (int)ow >= d.OW)
+      return zero;
+    return eo + rc.nbase + ((long)oh * d.OW + ow) * d.K + c.co;
+  }
   __device__ unsigned short elem(int row, int k) const {
     int co = k % d.K, t = k / d.K;
     int kw = t % d.KW, kh = t / d.KW;
@@ -206,6 +286,30 @@ struct DgradNT1 {
         (unsigned)ow >= (unsigned)d.OW)
       return zero;
     return eo + (((long)n * d.OH + oh) * d.OW + ow) * d.K + co;
+  }
+  struct RowC { long off0; int oh0, ow0; };
+  struct KC { int co, kw, kh; };
+  __device__ RowC rowc(int row) const {
+    unsigned t2 = divq((unsigned)row, d.mW);
+    unsigned iw = (unsigned)row - t2 * d.W;
+    unsigned n = divq(t2, d.mH);
+    unsigned ih = t2 - n * d.H;
+    int oh0 = (int)ih + d.PT, ow0 = (int)iw + d.PL;
+    return RowC{(((long)n * d.OH + oh0) * d.OW + ow0) * d.K, oh0, ow0};
+  }
+  __device__ KC kc(int k8) const {
+    unsigned k = (unsigned)k8 * 8;
+    unsigned t = divq(k, d.mK);
+    unsigned co = k - t * d.K;
+    unsigned kh = divq(t, d.mKW);
+    unsigned kw = t - kh * d.KW;
+    return KC{(int)co, (int)kw, (int)kh};
+  }
+  __device__ const void* addr_rk(const RowC& rc, const KC& c) const {
+    if (c.kh >= d.KH || (unsigned)(rc.oh0 - c.kh) >= (unsigned)d.OH ||
+        (unsigned)(rc.ow0 - c.kw) >= (unsigned)d.OW)
+      return zero;
+    return eo + rc.off0 - ((long)c.kh * d.OW + c.kw) * d.K + c.co;
   }
   __device__ unsigned short elem(int row, int k) const {
     int co = k % d.K, t = k / d.K;
