@@ -1,0 +1,187 @@
+"""Real-RCCL data-parallel tests.
+
+The CPU lane (test_parallel.py) proves the protocol over gloo; this
+lane runs the PRODUCTION backend — "nccl" (= RCCL on ROCm).
+
+RCCL REFUSES two ranks on one device ("Duplicate GPU detected", NCCL
+2.26 — measured here, not assumed), so the world=2 replica-equality
+test requires >= 2 visible GPUs and skips on the usual 1-GPU test box
+(the driver's 8-GPU scaling run exercises it for real).  What CAN run
+on one GPU and does: a world=1 RCCL process group driving real
+ncclAllReduce calls on the reducer's comm stream (API + stream
+interplay + teardown), which is exactly the plane bench.py uses.
+"""
+
+import os
+import pickle
+import socket
+import tempfile
+
+import numpy
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+torch = pytest.importorskip("torch")
+if not torch.cuda.is_available():
+    pytest.skip("no GPU", allow_module_level=True)
+
+import torch.multiprocessing as mp  # noqa: E402
+
+
+def _free_port():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+def _train_with_ctx(ctx, collect_weights=False):
+    """Shared body: small CIFAR CNN with the GradientReducer over ctx."""
+    from veles import prng
+    prng.seed_all(4242)                 # same seed: identical init
+    from veles.backends import ROCmDevice
+    from veles.config import root
+    from veles.dummy import DummyLauncher
+    from veles.loader.base import TRAIN
+    from veles.znicz_amd.parallel import GradientReducer
+    from veles.znicz_amd.models.cifar import CifarWorkflow
+    root.cifar.loader.update({"minibatch_size": 64, "n_train": 256,
+                              "n_valid": 0, "on_device": True,
+                              "device_dtype": "bf16", "seed": 31337})
+    root.cifar.decision.update({"max_epochs": 2,
+                                "fail_iterations": 10 ** 9})
+    wf = CifarWorkflow(DummyLauncher())
+    wf.dist = ctx
+    wf.loader.shard(ctx.rank, ctx.world)
+    dev = ROCmDevice(ctx.rank % max(1, torch.cuda.device_count()))
+    wf.initialize(device=dev)
+    wf.grad_reducer = GradientReducer(ctx, dev)
+    loader, ev, dec = wf.loader, wf.evaluator, wf.decision
+    gds_rev = [g for g in reversed(wf.gds)]
+    steps = 0
+    while not bool(dec.complete) and steps < 64:
+        loader.run()
+        for f in wf.forwards:
+            f.run()
+        ev.run()
+        dec.run()
+        if not bool(dec.complete):
+            for g in gds_rev:
+                g.run()
+            wf.comm_sync.run()
+        steps += 1
+    torch.cuda.synchronize()
+    out = {"backend": ctx.backend, "steps": steps,
+           "train_err_pt": dec.epoch_n_err_pt[TRAIN]}
+    if collect_weights:
+        ws = []
+        for f in wf.forwards:
+            w = getattr(f, "weights", None)
+            if w is not None and w:
+                w.map_read()
+                ws.append(w.mem.copy())
+        out["weights"] = ws
+    return out
+
+
+def _worker(rank, world, port, out_dir, backend="nccl"):
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+    torch.cuda.set_device(rank % max(1, torch.cuda.device_count()))
+    from veles.znicz_amd.parallel import DistributedContext
+    ctx = DistributedContext.create(backend=backend)
+    result = _train_with_ctx(ctx, collect_weights=True)
+    with open(os.path.join(out_dir, "rank%d.pkl" % rank), "wb") as f:
+        pickle.dump(result, f)
+    ctx.finalize()
+
+
+@pytest.mark.timeout(300)
+def test_nccl_world1_collectives_and_reducer():
+    """World=1 RCCL: a real nccl process group, explicit ncclAllReduce
+    on the comm stream, the GradientReducer's comm-stream fused-SGD
+    machinery, and clean teardown."""
+    import torch.distributed as dist
+    os.environ["RANK"] = "0"
+    os.environ["WORLD_SIZE"] = "1"
+    os.environ["LOCAL_RANK"] = "0"
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(_free_port())
+    from veles.backends import ROCmDevice
+    from veles.znicz_amd.parallel import DistributedContext
+    ctx = DistributedContext.create(backend="nccl")
+    assert ctx.backend == "nccl" and ctx.world == 1
+    dev = ROCmDevice(0)
+    # real collectives through RCCL (ctx.all_reduce_ shortcuts world=1,
+    # so drive the backend directly), on the comm stream as the
+    # reducer does
+    t = torch.full((1 << 20,), 2.0, device="cuda")
+    with torch.cuda.stream(dev.comm_stream):
+        dist.all_reduce(t)
+        s = torch.tensor([3.0], device="cuda")
+        dist.all_reduce(s)
+    torch.cuda.synchronize()
+    assert float(t[12345].item()) == 2.0 and float(s.item()) == 3.0
+    res = _train_with_ctx(ctx)
+    assert res["train_err_pt"] < 95.0
+    ctx.finalize()
+
+
+def _check_world2(res, backend):
+    assert res[0]["backend"] == res[1]["backend"] == backend
+    assert res[0]["steps"] == res[1]["steps"] > 4
+    assert res[0]["train_err_pt"] == pytest.approx(
+        res[1]["train_err_pt"], abs=1e-9)
+    assert res[0]["train_err_pt"] < 95.0
+    assert len(res[0]["weights"]) == len(res[1]["weights"]) > 0
+    for w0, w1 in zip(res[0]["weights"], res[1]["weights"]):
+        numpy.testing.assert_array_equal(w0, w1)
+
+
+@pytest.mark.timeout(420)
+def test_gloo_world2_gpu_flat_buckets():
+    """Two GPU ranks on one device over gloo (host-copy reduction):
+    exercises the reducer's persistent FLAT bucket — gradient tensors
+    reparented into one buffer, collective on the whole buffer, fused
+    SGD from the views — end-to-end on device memory."""
+    port = _free_port()
+    out_dir = tempfile.mkdtemp()
+    mp.spawn(_worker, args=(2, port, out_dir, "gloo"), nprocs=2,
+             join=True)
+    res = []
+    for r in range(2):
+        with open(os.path.join(out_dir, "rank%d.pkl" % r), "rb") as f:
+            res.append(pickle.load(f))
+    _check_world2(res, "gloo")
+
+
+@pytest.mark.timeout(420)
+def test_nccl_world2_training():
+    if torch.cuda.device_count() < 2:
+        pytest.skip("RCCL refuses 2 ranks on one device (Duplicate GPU "
+                    "detected); needs >= 2 GPUs")
+    port = _free_port()
+    out_dir = tempfile.mkdtemp()
+    mp.spawn(_worker, args=(2, port, out_dir), nprocs=2, join=True)
+    res = []
+    for r in range(2):
+        with open(os.path.join(out_dir, "rank%d.pkl" % r), "rb") as f:
+            res.append(pickle.load(f))
+    assert res[0]["backend"] == res[1]["backend"] == "nccl"
+    assert res[0]["steps"] == res[1]["steps"] > 4
+    # global Decision metrics agree
+    assert res[0]["train_err_pt"] == pytest.approx(
+        res[1]["train_err_pt"], abs=1e-9)
+    # training actually learned something
+    assert res[0]["train_err_pt"] < 95.0
+    # replicas stay bitwise-identical: the all-reduced gradients and the
+    # fused SGD update are the same on every rank
+    assert len(res[0]["weights"]) == len(res[1]["weights"]) > 0
+    for w0, w1 in zip(res[0]["weights"], res[1]["weights"]):
+        numpy.testing.assert_array_equal(w0, w1)

@@ -51,7 +51,10 @@ class DistributedContext(object):
             os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
             os.environ.setdefault("MASTER_PORT", "29500")
             if backend == "nccl":
-                torch.cuda.set_device(local_rank)
+                # wrap at the device count: the correctness test lane
+                # runs world=2 with both ranks sharing one MI355X
+                torch.cuda.set_device(
+                    local_rank % max(1, torch.cuda.device_count()))
             dist.init_process_group(backend=backend, rank=rank,
                                     world_size=world)
         return cls(rank, world, local_rank, backend)
