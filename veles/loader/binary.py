@@ -1,0 +1,172 @@
+"""Binary dataset ingestion — MNIST idx and CIFAR-10 bin formats.
+
+Parity target: the reference samples read MNIST's idx-ubyte files and
+CIFAR's python pickles directly from disk (``znicz/samples/MNIST``,
+``znicz/samples/CIFAR10`` loader code, ``veles/loader/pickles.py``).
+This module covers the two classic binary formats end-to-end so a real
+dataset dropped on disk trains without conversion:
+
+* idx (``MnistIdxLoader``): the idx3-ubyte image / idx1-ubyte label
+  format (optionally gzip-compressed, as distributed);
+* CIFAR bin (``CifarBinLoader``): record-per-sample binaries — 1 label
+  byte + 3072 CHW bytes per record.
+
+No hdf5 path: h5py is not in this environment (and the reference's
+hdf5 loader was a low-confidence sighting); idx/bin/pickles cover the
+formats its samples actually used.
+"""
+
+import gzip
+import os
+import struct
+
+import numpy
+
+from veles.loader.fullbatch import FullBatchLoader
+from veles.loader.image import pad_channels_to8
+
+
+def _open_maybe_gz(path):
+    return gzip.open(path, "rb") if path.endswith(".gz") else \
+        open(path, "rb")
+
+
+def read_idx(path):
+    """Read an idx-format array (images: uint8 [N,H,W]; labels: [N])."""
+    with _open_maybe_gz(path) as f:
+        magic = struct.unpack(">I", f.read(4))[0]
+        dtype_code = (magic >> 8) & 0xFF
+        ndim = magic & 0xFF
+        dtypes = {0x08: numpy.uint8, 0x09: numpy.int8,
+                  0x0B: numpy.int16, 0x0C: numpy.int32,
+                  0x0D: numpy.float32, 0x0E: numpy.float64}
+        if dtype_code not in dtypes or (magic >> 16) != 0:
+            raise ValueError("%s: not an idx file (magic 0x%08x)" %
+                             (path, magic))
+        shape = struct.unpack(">" + "I" * ndim, f.read(4 * ndim))
+        dt = numpy.dtype(dtypes[dtype_code]).newbyteorder(">")  # idx is BE
+        data = numpy.frombuffer(f.read(), dtype=dt)
+        return data.astype(dtypes[dtype_code]).reshape(shape)
+
+
+def write_idx(path, arr):
+    """Inverse of read_idx (used by tests and dataset tooling)."""
+    codes = {numpy.dtype(numpy.uint8): 0x08,
+             numpy.dtype(numpy.int8): 0x09,
+             numpy.dtype(numpy.int32): 0x0C,
+             numpy.dtype(numpy.float32): 0x0D}
+    code = codes[arr.dtype]
+    opener = gzip.open if path.endswith(".gz") else open
+    with opener(path, "wb") as f:
+        f.write(struct.pack(">I", (code << 8) | arr.ndim))
+        f.write(struct.pack(">" + "I" * arr.ndim, *arr.shape))
+        a = arr
+        if arr.dtype.itemsize > 1:
+            a = arr.astype(arr.dtype.newbyteorder(">"))
+        f.write(a.tobytes())
+
+
+class MnistIdxLoader(FullBatchLoader):
+    """MNIST-style idx files → full-batch dataset.
+
+    kwargs: ``train_images``/``train_labels`` (paths, .gz ok), optional
+    ``validation_images``/``validation_labels``; pixel range is mapped
+    to ``normalize_range`` and images become NHWC with C padded to 8
+    (``pad_channels``) for the HIP conv path — or stay flat [N, H*W]
+    with ``flatten=True`` for MLPs.
+    """
+
+    def __init__(self, workflow, **kwargs):
+        super(MnistIdxLoader, self).__init__(workflow, **kwargs)
+        self.train_images = kwargs["train_images"]
+        self.train_labels = kwargs["train_labels"]
+        self.validation_images = kwargs.get("validation_images")
+        self.validation_labels = kwargs.get("validation_labels")
+        self.normalize_range = kwargs.get("normalize_range", (-1.0, 1.0))
+        self.flatten = kwargs.get("flatten", True)
+        self.pad_channels = kwargs.get("pad_channels", True)
+
+    def _load_set(self, images_path, labels_path):
+        imgs = read_idx(images_path)
+        labels = read_idx(labels_path).astype(numpy.int32)
+        if imgs.shape[0] != labels.shape[0]:
+            raise ValueError("idx image/label count mismatch: %d vs %d" %
+                             (imgs.shape[0], labels.shape[0]))
+        lo, hi = self.normalize_range
+        data = imgs.astype(numpy.float32) / 255.0 * (hi - lo) + lo
+        if self.flatten:
+            data = data.reshape(data.shape[0], -1)
+        else:
+            data = data[..., None]          # [N,H,W,1]
+            if self.pad_channels:
+                n, h, w, _ = data.shape
+                out = numpy.zeros((n, h, w, 8), numpy.float32)
+                out[..., :1] = data
+                data = out
+        return data, labels
+
+    def load_data(self):
+        tr_d, tr_l = self._load_set(self.train_images, self.train_labels)
+        va_d = va_l = None
+        if self.validation_images:
+            va_d, va_l = self._load_set(self.validation_images,
+                                        self.validation_labels)
+        if va_d is not None:
+            self.original_data.mem = numpy.concatenate([va_d, tr_d])
+            self.original_labels.mem = numpy.concatenate([va_l, tr_l])
+            self.class_lengths = [0, len(va_d), len(tr_d)]
+        else:
+            self.original_data.mem = tr_d
+            self.original_labels.mem = tr_l
+            self.class_lengths = [0, 0, len(tr_d)]
+
+
+class CifarBinLoader(FullBatchLoader):
+    """CIFAR-10 binary batches → full-batch NHWC dataset.
+
+    kwargs: ``train_files`` (list of data_batch_*.bin paths),
+    optional ``validation_files`` (e.g. [test_batch.bin]).
+    Records: 1 label byte + 32*32*3 CHW bytes.
+    """
+
+    RECORD = 1 + 3 * 32 * 32
+
+    def __init__(self, workflow, **kwargs):
+        super(CifarBinLoader, self).__init__(workflow, **kwargs)
+        self.train_files = kwargs["train_files"]
+        self.validation_files = kwargs.get("validation_files", [])
+        self.normalize_range = kwargs.get("normalize_range", (-1.0, 1.0))
+        self.pad_channels = kwargs.get("pad_channels", True)
+
+    def _load_files(self, paths):
+        datas, labels = [], []
+        for p in paths:
+            raw = numpy.fromfile(p, dtype=numpy.uint8)
+            if raw.size % self.RECORD:
+                raise ValueError("%s: size %d is not a whole number of "
+                                 "CIFAR records" % (p, raw.size))
+            rec = raw.reshape(-1, self.RECORD)
+            labels.append(rec[:, 0].astype(numpy.int32))
+            imgs = rec[:, 1:].reshape(-1, 3, 32, 32)       # CHW
+            imgs = imgs.transpose(0, 2, 3, 1)              # -> NHWC
+            lo, hi = self.normalize_range
+            d = imgs.astype(numpy.float32) / 255.0 * (hi - lo) + lo
+            if self.pad_channels:
+                n, h, w, c = d.shape
+                out = numpy.zeros((n, h, w, 8), numpy.float32)
+                out[..., :c] = d
+                d = out
+            datas.append(d)
+        return numpy.concatenate(datas), numpy.concatenate(labels)
+
+    def load_data(self):
+        tr_d, tr_l = self._load_files(self.train_files)
+        if self.validation_files:
+            va_d, va_l = self._load_files(self.validation_files)
+            self.original_data.mem = numpy.concatenate([va_d, tr_d])
+            self.original_labels.mem = numpy.concatenate([va_l, tr_l])
+            self.class_lengths = [0, len(va_d), len(tr_d)]
+        else:
+            self.original_data.mem = tr_d
+            self.original_labels.mem = tr_l
+            self.class_lengths = [0, 0, len(tr_d)]

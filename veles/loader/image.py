@@ -16,6 +16,7 @@ import numpy
 
 from veles.loader.base import Loader
 from veles.loader.fullbatch import FullBatchLoader
+from veles.loader.fullbatch import FullBatchLoaderMSE as _FBMSE
 
 
 def pad_channels_to8(img):
@@ -35,7 +36,10 @@ class ImageLoader(Loader):
     def __init__(self, workflow, **kwargs):
         super(ImageLoader, self).__init__(workflow, **kwargs)
         self.scale = kwargs.get("scale", None)        # (w, h) or float
-        self.crop = kwargs.get("crop", None)          # (w, h) center crop
+        self.crop = kwargs.get("crop", None)          # (w, h) crop size
+        #: "center" (deterministic) or "random" (augmentation; uses the
+        #: loader's seeded rng so runs stay reproducible)
+        self.crop_mode = kwargs.get("crop_mode", "center")
         self.mirror = kwargs.get("mirror", False)     # random h-flip
         self.grayscale = kwargs.get("grayscale", False)
         self.normalize_range = kwargs.get("normalize_range", (-1.0, 1.0))
@@ -60,7 +64,11 @@ class ImageLoader(Loader):
         if self.crop is not None:
             cw, ch = self.crop
             w, h = img.size
-            x0, y0 = (w - cw) // 2, (h - ch) // 2
+            if self.crop_mode == "random" and rng is not None:
+                x0 = rng.randint(0, max(1, w - cw + 1))
+                y0 = rng.randint(0, max(1, h - ch + 1))
+            else:
+                x0, y0 = (w - cw) // 2, (h - ch) // 2
             img = img.crop((x0, y0, x0 + cw, y0 + ch))
         arr = numpy.asarray(img, numpy.float32)
         if arr.ndim == 2:
@@ -129,3 +137,52 @@ class FileImageLoader(ImageLoader, FullBatchLoader):
 
 
 FullBatchImageLoader = FileImageLoader  # reference alias
+
+
+class FileImageLoaderMSE(FileImageLoader, _FBMSE):
+    """Input-image / target-image dataset pairs for MSE training
+    (reference ``fullbatch_image.py`` + ``image_mse.py``: denoising /
+    reconstruction autoencoders train against a TARGET image per
+    sample, not a class label).
+
+    ``target_paths`` mirrors ``*_paths``: one glob list whose files
+    pair with the input files by basename (the reference paired by
+    file order within matching directory trees; basename pairing is
+    the robust version of that contract).  Targets go through the same
+    scale/crop pipeline (deterministic: no mirror/random-crop skew
+    between input and target — augmentation is applied to neither).
+    """
+
+    def __init__(self, workflow, **kwargs):
+        super(FileImageLoaderMSE, self).__init__(workflow, **kwargs)
+        self.target_paths = kwargs.get("target_paths", [])
+
+    def load_data(self):
+        # inputs first (fills original_data + class_lengths)
+        saved_mirror, self.mirror = self.mirror, False
+        saved_mode, self.crop_mode = self.crop_mode, "center"
+        try:
+            super(FileImageLoaderMSE, self).load_data()
+            tfiles = self._expand(self.target_paths)
+            if not tfiles:
+                raise ValueError("FileImageLoaderMSE: no target files")
+            by_base = {os.path.basename(p): p for p in tfiles}
+            sets = [self._expand(p) for p in (self.test_paths,
+                                              self.validation_paths,
+                                              self.train_paths)]
+            order = [p for s in sets for p in s]
+            first = self.load_image(tfiles[0])
+            targets = numpy.zeros((len(order),) + first.shape,
+                                  numpy.float32)
+            for i, path in enumerate(order):
+                base = os.path.basename(path)
+                if base not in by_base:
+                    raise ValueError(
+                        "no target image pairs with input %r" % path)
+                targets[i] = self.load_image(by_base[base])
+            self.original_targets.mem = targets
+        finally:
+            self.mirror = saved_mirror
+            self.crop_mode = saved_mode
+    # minibatch_targets machinery (create/initialize/fill) comes from
+    # FullBatchLoaderMSE via the MRO

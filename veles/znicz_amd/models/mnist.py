@@ -35,9 +35,25 @@ class MnistWorkflow(StandardWorkflow):
                     "weights_decay": cfg.weights_decay}},
         ]
         kwargs.setdefault("layers", layers)
-        kwargs.setdefault("loader_factory",
-                          lambda wf, **kw: SyntheticClassLoader(wf, **kw))
-        kwargs.setdefault("loader_config", cfg.loader.to_dict())
+        # real-file ingestion: point root.mnist.loader.idx_files at the
+        # idx-ubyte files (train_images/train_labels[/validation_*],
+        # .gz accepted) and the sample trains on them; otherwise it
+        # runs on the synthetic stand-in (BASELINE: no network).
+        loader_cfg = cfg.loader.to_dict()
+        idx_files = loader_cfg.pop("idx_files", None)
+        if idx_files:
+            from veles.loader.binary import MnistIdxLoader
+            for k in ("sample_shape", "n_train", "n_valid", "n_classes"):
+                loader_cfg.pop(k, None)
+            loader_cfg.update(idx_files)
+            kwargs.setdefault(
+                "loader_factory",
+                lambda wf, **kw: MnistIdxLoader(wf, **kw))
+        else:
+            kwargs.setdefault(
+                "loader_factory",
+                lambda wf, **kw: SyntheticClassLoader(wf, **kw))
+        kwargs.setdefault("loader_config", loader_cfg)
         kwargs.setdefault("decision_config", cfg.decision.to_dict())
         kwargs.setdefault("snapshotter_config", cfg.snapshotter.to_dict())
         kwargs.setdefault("loss_function", "softmax")
