@@ -366,3 +366,47 @@ def test_cifar_cnn_trains_on_gpu():
     # prototype data with 10 classes: must be well below chance (90%)
     assert wf.decision.epoch_n_err_pt[TRAIN] < 60.0, \
         wf.decision.epoch_n_err_pt
+
+
+def test_histogram_kernel_matches_numpy():
+    """LDS histogram vs numpy.histogram, bf16 and f32, accumulation."""
+    import torch
+    from veles.znicz_amd import ops
+    rng = numpy.random.RandomState(17)
+    host = rng.normal(0, 2.0, 100000).astype(numpy.float32)
+    lo, hi = -3.0, 3.0
+    want, _ = numpy.histogram(host, bins=32, range=(lo, hi))
+    # f32
+    x = torch.from_numpy(host).cuda()
+    counts = torch.zeros(32, dtype=torch.int32, device="cuda")
+    ops.histogram(x, counts, lo, hi)
+    numpy.testing.assert_array_equal(counts.cpu().numpy(), want)
+    # bf16 (bins computed from the bf16-rounded values)
+    xb = x.bfloat16()
+    wantb, _ = numpy.histogram(xb.float().cpu().numpy(), bins=32,
+                               range=(lo, hi))
+    counts.zero_()
+    ops.histogram(xb, counts, lo, hi)
+    numpy.testing.assert_array_equal(counts.cpu().numpy(), wantb)
+    # accumulation: second call adds
+    ops.histogram(xb, counts, lo, hi)
+    numpy.testing.assert_array_equal(counts.cpu().numpy(), 2 * wantb)
+
+
+def test_multi_histogram_unit_device_path():
+    import torch
+    from veles.dummy import DummyWorkflow
+    from veles.memory import Array
+    from veles.znicz_amd.multi_hist import MultiHistogram
+    from veles.backends import ROCmDevice
+    rng = numpy.random.RandomState(3)
+    w = rng.normal(0, 1, (6, 500)).astype(numpy.float32)
+    unit = MultiHistogram(DummyWorkflow(), hist_number=6, n_bars=16)
+    unit.input = Array(w.copy())
+    unit.initialize(device=ROCmDevice(0))
+    unit.input.initialize(ROCmDevice(0))
+    unit.run()
+    for i in range(6):
+        lo, hi = unit.limits.mem[i]
+        want, _ = numpy.histogram(w[i], bins=16, range=(lo, hi))
+        numpy.testing.assert_array_equal(unit.output.mem[i], want)

@@ -32,6 +32,20 @@ class RangeAccumulator(Unit):
         self.y_out.mem = numpy.zeros(self.bars, numpy.float64)
 
     def run(self):
+        if self.input.devmem is not None and self.input.on_gpu:
+            # device path: running range via
+            # aminmax, samples buffered ON DEVICE, histogrammed by the
+            # LDS hist kernel at flush with the final range — same
+            # semantics as the host path, no full-tensor downloads
+            import torch
+            v = self.input.devmem.reshape(-1)
+            mn, mx = torch.aminmax(v)
+            self.min_value = min(self.min_value, float(mn))
+            self.max_value = max(self.max_value, float(mx))
+            self._samples.append(v.detach().clone())
+            if len(self._samples) > 64:
+                self._flush()
+            return
         self.input.map_read()
         v = self.input.mem.reshape(-1)
         self.min_value = min(self.min_value, float(v.min()))
@@ -43,11 +57,22 @@ class RangeAccumulator(Unit):
     def _flush(self):
         if not self._samples:
             return
-        allv = numpy.concatenate(self._samples)
-        hist, edges = numpy.histogram(
-            allv, bins=self.bars, range=(self.min_value,
-                                         self.max_value or 1.0))
-        self.y_out.mem += hist
+        lo = self.min_value
+        hi = self.max_value if self.max_value > lo else lo + 1e-8
+        if isinstance(self._samples[0], numpy.ndarray):
+            allv = numpy.concatenate(self._samples)
+            hist, edges = numpy.histogram(allv, bins=self.bars,
+                                          range=(lo, hi))
+            self.y_out.mem += hist
+        else:
+            import torch
+            from veles.znicz_amd import ops
+            counts = torch.zeros(self.bars, dtype=torch.int32,
+                                 device=self._samples[0].device)
+            for s in self._samples:
+                ops.histogram(s, counts, lo, hi)
+            self.y_out.mem += counts.cpu().numpy().astype(numpy.float64)
+            edges = numpy.linspace(lo, hi, self.bars + 1)
         self.x_out.mem[...] = 0.5 * (edges[:-1] + edges[1:])
         self._samples = []
 

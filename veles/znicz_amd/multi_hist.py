@@ -28,6 +28,25 @@ class MultiHistogram(Unit):
                                       numpy.float64)
 
     def run(self):
+        if self.input.devmem is not None and self.input.on_gpu:
+            import torch
+            from veles.znicz_amd import ops
+            w = self.input.devmem
+            rows = min(self.hist_number, w.shape[0])
+            counts = torch.zeros((rows, self.n_bars),
+                                 dtype=torch.int32, device=w.device)
+            lims = []
+            for i in range(rows):
+                v = w[i].reshape(-1).contiguous()
+                mn, mx = torch.aminmax(v)
+                lo, hi = float(mn), float(mx)
+                if lo == hi:
+                    hi = lo + 1e-8
+                ops.histogram(v, counts[i], lo, hi)
+                lims.append((lo, hi))
+            self.output.mem[:rows] = counts.cpu().numpy()
+            self.limits.mem[:rows] = lims
+            return
         self.input.map_read()
         w = self.input.mem
         rows = min(self.hist_number, w.shape[0])

@@ -23,9 +23,32 @@ def _load():
         import torch  # noqa: F401  (must be imported before the ext .so)
         from veles.znicz_amd.ops import _mzk  # built in-tree
         _ext = _mzk
+        _apply_routing(_ext)
     except ImportError as e:
         _ext_err = e
     return _ext
+
+
+def _apply_routing(mod):
+    """Load the swept gfx950 routing table (SURVEY §2b autotune-DB row)
+    and override the kernel launchers' built-in heuristics.  Regenerate
+    with tools/gen_routing.py on an MI355X; absent file = defaults."""
+    import json
+    import os
+    path = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                        "gfx950_routing.json")
+    if not os.path.exists(path):
+        return
+    try:
+        with open(path) as f:
+            r = json.load(f)
+        mod.set_routing(int(r.get("nt_split_budget", -1)),
+                        int(r.get("tn_split_budget", -1)),
+                        int(r.get("tn2_split_budget", -1)),
+                        int(r.get("ntn_max_n", -1)))
+    except Exception as e:                       # never block import
+        import logging
+        logging.getLogger("ops").warning("routing table ignored: %r", e)
 
 
 def ext():
@@ -253,6 +276,12 @@ def eval_softmax(probs, labels, err_out, n_err, batch):
 def confusion_accum(max_idx, labels, cm, batch):
     """cm[pred, label] += 1 on device (int32) — no host sync."""
     require().confusion_accum(max_idx, labels, cm, int(batch))
+
+
+def histogram(x, counts, lo, hi):
+    """counts[nbins] (int32, device) += histogram of x over [lo, hi]
+    with numpy.histogram edge semantics (LDS hist kernel)."""
+    require().histogram(x, counts, float(lo), float(hi))
 
 
 def eval_mse(y, target, err_out, mse, metrics, batch, root):

@@ -53,6 +53,9 @@ void dropout_fwd_launch(const unsigned short*, unsigned short*,
                         unsigned long long*, hipStream_t);
 void dropout_bwd_launch(unsigned short*, const unsigned char*, long, float,
                         hipStream_t);
+void histogram_launch(const void*, bool, long, float, float, int,
+                      unsigned int*, hipStream_t);
+void set_routing(int, int, int, int);
 // from softmax.hip
 void softmax_launch(float*, int*, int, int, hipStream_t);
 void eval_softmax_launch(const float*, const int*, unsigned short*, int*,
@@ -385,6 +388,25 @@ void mean_disp(torch::Tensor x, torch::Tensor mean, torch::Tensor rdisp,
                    mean.numel(), cur_stream());
 }
 
+void set_routing_py(int64_t nt_split, int64_t tn_split,
+                    int64_t tn2_split, int64_t ntn_max_n) {
+  set_routing((int)nt_split, (int)tn_split, (int)tn2_split,
+              (int)ntn_max_n);
+}
+
+void histogram(torch::Tensor x, torch::Tensor counts, double lo,
+               double hi) {
+  TORCH_CHECK(x.is_cuda() && counts.is_cuda(), "histogram: cuda only");
+  TORCH_CHECK(counts.dtype() == torch::kInt32, "counts must be int32");
+  TORCH_CHECK(x.is_contiguous() && counts.is_contiguous());
+  const bool bf = x.dtype() == torch::kBFloat16;
+  TORCH_CHECK(bf || x.dtype() == torch::kFloat32,
+              "histogram: bf16 or f32 input");
+  histogram_launch(x.data_ptr(), bf, x.numel(), (float)lo, (float)hi,
+                   (int)counts.numel(),
+                   (unsigned int*)counts.data_ptr<int>(), cur_stream());
+}
+
 void confusion_accum(torch::Tensor max_idx, torch::Tensor labels,
                      torch::Tensor cm, int64_t batch) {
   confusion_launch(max_idx.data_ptr<int>(), labels.data_ptr<int>(),
@@ -425,4 +447,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("lrn_bwd", &lrn_bwd);
   m.def("dropout_fwd", &dropout_fwd);
   m.def("dropout_bwd", &dropout_bwd);
+  m.def("histogram", &histogram, "LDS histogram (counts int32)");
+  m.def("set_routing", &set_routing_py,
+        "override GEMM routing constants (gfx950_routing.json)");
 }

@@ -618,3 +618,52 @@ void dropout_bwd_launch(unsigned short* err, const unsigned char* mask,
                      mask, n, 1.f / (1.f - ratio));
   HIP_CHECK_LAST();
 }
+
+// ---------------------------------------------------------------------------
+// LDS histogram (the reference's ocl/accumulator.cl
+// range/hist kernels) — per-block int32 bins in LDS, one global atomic
+// per (block, bin) flush; bf16 and f32 inputs.
+// ---------------------------------------------------------------------------
+template <typename T>
+__global__ void hist_k(const T* __restrict__ x, long n, float lo,
+                       float hi, float inv_width, int nbins,
+                       unsigned int* __restrict__ out) {
+  extern __shared__ unsigned int bins[];
+  for (int b = threadIdx.x; b < nbins; b += blockDim.x) bins[b] = 0u;
+  __syncthreads();
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long p = i; p < n; p += stride) {
+    float v;
+    if constexpr (sizeof(T) == 2)
+      v = bf2f(((const unsigned short*)x)[p]);
+    else
+      v = ((const float*)x)[p];
+    int b = (int)((v - lo) * inv_width);
+    // numpy.histogram semantics: the right edge is inclusive (lands in
+    // the last bin); out-of-range values are dropped
+    if (v == hi) b = nbins - 1;
+    if (v >= lo && b >= 0 && b < nbins) atomicAdd(&bins[b], 1u);
+  }
+  __syncthreads();
+  for (int b = threadIdx.x; b < nbins; b += blockDim.x)
+    if (bins[b]) atomicAdd(&out[b], bins[b]);
+}
+
+void histogram_launch(const void* x, bool is_bf16, long n, float lo,
+                      float hi, int nbins, unsigned int* out,
+                      hipStream_t s) {
+  if (hi <= lo) hi = lo + 1e-8f;
+  const float inv_width = (float)nbins / (hi - lo);
+  int blocks = (int)min((n + 255) / 256, (long)2048);
+  const size_t lds = (size_t)nbins * 4;
+  if (is_bf16)
+    hipLaunchKernelGGL(hist_k<unsigned short>, dim3(blocks), dim3(256),
+                       lds, s, (const unsigned short*)x, n, lo, hi,
+                       inv_width, nbins, out);
+  else
+    hipLaunchKernelGGL(hist_k<float>, dim3(blocks), dim3(256), lds, s,
+                       (const float*)x, n, lo, hi, inv_width, nbins,
+                       out);
+  HIP_CHECK_LAST();
+}

@@ -1132,6 +1132,26 @@ __global__ __launch_bounds__(256, 2) void gemm_tn2n_k(
 
 static inline int cdiv(int a, int b) { return (a + b - 1) / b; }
 
+// ---------------------------------------------------------------------------
+// tunable routing table (SURVEY §2b autotune-DB row): defaults are the
+// measured round-1/2 heuristics; veles.znicz_amd.ops loads the swept
+// gfx950_routing.json at import and overrides them via set_routing()
+// (regenerate with tools/gen_routing.py on an MI355X).
+// ---------------------------------------------------------------------------
+int g_nt_split_budget = 512;    // FC NT split-K: blocks to fill toward
+int g_tn_split_budget = 1024;   // TN wgrad split-K budget
+int g_tn2_split_budget = 1024;  // TN2 wgrad split-K budget
+int g_ntn_max_n = 96;           // widest narrow-N NT tile routed
+
+void set_routing(int nt_split, int tn_split, int tn2_split,
+                 int ntn_max_n) {
+  if (nt_split > 0) g_nt_split_budget = nt_split;
+  if (tn_split > 0) g_tn_split_budget = tn_split;
+  if (tn2_split > 0) g_tn2_split_budget = tn2_split;
+  if (ntn_max_n >= 0) g_ntn_max_n = ntn_max_n;
+}
+
+
 template <typename LA, typename LB, typename OUT_T>
 void launch_gemm_nt(const LA& la, const LB& lb, OUT_T* C, const float* bias,
                     int M, int N, int K, int act, hipStream_t stream) {
@@ -1159,7 +1179,7 @@ void launch_gemm_nt(const LA& la, const LB& lb, OUT_T* C, const float* bias,
     HIP_CHECK_LAST();
     return;
   }
-  if (N <= 96) {
+  if (N <= 96 && g_ntn_max_n >= 96) {
     // AlexNet conv1 class (K=96 kernels): the 128-wide tile would
     // waste 25% of the MFMA columns
     const int tn = cdiv(N, 96);
@@ -1189,7 +1209,7 @@ void launch_gemm_tn(const LA& la, const LB& lb, float* C, int M, int N,
   const int tn = cdiv(N, BN);
   const int tiles = cdiv(M, BM) * tn;
   const int nk = cdiv(K, BK);
-  const int splits = max(1, min(nk, 1024 / tiles));
+  const int splits = max(1, min(nk, g_tn_split_budget / tiles));
   if (splits > 1) {
     if (beta == 0.f) {
       void zero_f32_launch(float*, long, hipStream_t);
@@ -1219,7 +1239,7 @@ bool launch_gemm_tn2(const LA& la, const LB& lb, float* C, int M, int N,
   if (bmt == BM) bias_out = nullptr;
   const int tiles = cdiv(M, bmt) * tn;
   const int nk = cdiv(K, TBK);
-  const int splits = max(1, min(nk, 1024 / tiles));
+  const int splits = max(1, min(nk, g_tn2_split_budget / tiles));
   if (splits > 1) {
     if (beta == 0.f) {
       void zero_f32_launch(float*, long, hipStream_t);
@@ -1263,7 +1283,7 @@ void gemm_nt_dense(const unsigned short* A, const unsigned short* B,
   DenseNT la{A, zero, K, K}, lb{B, zero, K, K};
   const int tiles = cdiv(M, BM) * cdiv(N, BN);
   const int nk = cdiv(K, BK);
-  const int splits = max(1, min(nk, 512 / tiles));
+  const int splits = max(1, min(nk, g_nt_split_budget / tiles));
   if (splits > 1) {
     // fp32 workspace + atomic partials, then one bias/act/cast pass.
     // ONE process-global workspace: every FC split-K launch in this
