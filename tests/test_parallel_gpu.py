@@ -135,7 +135,7 @@ def test_nccl_world1_collectives_and_reducer():
 
 def _check_world2(res, backend):
     assert res[0]["backend"] == res[1]["backend"] == backend
-    assert res[0]["steps"] == res[1]["steps"] > 4
+    assert res[0]["steps"] == res[1]["steps"] >= 4
     assert res[0]["train_err_pt"] == pytest.approx(
         res[1]["train_err_pt"], abs=1e-9)
     assert res[0]["train_err_pt"] < 95.0
@@ -174,7 +174,7 @@ def test_nccl_world2_training():
         with open(os.path.join(out_dir, "rank%d.pkl" % r), "rb") as f:
             res.append(pickle.load(f))
     assert res[0]["backend"] == res[1]["backend"] == "nccl"
-    assert res[0]["steps"] == res[1]["steps"] > 4
+    assert res[0]["steps"] == res[1]["steps"] >= 4
     # global Decision metrics agree
     assert res[0]["train_err_pt"] == pytest.approx(
         res[1]["train_err_pt"], abs=1e-9)
