@@ -367,26 +367,65 @@ void col_sum_launch(const unsigned short* a, float* out, int M, int N,
 }
 
 // -- bf16 2-D transpose (LDS 32x32 tiles, +1 pad) ---------------------------
+// 64×64 tiles, 16-B vector global access on BOTH sides: each thread
+// vector-loads one short8 chunk, scatters its 8 elements to TRANSPOSED
+// LDS positions (the scalar cost sits on the cheap LDS-write side),
+// then vector-reads the transposed image and stores coalesced.  The
+// old 32×32 scalar-u16 version ran ~2.5 TB/s (2-B store grain).
 __global__ void transpose2d_k(const unsigned short* __restrict__ in,
                               unsigned short* __restrict__ out, int R,
                               int C) {
-  __shared__ unsigned short tile[32][33];
-  int c0 = blockIdx.x * 32, r0 = blockIdx.y * 32;
-  int tx = threadIdx.x & 31, ty = threadIdx.x >> 5;   // 256 thr: 8 rows
-  for (int i = 0; i < 32; i += 8) {
-    int r = r0 + ty + i, c = c0 + tx;
-    tile[ty + i][tx] = (r < R && c < C) ? in[(long)r * C + c] : 0;
+  constexpr int TP = 64 + 8;             // pitch: 16-B aligned, detunes
+  __shared__ unsigned short tt[64 * TP]; // banks across the 8 writes
+  const int c0 = blockIdx.x * 64, r0 = blockIdx.y * 64;
+  const int tx = threadIdx.x & 7;        // chunk (8 cols) within row
+  const int ty = threadIdx.x >> 3;       // row 0..31; two row-halves
+  const bool interior = (r0 + 64 <= R) && (c0 + 64 <= C);
+#pragma unroll
+  for (int h = 0; h < 2; ++h) {
+    const int r = ty + h * 32;           // tile row
+    const int c = tx * 8;                // tile col of this chunk
+    unsigned short v[8];
+    if (interior) {
+      *(short8*)v = *(const short8*)(in + (long)(r0 + r) * C + c0 + c);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int rr = r0 + r, cc = c0 + c + j;
+        v[j] = (rr < R && cc < C) ? in[(long)rr * C + cc]
+                                  : (unsigned short)0;
+      }
+    }
+    // row index XORed with the column-octet spreads the 8 scatter
+    // writes of a wave over 32 distinct banks (unswizzled, every lane
+    // of an instruction hits the same 4-bank set: 16-way conflict)
+    const int swz = (tx & 7) << 3;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      tt[(c + j) * TP + (r ^ swz)] = v[j];
   }
   __syncthreads();
-  for (int i = 0; i < 32; i += 8) {
-    int r = c0 + ty + i, c = r0 + tx;   // transposed coords
-    if (r < C && c < R) out[(long)r * R + c] = tile[tx][ty + i];
+#pragma unroll
+  for (int h = 0; h < 2; ++h) {
+    const int tr = ty + h * 32;          // transposed-tile row (= col c)
+    const int tc = tx * 8;
+    const int swz = ((tr >> 3) & 7) << 3;
+    short8 v = *(const short8*)(tt + tr * TP + (tc ^ swz));
+    const int orow = c0 + tr, ocol = r0 + tc;
+    if (interior) {
+      *(short8*)(out + (long)orow * R + ocol) = v;
+    } else if (orow < C) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        if (ocol + j < R)
+          out[(long)orow * R + ocol + j] = ((unsigned short*)&v)[j];
+    }
   }
 }
 
 void transpose2d_launch(const unsigned short* in, unsigned short* out,
                         int R, int C, hipStream_t s) {
-  dim3 grid((C + 31) / 32, (R + 31) / 32);
+  dim3 grid((C + 63) / 64, (R + 63) / 64);
   hipLaunchKernelGGL(transpose2d_k, grid, dim3(256), 0, s, in, out, R, C);
   HIP_CHECK_LAST();
 }

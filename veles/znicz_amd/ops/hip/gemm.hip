@@ -1179,7 +1179,7 @@ void launch_gemm_nt(const LA& la, const LB& lb, OUT_T* C, const float* bias,
     HIP_CHECK_LAST();
     return;
   }
-  if (N <= 96 && g_ntn_max_n >= 96) {
+  if (N <= 96 && g_ntn_max_n >= 96 && LA::N96_OK) {
     // AlexNet conv1 class (K=96 kernels): the 128-wide tile would
     // waste 25% of the MFMA columns
     const int tn = cdiv(N, 96);

@@ -14,6 +14,7 @@
 // dense row-major [rows][ld], K-innermost; k past Klim reads the zero page
 struct DenseNT {
   static constexpr bool MAY_SLOW = false;   // addr() never nullptr
+  static constexpr bool N96_OK = true;
   const unsigned short* base;
   const unsigned short* zero;
   long ld;
@@ -44,6 +45,7 @@ struct DenseNT {
 // (Im2colNTP below handles C%4).
 struct Im2colNT {
   static constexpr bool MAY_SLOW = false;
+  static constexpr bool N96_OK = true;
   const unsigned short* x;
   const unsigned short* zero;
   ConvDesc d;
@@ -114,6 +116,7 @@ struct Im2colNT {
 // TYPE so the fallback machinery costs the C%8 kernels nothing.
 struct Im2colNTP {
   static constexpr bool MAY_SLOW = true;
+  static constexpr bool N96_OK = true;
   const unsigned short* x;
   const unsigned short* zero;
   ConvDesc d;
@@ -196,6 +199,8 @@ struct Im2colNTP {
 // input, kdim=(kh,kw,co) co-inner; strided positions read the zero page.
 struct DgradNT {
   static constexpr bool MAY_SLOW = false;
+  // measured: the 96-wide narrow tile LOSES ~6% on the dgrad gather
+  static constexpr bool N96_OK = false;
   const unsigned short* eo;
   const unsigned short* zero;
   ConvDesc d;
@@ -267,6 +272,7 @@ struct DgradNT {
 // stride divisibility checks and divisions of the generic DgradNT.
 struct DgradNT1 {
   static constexpr bool MAY_SLOW = false;
+  static constexpr bool N96_OK = false;
   const unsigned short* eo;
   const unsigned short* zero;
   ConvDesc d;
