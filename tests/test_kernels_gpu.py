@@ -410,3 +410,25 @@ def test_multi_histogram_unit_device_path():
         lo, hi = unit.limits.mem[i]
         want, _ = numpy.histogram(w[i], bins=16, range=(lo, hi))
         numpy.testing.assert_array_equal(unit.output.mem[i], want)
+
+
+def test_dgrad_fused_act_derivative():
+    """gemm_nt's dact epilogue == separate GEMM + act_bwd pass."""
+    import torch
+    from veles.znicz_amd import ops
+    from veles.znicz_amd import activation_defs as act
+    torch.manual_seed(5)
+    M, N, K = 192, 96, 128
+    eo = (torch.randn(M, K, device="cuda") * 0.3).bfloat16()
+    w = (torch.randn(N, K, device="cuda") * 0.3).bfloat16()
+    y = torch.rand(M, N, device="cuda").bfloat16()  # fwd output
+    fused = torch.empty(M, N, device="cuda", dtype=torch.bfloat16)
+    ops.gemm_nt(eo, w, fused, activation=act.ACT_TANH, dact=y)
+    # reference: GEMM then the standalone act_bwd kernel
+    ref = torch.empty_like(fused)
+    ops.gemm_nt(eo, w, ref)
+    ops.act_bwd_inplace(ref, y, act.ACT_TANH)
+    torch.cuda.synchronize()
+    numpy.testing.assert_allclose(fused.float().cpu().numpy(),
+                                  ref.float().cpu().numpy(),
+                                  rtol=2e-2, atol=2e-2)

@@ -82,7 +82,10 @@ class GradientDescent(GradientDescentBase):
     def hip_run(self):
         from veles.znicz_amd import ops
         eo = ops.as2d_bf16(self.err_output)
-        if self.ACTIVATION != act.ACT_LINEAR:
+        if self.ACTIVATION != act.ACT_LINEAR and \
+                not getattr(self, "err_output_preactivated", False):
+            # (when the DOWNSTREAM gd fused act' into the dgrad that
+            # produced err_output, this pass is already done)
             ops.act_bwd_inplace(eo, ops.as2d_bf16(self.output),
                                 self.ACTIVATION)
         batch = eo.shape[0]
@@ -91,7 +94,15 @@ class GradientDescent(GradientDescentBase):
             else self.weights.bf16_t
         if self.need_err_input:
             ei = ops.as2d_bf16(self.err_input)
-            ops.gemm_nt(eo, w_dgrad, ei)
+            in_act = getattr(self, "input_activation", None)
+            if in_act:
+                # fuse the PREVIOUS layer's act' into this dgrad's
+                # epilogue: ei = (eo·Wᵀ) ∘ act'(input) — saves the
+                # upstream gd's full read+write act_bwd pass
+                ops.gemm_nt(eo, w_dgrad, ei, activation=in_act,
+                            dact=ops.as2d_bf16(self.input))
+            else:
+                ops.gemm_nt(eo, w_dgrad, ei)
             self.err_input.touch_device()
         x = ops.as2d_bf16(self.input)
         beta = 1.0 if self.accumulate_gradient else 0.0

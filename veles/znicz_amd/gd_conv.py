@@ -110,14 +110,18 @@ class GradientDescentConv(GradientDescentBase):
         fwd = self._fwd()
         desc = fwd.conv_desc()
         eo = self.err_output.devmem
-        if self.ACTIVATION != act.ACT_LINEAR:
+        if self.ACTIVATION != act.ACT_LINEAR and \
+                not getattr(self, "err_output_preactivated", False):
             ops.act_bwd_inplace(eo.view(eo.shape[0], -1),
                                 ops.as2d_bf16(self.output),
                                 self.ACTIVATION)
         batch = int(self.input.shape[0])
         if self.need_err_input:
+            in_act = getattr(self, "input_activation", None)
             ops.conv_dgrad(eo, self.weights.bf16_dgrad,
-                           self.err_input.devmem, desc)
+                           self.err_input.devmem, desc,
+                           activation=in_act or 0,
+                           dact=self.input.devmem if in_act else None)
             self.err_input.touch_device()
         want_bias = self.include_bias and self.bias is not None and \
             self.bias

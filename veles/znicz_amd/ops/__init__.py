@@ -149,9 +149,14 @@ def labels_on_device(labels_arr, device):
 # kernel wrappers (1:1 with bindings.cpp)
 # ---------------------------------------------------------------------------
 
-def gemm_nt(a, b, out, bias=None, activation=0):
-    """out[M,N] = act(a[M,K] · b[N,K]ᵀ + bias).  a,b bf16; out bf16|f32."""
-    require().gemm_nt(a, b, out, bias, int(activation))
+def gemm_nt(a, b, out, bias=None, activation=0, dact=None):
+    """out[M,N] = act(a[M,K] · b[N,K]ᵀ + bias).  a,b bf16; out bf16|f32.
+
+    With ``dact`` (bf16, same shape as out) the epilogue instead FUSES
+    the consumer layer's activation derivative: out = (a·bᵀ) ∘
+    act'(dact) — replaces the separate act_bwd pass on the dgrad path
+    (``activation`` is then the derivative code)."""
+    require().gemm_nt(a, b, out, bias, int(activation), dact)
 
 
 def gemm_tn(a, b, out, beta=0.0, bias_out=None):
@@ -295,8 +300,13 @@ def conv_fwd(x, w, out, bias, conv_desc, activation=0):
     require().conv_fwd(x, w, out, bias, conv_desc, int(activation))
 
 
-def conv_dgrad(err_out, w_dgrad, err_in, conv_desc):
-    require().conv_dgrad(err_out, w_dgrad, err_in, conv_desc)
+def conv_dgrad(err_out, w_dgrad, err_in, conv_desc, activation=0,
+               dact=None):
+    """err_in = col2im-gather(err_out · Wᵀ); with ``dact`` the consumer
+    layer's activation derivative is fused into the epilogue (see
+    gemm_nt)."""
+    require().conv_dgrad(err_out, w_dgrad, err_in, conv_desc,
+                         int(activation), dact)
 
 
 def conv_wgrad(err_out, x, grad_w, conv_desc, beta=0.0, bias_out=None):

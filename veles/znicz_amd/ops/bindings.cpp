@@ -10,7 +10,8 @@
 // from gemm.hip
 void gemm_nt_dense(const unsigned short*, const unsigned short*, void*,
                    bool, const float*, const unsigned short*, int, int,
-                   int, int, hipStream_t);
+                   int, int, hipStream_t,
+                   const unsigned short* dact = nullptr);
 bool gemm_tn_dense(const unsigned short*, const unsigned short*, float*,
                    int, int, int, float, hipStream_t, float*);
 void conv_fwd_gemm(const unsigned short*, const unsigned short*, void*,
@@ -18,7 +19,8 @@ void conv_fwd_gemm(const unsigned short*, const unsigned short*, void*,
                    const ConvDesc&, int, hipStream_t);
 void conv_dgrad_gemm(const unsigned short*, const unsigned short*,
                      unsigned short*, const unsigned short*,
-                     const ConvDesc&, hipStream_t);
+                     const ConvDesc&, hipStream_t, int act = 0,
+                     const unsigned short* dact = nullptr);
 bool conv_wgrad_gemm(const unsigned short*, const unsigned short*, float*,
                      const ConvDesc&, float, hipStream_t, float*);
 // from elementwise.hip
@@ -112,7 +114,8 @@ ConvDesc make_conv_desc(const std::vector<int64_t>& v) {
 }
 
 void gemm_nt(torch::Tensor a, torch::Tensor b, torch::Tensor out,
-             c10::optional<torch::Tensor> bias, int64_t act) {
+             c10::optional<torch::Tensor> bias, int64_t act,
+             c10::optional<torch::Tensor> dact) {
   const int M = a.size(0), K = a.size(1), N = b.size(0);
   TORCH_CHECK(b.size(1) == K, "gemm_nt: K mismatch");
   TORCH_CHECK(out.size(0) == M && out.size(1) == N, "gemm_nt: out shape");
@@ -124,8 +127,10 @@ void gemm_nt(torch::Tensor a, torch::Tensor b, torch::Tensor out,
   bool f32out = out.scalar_type() == torch::kFloat32;
   void* op = f32out ? (void*)out.data_ptr<float>()
                     : (void*)ubf_mut(out);
+  const unsigned short* dp = dact.has_value() ? ubf(*dact) : nullptr;
+  if (dp) TORCH_CHECK(dact->numel() == out.numel(), "gemm_nt: dact shape");
   gemm_nt_dense(ubf(a), ubf(b), op, f32out, bp, zero_page(a.device()), M,
-                N, K, (int)act, cur_stream());
+                N, K, (int)act, cur_stream(), dp);
 }
 
 bool gemm_tn(torch::Tensor a, torch::Tensor b, torch::Tensor out,
@@ -157,10 +162,14 @@ void conv_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor out,
 }
 
 void conv_dgrad(torch::Tensor eo, torch::Tensor wdg, torch::Tensor ei,
-                std::vector<int64_t> desc) {
+                std::vector<int64_t> desc, int64_t act,
+                c10::optional<torch::Tensor> dact) {
   ConvDesc d = make_conv_desc(desc);
+  const unsigned short* dp = dact.has_value() ? ubf(*dact) : nullptr;
+  if (dp) TORCH_CHECK(dact->numel() == ei.numel(),
+                      "conv_dgrad: dact shape");
   conv_dgrad_gemm(ubf(eo), ubf(wdg), ubf_mut(ei), zero_page(eo.device()),
-                  d, cur_stream());
+                  d, cur_stream(), (int)act, dp);
 }
 
 bool conv_wgrad(torch::Tensor eo, torch::Tensor x, torch::Tensor gw,
@@ -425,10 +434,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("row_argmin", &row_argmin);
   m.def("mean_disp", &mean_disp);
   m.doc() = "veles.znicz_amd hand-written CDNA4 (gfx950) kernels";
-  m.def("gemm_nt", &gemm_nt, "C=act(A·Bᵀ+bias), bf16 MFMA");
+  m.def("gemm_nt", &gemm_nt, "C=act(A·Bᵀ+bias), bf16 MFMA",
+        py::arg("a"), py::arg("b"), py::arg("out"),
+        py::arg("bias") = c10::nullopt, py::arg("act") = 0,
+        py::arg("dact") = c10::nullopt);
   m.def("gemm_tn", &gemm_tn, "C=AᵀB (K-major wgrad core)");
   m.def("conv_fwd", &conv_fwd, "implicit-im2col conv forward");
-  m.def("conv_dgrad", &conv_dgrad, "conv backward-data (gather)");
+  m.def("conv_dgrad", &conv_dgrad, "conv backward-data (gather)",
+        py::arg("eo"), py::arg("wdg"), py::arg("ei"), py::arg("desc"),
+        py::arg("act") = 0, py::arg("dact") = c10::nullopt);
   m.def("conv_wgrad", &conv_wgrad, "conv backward-weights (K-major)");
   m.def("act_fwd", &act_fwd);
   m.def("act_bwd", &act_bwd);

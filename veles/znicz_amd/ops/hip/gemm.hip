@@ -46,7 +46,8 @@
 template <typename LA, typename LB, typename OUT_T, int ACT>
 __global__ __launch_bounds__(256, 2) void gemm_nt_k(
     LA la, LB lb, OUT_T* __restrict__ C, const float* __restrict__ bias,
-    int M, int N, int K, int tiles_n) {
+    const unsigned short* __restrict__ dact, int M, int N, int K,
+    int tiles_n) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const int tid = threadIdx.x;
   const int wave = tid >> 6, lane = tid & 63;
@@ -243,7 +244,12 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_k(
                           acc[mi][ni][r]);
           continue;
         }
-        float v = act_fwd<ACT>(acc[mi][ni][r] + bv);
+        // dact != null: this is a dgrad whose consumer's activation
+        // derivative is FUSED here (ACT is the derivative code, dact
+        // the forward output y) — replaces the separate act_bwd pass
+        float v = dact ? acc[mi][ni][r] *
+                             act_dy<ACT>(bf2f(dact[(long)row * N + col]))
+                       : act_fwd<ACT>(acc[mi][ni][r] + bv);
         if constexpr (sizeof(OUT_T) == 2)
           C[(long)row * N + col] = (OUT_T)f2bf(v);
         else
@@ -264,7 +270,8 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_k(
 template <typename LA, typename LB, typename OUT_T, int ACT, int BNT>
 __global__ __launch_bounds__(256, 2) void gemm_ntn_k(
     LA la, LB lb, OUT_T* __restrict__ C, const float* __restrict__ bias,
-    int M, int N, int K, int tiles_n) {
+    const unsigned short* __restrict__ dact, int M, int N, int K,
+    int tiles_n) {
   constexpr int LDS_B = BNT * BK * 2;
   constexpr int LDS_BUF = NT_LDS_A + LDS_B;
   extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -406,7 +413,9 @@ __global__ __launch_bounds__(256, 2) void gemm_ntn_k(
       for (int r = 0; r < 4; ++r) {
         const int row = rbase + r;
         if (row >= M) continue;
-        float v = act_fwd<ACT>(acc[mi][ni][r] + bv);
+        float v = dact ? acc[mi][ni][r] *
+                             act_dy<ACT>(bf2f(dact[(long)row * N + col]))
+                       : act_fwd<ACT>(acc[mi][ni][r] + bv);
         if constexpr (sizeof(OUT_T) == 2)
           C[(long)row * N + col] = (OUT_T)f2bf(v);
         else
@@ -420,13 +429,17 @@ __global__ __launch_bounds__(256, 2) void gemm_ntn_k(
 template <typename OUT_T, int ACT>
 __global__ void bias_act_cast_k(const float* __restrict__ ws,
                                 OUT_T* __restrict__ out,
-                                const float* __restrict__ bias, long n,
-                                int N) {
+                                const float* __restrict__ bias,
+                                const unsigned short* __restrict__ dact,
+                                long n, int N) {
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   long stride = (long)gridDim.x * blockDim.x;
   for (long p = i; p < n; p += stride) {
-    float v = ws[p] + (bias ? bias[p % N] : 0.f);
-    v = act_fwd<ACT>(v);
+    float v;
+    if (dact)
+      v = ws[p] * act_dy<ACT>(bf2f(dact[p]));
+    else
+      v = act_fwd<ACT>(ws[p] + (bias ? bias[p % N] : 0.f));
     if constexpr (sizeof(OUT_T) == 2)
       out[p] = (OUT_T)f2bf(v);
     else
@@ -1154,14 +1167,15 @@ void set_routing(int nt_split, int tn_split, int tn2_split,
 
 template <typename LA, typename LB, typename OUT_T>
 void launch_gemm_nt(const LA& la, const LB& lb, OUT_T* C, const float* bias,
-                    int M, int N, int K, int act, hipStream_t stream) {
+                    int M, int N, int K, int act, hipStream_t stream,
+                    const unsigned short* dact = nullptr) {
   if (N <= 32) {
     const int tn = cdiv(N, 32);
     dim3 grid(cdiv(M, BM) * tn), block(256);
 #define GO32(A)                                                              \
     hipLaunchKernelGGL((gemm_ntn_k<LA, LB, OUT_T, A, 32>), grid, block,      \
                        (NT_LDS_A + 32 * BK * 2) * 2, stream, la, lb, C,      \
-                       bias, M, N, K, tn)
+                       bias, dact, M, N, K, tn)
     ACT_SWITCH(act, GO32);
 #undef GO32
     HIP_CHECK_LAST();
@@ -1173,7 +1187,7 @@ void launch_gemm_nt(const LA& la, const LB& lb, OUT_T* C, const float* bias,
 #define GO64(A)                                                              \
     hipLaunchKernelGGL((gemm_ntn_k<LA, LB, OUT_T, A, 64>), grid, block,      \
                        (NT_LDS_A + 64 * BK * 2) * 2, stream, la, lb, C,      \
-                       bias, M, N, K, tn)
+                       bias, dact, M, N, K, tn)
     ACT_SWITCH(act, GO64);
 #undef GO64
     HIP_CHECK_LAST();
@@ -1187,7 +1201,7 @@ void launch_gemm_nt(const LA& la, const LB& lb, OUT_T* C, const float* bias,
 #define GO96(A)                                                              \
     hipLaunchKernelGGL((gemm_ntn_k<LA, LB, OUT_T, A, 96>), grid, block,      \
                        (NT_LDS_A + 96 * BK * 2) * 2, stream, la, lb, C,      \
-                       bias, M, N, K, tn)
+                       bias, dact, M, N, K, tn)
     ACT_SWITCH(act, GO96);
 #undef GO96
     HIP_CHECK_LAST();
@@ -1197,7 +1211,8 @@ void launch_gemm_nt(const LA& la, const LB& lb, OUT_T* C, const float* bias,
   dim3 grid(cdiv(M, BM) * tn), block(256);
 #define GO(A)                                                                \
   hipLaunchKernelGGL((gemm_nt_k<LA, LB, OUT_T, A>), grid, block,             \
-                     NT_LDS_TOTAL, stream, la, lb, C, bias, M, N, K, tn)
+                     NT_LDS_TOTAL, stream, la, lb, C, bias, dact, M, N, K, \
+                     tn)
   ACT_SWITCH(act, GO);
 #undef GO
   HIP_CHECK_LAST();
@@ -1279,7 +1294,8 @@ bool launch_gemm_tn2(const LA& la, const LB& lb, float* C, int M, int N,
 void gemm_nt_dense(const unsigned short* A, const unsigned short* B,
                    void* C, bool c_is_f32, const float* bias,
                    const unsigned short* zero, int M, int N, int K,
-                   int act, hipStream_t s) {
+                   int act, hipStream_t s,
+                   const unsigned short* dact = nullptr) {
   DenseNT la{A, zero, K, K}, lb{B, zero, K, K};
   const int tiles = cdiv(M, BM) * cdiv(N, BN);
   const int nk = cdiv(K, BK);
@@ -1315,7 +1331,9 @@ void gemm_nt_dense(const unsigned short* A, const unsigned short* B,
     dim3 grid(tiles, splits), block(256);
     hipLaunchKernelGGL((gemm_nt_k<DenseNT, DenseNT, float, ACT_LINEAR>),
                        grid, block, NT_LDS_TOTAL, s, la, lb, ws,
-                       (const float*)nullptr, M, N, K, cdiv(N, BN));
+                       (const float*)nullptr,
+                       (const unsigned short*)nullptr, M, N, K,
+                       cdiv(N, BN));
     HIP_CHECK_LAST();
     long n = (long)M * N;
     int b = (int)min((n + 255) / 256, (long)2048);
@@ -1323,11 +1341,12 @@ void gemm_nt_dense(const unsigned short* A, const unsigned short* B,
     do {                                                                     \
       if (c_is_f32)                                                          \
         hipLaunchKernelGGL((bias_act_cast_k<float, A_>), dim3(b),            \
-                           dim3(256), 0, s, ws, (float*)C, bias, n, N);      \
+                           dim3(256), 0, s, ws, (float*)C, bias, dact, n,    \
+                           N);                                               \
       else                                                                   \
         hipLaunchKernelGGL((bias_act_cast_k<unsigned short, A_>), dim3(b),   \
                            dim3(256), 0, s, ws, (unsigned short*)C, bias,    \
-                           n, N);                                            \
+                           dact, n, N);                                      \
     } while (0)
     ACT_SWITCH(act, FIN);
 #undef FIN
@@ -1335,9 +1354,10 @@ void gemm_nt_dense(const unsigned short* A, const unsigned short* B,
     return;
   }
   if (c_is_f32)
-    launch_gemm_nt(la, lb, (float*)C, bias, M, N, K, act, s);
+    launch_gemm_nt(la, lb, (float*)C, bias, M, N, K, act, s, dact);
   else
-    launch_gemm_nt(la, lb, (unsigned short*)C, bias, M, N, K, act, s);
+    launch_gemm_nt(la, lb, (unsigned short*)C, bias, M, N, K, act, s,
+                   dact);
 }
 
 bool gemm_tn_dense(const unsigned short* A, const unsigned short* B,
@@ -1502,24 +1522,27 @@ void conv_fwd_gemm(const unsigned short* x, const unsigned short* w,
 
 void conv_dgrad_gemm(const unsigned short* eo, const unsigned short* wdg,
                      unsigned short* ei, const unsigned short* zero,
-                     const ConvDesc& d, hipStream_t s) {
+                     const ConvDesc& d, hipStream_t s,
+                     int act = ACT_LINEAR,
+                     const unsigned short* dact = nullptr) {
   const int M = d.N * d.H * d.W;       // input pixels
   const int N = d.C;                   // input channels
   const int K = d.KH * d.KW * d.K;     // co-inner gather dim
   static const bool use8 = getenv("VELES_NT8") != nullptr;
-  if (use8 && M >= 2048 && N >= 192 && (K & 7) == 0) {
+  if (use8 && M >= 2048 && N >= 192 && (K & 7) == 0 && !dact) {
     conv_dgrad_gemm8(eo, wdg, ei, zero, d, s);
     return;
   }
   DenseNT lb{wdg, zero, K, K};         // wdg: [C][KH][KW][K]
   if (d.SY == 1 && d.SX == 1) {
     DgradNT1 la{eo, zero, d};
-    launch_gemm_nt(la, lb, ei, (const float*)nullptr, M, N, K,
-                   ACT_LINEAR, s);
+    launch_gemm_nt(la, lb, ei, (const float*)nullptr, M, N, K, act, s,
+                   dact);
     return;
   }
   DgradNT la{eo, zero, d};
-  launch_gemm_nt(la, lb, ei, (const float*)nullptr, M, N, K, ACT_LINEAR, s);
+  launch_gemm_nt(la, lb, ei, (const float*)nullptr, M, N, K, act, s,
+                 dact);
 }
 
 bool conv_wgrad_gemm(const unsigned short* eo, const unsigned short* x,

@@ -210,7 +210,38 @@ class StandardWorkflow(StandardWorkflowBase):
             unit.gate_skip = self.decision.gd_skip
             self.gds[i] = unit
             prev = unit
+        self._fuse_act_bwd_pairs()
         return prev
+
+    def _fuse_act_bwd_pairs(self):
+        """HIP-path fusion (reference's err_y_update fold, SURVEY §2d
+        row 5): where gd[i]'s err_input IS gd[i-1]'s err_output and
+        layer i-1 has an activation, gd[i]'s dgrad GEMM epilogue
+        multiplies by act'(input) and gd[i-1] skips its separate
+        act_bwd pass.  Only GEMM-backed gd pairs qualify (pooling/LRN/
+        dropout gds forward err untouched and break the chain).  The
+        numpy reference path ignores both flags.
+
+        Default OFF (opt-in VELES_DGRAD_FUSE=1): same-box A/B measured
+        the fusion NET-NEGATIVE on every benched model (VGG -2.1%,
+        AlexNet -0.8%) — the standalone act_bwd kernel streams
+        octet-vectorized at ~6.5 TB/s while the fused epilogue's
+        per-element 2-B dact reads serialize the GEMM tail.  Kept as a
+        tested capability (tests/test_kernels_gpu.py)."""
+        import os
+        if not os.environ.get("VELES_DGRAD_FUSE"):
+            return
+        from veles.znicz_amd import activation_defs as act
+        from veles.znicz_amd.gd import GradientDescent
+        from veles.znicz_amd.gd_conv import GradientDescentConv
+        gemmlike = (GradientDescent, GradientDescentConv)
+        for i in range(1, len(self.gds)):
+            g, up = self.gds[i], self.gds[i - 1]
+            if (isinstance(g, gemmlike) and isinstance(up, gemmlike) and
+                    g.need_err_input and
+                    up.ACTIVATION != act.ACT_LINEAR):
+                g.input_activation = up.ACTIVATION
+                up.err_output_preactivated = True
 
     def link_loop(self, last_unit):
         # CommSync: the compute stream joins the comm stream's pending
