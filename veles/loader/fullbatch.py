@@ -86,6 +86,34 @@ class FullBatchLoader(Loader):
         self._idx_key = key
         return self._idx_dev[:size]
 
+    #: GraphStepper sets this: the host side of fill_minibatch (index
+    #: bookkeeping + pinned upload) stays eager, but the device gather
+    #: is deferred to :meth:`device_gather`, which the stepper calls
+    #: INSIDE the captured region — the index buffer has a fixed
+    #: address, so replays read each step's freshly-uploaded indices.
+    defer_device_gather = False
+
+    _gather_pending = False
+
+    def device_gather_if_pending(self):
+        if self._gather_pending:
+            self._gather_pending = False
+            self.device_gather()
+
+    def device_gather(self):
+        """The device-side portion of fill_minibatch (full minibatches
+        only — the graph path never runs short/valid batches)."""
+        import torch
+        size = self.minibatch_size
+        dev_idx = self._idx_dev[:size]
+        torch.index_select(self.original_data.devmem, 0, dev_idx,
+                           out=self.minibatch_data.devmem[:size])
+        self.minibatch_data.touch_device()
+        if self.original_labels.devmem is not None:
+            torch.index_select(self.original_labels.devmem, 0, dev_idx,
+                               out=self.minibatch_labels.devmem[:size])
+            self.minibatch_labels.touch_device()
+
     def fill_minibatch(self):
         size = self.minibatch_size
         idx = self.minibatch_indices.mem[:size]
@@ -93,6 +121,15 @@ class FullBatchLoader(Loader):
             import torch
             dev_idx = self._dev_indices(
                 idx, self.original_data.devmem.device)
+            # host mirrors for Decision/debug paths
+            self.minibatch_labels.mem[:size] = \
+                self.original_labels.mem[idx]
+            from veles.loader.base import TRAIN
+            if self.defer_device_gather and \
+                    size == self.max_minibatch_size and \
+                    int(self.minibatch_class) == TRAIN:
+                self._gather_pending = True
+                return                   # stepper gathers inside the graph
             # device-side gather straight into the minibatch buffer
             torch.index_select(
                 self.original_data.devmem, 0, dev_idx,
@@ -103,9 +140,6 @@ class FullBatchLoader(Loader):
                     self.original_labels.devmem, 0, dev_idx,
                     out=self.minibatch_labels.devmem[:size])
                 self.minibatch_labels.touch_device()
-            # host mirrors for Decision/debug paths
-            self.minibatch_labels.mem[:size] = \
-                self.original_labels.mem[idx]
         else:
             self.minibatch_data.map_invalidate()
             self.minibatch_data.mem[:size] = self.original_data.mem[idx]

@@ -59,6 +59,11 @@ class GraphStepper(object):
         self._sig = None
         self._touched = ()
         self._disabled = not enabled or not self._graphs_supported()
+        # pull the loader's device gather INTO the captured region: its
+        # index buffer has a fixed address and the pinned upload stays
+        # eager, so each replay gathers that step's fresh minibatch
+        if hasattr(self.loader, "device_gather"):
+            self.loader.defer_device_gather = True
 
     @staticmethod
     def _graphs_supported():
@@ -95,6 +100,8 @@ class GraphStepper(object):
         self.decision.run()
 
     def _middle(self):
+        if hasattr(self.loader, "device_gather_if_pending"):
+            self.loader.device_gather_if_pending()
         for f in self.forwards:
             f.run()
         self.evaluator.run()
@@ -103,6 +110,8 @@ class GraphStepper(object):
         self.comm_sync.run()
 
     def _eager(self, check_skip=False):
+        if hasattr(self.loader, "device_gather_if_pending"):
+            self.loader.device_gather_if_pending()
         for f in self.forwards:
             f.run()
         self.evaluator.run()
