@@ -585,6 +585,156 @@ void launch_gemm_ntn8(const LA& la, const LB& lb, OUT_T* C,
   HIP_CHECK_LAST();
 }
 
+
+// ---------------------------------------------------------------------------
+// deep-pipelined WIDE core: 8 waves (512 thr), BM=256 × BN=128, three
+// LDS buffers, 2-tile prefetch (counted vmcnt(12)) — the ntn8 schedule
+// at full width for the big conv fwd/dgrad shapes.  48 chunks/buffer
+// (32 A + 16 B, all real) = 48 KiB; ×3 = 144 KiB, 1 block/CU, 2
+// waves/SIMD (same as the 2-block 128² core, but with twice the
+// prefetch depth).  Waves stack 8M×1N: each wave owns 32 rows × 128
+// cols.
+// ---------------------------------------------------------------------------
+template <typename LA, typename LB, typename OUT_T, int ACT>
+__global__ __launch_bounds__(512, 1) void gemm_nt8d_k(
+    LA la, LB lb, OUT_T* __restrict__ C, const float* __restrict__ bias,
+    const unsigned short* __restrict__ dact, int M, int N, int K,
+    int tiles_n) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int tile = xcd_tile(blockIdx.x, gridDim.x);
+  const int tm = tile / tiles_n, tn = tile % tiles_n;
+  const int m0 = tm * 256, n0 = tn * 128;
+  const int nk = (K + BK - 1) / BK;
+  const int rl = lane >> 3;
+  const int k8 = lane & 7;
+
+  typename LA::RowC rca[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+    rca[i] = la.rowc(min(m0 + (wave * 4 + i) * 8 + rl, M - 1));
+  typename LB::RowC rcb[2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+    rcb[i] = lb.rowc(min(n0 + (wave + i * 8) * 8 + rl, N - 1));
+
+  auto stage = [&](int buf, int kt) {
+    char* base = smem + buf * N8_BUF;
+    const int k8base = kt * (BK / 8);
+    const auto kca0 = la.kc(k8base + (k8 ^ rl));
+    const auto kca1 = la.kc(k8base + (k8 ^ rl ^ 1));
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int chunk = wave * 4 + i;
+      const void* sa = la.addr_rk(rca[i], (i & 1) ? kca1 : kca0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)sa,
+          (__attribute__((address_space(3))) void*)(base + chunk * 1024),
+          16, 0, 0);
+    }
+    const auto kcb0 = lb.kc(k8base + (k8 ^ rl));
+    const auto kcb1 = lb.kc(k8base + (k8 ^ rl ^ 1));
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      const int bc = wave + i * 8;                 // B chunks 0..15
+      const int chunk = 32 + bc;
+      const void* sb = lb.addr_rk(rcb[i], (bc & 1) ? kcb1 : kcb0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)sb,
+          (__attribute__((address_space(3))) void*)(base + chunk * 1024),
+          16, 0, 0);
+    }
+  };
+
+  f32x4 acc[2][8] = {};
+
+  if (nk > 0) stage(0, 0);
+  if (nk > 1) stage(1, 1);
+  int buf = 0;
+  for (int t = 0; t < nk; ++t) {
+    __builtin_amdgcn_s_barrier();
+    if (t + 2 < nk) {
+      stage(buf + 2 >= 3 ? buf - 1 : buf + 2, t + 2);
+      asm volatile("s_waitcnt vmcnt(12)" ::: "memory");
+    } else if (t + 1 < nk) {
+      asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+    const char* As = smem + buf * N8_BUF;
+    const char* Bs = As + 32 * 1024;
+    const int fr = lane & 15;
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      const int ko = ks * 4 + (lane >> 4);
+      short8 af[2], bf[8];
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        const int row = wave * 32 + i * 16 + fr;
+        const int slot = ko ^ (row & 7) ^ ((row >> 3) & 1);
+        af[i] = *(const short8*)(As + row * 128 + slot * 16);
+      }
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        const int row = i * 16 + fr;
+        const int slot = ko ^ (row & 7) ^ ((row >> 3) & 1);
+        bf[i] = *(const short8*)(Bs + row * 128 + slot * 16);
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 8; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+    buf = buf + 1 == 3 ? 0 : buf + 1;
+  }
+
+  const int colw = n0 + (lane & 15);
+#pragma unroll
+  for (int ni = 0; ni < 8; ++ni) {
+    const int col = colw + ni * 16;
+    if (col >= N) continue;
+    const float bv = bias ? bias[col] : 0.f;
+#pragma unroll
+    for (int mi = 0; mi < 2; ++mi) {
+      const int rbase = m0 + wave * 32 + mi * 16 + ((lane >> 4) << 2);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = rbase + r;
+        if (row >= M) continue;
+        float v = dact ? acc[mi][ni][r] *
+                             act_dy<ACT>(bf2f(dact[(long)row * N + col]))
+                       : act_fwd<ACT>(acc[mi][ni][r] + bv);
+        if constexpr (sizeof(OUT_T) == 2)
+          C[(long)row * N + col] = (OUT_T)f2bf(v);
+        else
+          C[(long)row * N + col] = (OUT_T)v;
+      }
+    }
+  }
+}
+
+template <typename LA, typename LB, typename OUT_T>
+void launch_gemm_nt8d(const LA& la, const LB& lb, OUT_T* C,
+                      const float* bias, int M, int N, int K, int act,
+                      hipStream_t stream,
+                      const unsigned short* dact = nullptr) {
+  const int tn = (N + 127) / 128;
+  dim3 grid(((M + 255) / 256) * tn), block(512);
+#define GO8D(A)                                                              \
+  hipLaunchKernelGGL((gemm_nt8d_k<LA, LB, OUT_T, A>), grid, block,           \
+                     3 * N8_BUF, stream, la, lb, C, bias, dact, M, N, K,     \
+                     tn)
+  ACT_SWITCH(act, GO8D);
+#undef GO8D
+  HIP_CHECK_LAST();
+}
+
 // split-K finisher: out = act(ws + bias), bf16 or fp32 out
 template <typename OUT_T, int ACT>
 __global__ void bias_act_cast_k(const float* __restrict__ ws,
@@ -1686,6 +1836,15 @@ void conv_fwd_gemm(const unsigned short* x, const unsigned short* w,
   }
   Im2colNT la{x, zero, d};
   DenseNT lb{w, zero, K, K};
+  static const bool nt8d = getenv("VELES_NT8D") != nullptr;
+  if (nt8d && d.K > 96 && (K & 7) == 0 && K >= 512 && M >= 8192) {
+    if (out_f32)
+      launch_gemm_nt8d(la, lb, (float*)out, bias, M, d.K, K, act, s);
+    else
+      launch_gemm_nt8d(la, lb, (unsigned short*)out, bias, M, d.K, K,
+                       act, s);
+    return;
+  }
   if (out_f32)
     launch_gemm_nt(la, lb, (float*)out, bias, M, d.K, K, act, s);
   else
@@ -1712,6 +1871,12 @@ void conv_dgrad_gemm(const unsigned short* eo, const unsigned short* wdg,
     if (ntn8 && N > 64 && N <= 96 && (N & 7) == 0 && (K & 7) == 0 &&
         K >= 512 && M >= 4096) {
       launch_gemm_ntn8(la, lb, ei, (const float*)nullptr, M, N, K, act,
+                       s, dact);
+      return;
+    }
+    static const bool nt8d = getenv("VELES_NT8D") != nullptr;
+    if (nt8d && N > 96 && (K & 7) == 0 && K >= 512 && M >= 8192) {
+      launch_gemm_nt8d(la, lb, ei, (const float*)nullptr, M, N, K, act,
                        s, dact);
       return;
     }
