@@ -425,3 +425,63 @@ def test_yale_faces_trains_on_gpu():
     assert wf.is_finished
     assert wf.decision.epoch_n_err_pt[TRAIN] < 60.0, \
         wf.decision.epoch_n_err_pt
+
+
+def test_nn_rollback_restores_device_weights():
+    """the rollback must reach the DEVICE — after restore,
+    the forward output must change back (fp32 master pushed by unmap,
+    bf16/bf16_t/bf16_dgrad working copies rebuilt)."""
+    from veles.config import root as _root
+    from veles.znicz_amd.models.cifar import CifarWorkflow
+    from veles.znicz_amd.nn_rollback import NNRollback
+    prng.seed_all(77)
+    _root.cifar.loader.update({"minibatch_size": 32, "n_train": 64,
+                               "n_valid": 0, "on_device": True,
+                               "device_dtype": "bf16", "seed": 5})
+    _root.cifar.decision.update({"max_epochs": 10 ** 9,
+                                 "fail_iterations": 10 ** 9})
+    wf = CifarWorkflow(DummyLauncher())
+    dev = ROCmDevice(0)
+    wf.initialize(device=dev)
+    rb = NNRollback(wf, rollback_to_best=1, lr_cut_ratio=0.5)
+    rb.gds = wf.gds
+    rb.improved = True
+
+    wf.loader.run()                    # ONE fixed minibatch for all
+                                       # forward probes (loader not
+                                       # re-run: data stays identical)
+
+    def forward_out():
+        for f in wf.forwards:
+            f.run()
+        torch.cuda.synchronize()
+        out = wf.forwards[-1].output
+        out.map_read()
+        return out.mem.copy()
+
+    y0 = forward_out()
+    rb.run()                           # improved: snapshot best weights
+    # train a few steps so the device weights move away from "best"
+    # (same minibatch every step — fine for moving weights)
+    for _ in range(4):
+        for f in wf.forwards:
+            f.run()
+        wf.evaluator.run()
+        wf.decision.run()
+        for g in reversed(wf.gds):
+            g.run()
+        wf.comm_sync.run()
+    y_trained = forward_out()
+    assert numpy.abs(y_trained - y0).max() > 1e-5, \
+        "training did not move the weights (test is vacuous)"
+    # now "diverge": rollback must restore the snapshot ON DEVICE
+    rb.improved = False
+    rb.run()
+    y_rolled = forward_out()
+    conv_w = wf.forwards[0].weights
+    conv_w.map_read()
+    # the restored forward runs on the restored bf16 copies; outputs
+    # must be much closer to y0 than the trained outputs were
+    d_rolled = numpy.abs(y_rolled - y0).mean()
+    d_trained = numpy.abs(y_trained - y0).mean()
+    assert d_rolled < d_trained * 0.5, (d_rolled, d_trained)
