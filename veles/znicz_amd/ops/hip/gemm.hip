@@ -437,31 +437,36 @@ __global__ __launch_bounds__(256, 2) void gemm_ntn_k(
 #define N8_CHUNKS 48
 #define N8_BUF (N8_CHUNKS * 1024)
 
-template <typename LA, typename LB, typename OUT_T, int ACT>
+template <typename LA, typename LB, typename OUT_T, int ACT,
+          int BNT = 96>
 __global__ __launch_bounds__(512, 1) void gemm_ntn8_k(
     LA la, LB lb, OUT_T* __restrict__ C, const float* __restrict__ bias,
     const unsigned short* __restrict__ dact, int M, int N, int K,
     int tiles_n) {
+  constexpr int BCH = BNT / 8;          // real B chunks
+  constexpr int BSTG = BNT > 64 ? 2 : 1;  // B chunks staged per wave
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const int tid = threadIdx.x;
   const int wave = tid >> 6, lane = tid & 63;
   const int tile = xcd_tile(blockIdx.x, gridDim.x);
   const int tm = tile / tiles_n, tn = tile % tiles_n;
-  const int m0 = tm * 256, n0 = tn * 96;
+  const int m0 = tm * 256, n0 = tn * BNT;
   const int nk = (K + BK - 1) / BK;
   const int rl = lane >> 3;
   const int k8 = lane & 7;
 
   typename LA::RowC rca[4];
-  typename LB::RowC rcb;
+  typename LB::RowC rcb[BSTG];
 #pragma unroll
   for (int i = 0; i < 4; ++i)
     rca[i] = la.rowc(min(m0 + (wave * 4 + i) * 8 + rl, M - 1));
-  // B: wave w stages chunks w and 8+w (8..11 real, 12..15 zero pads)
-  const int bc2 = 8 + wave;
-  rcb = lb.rowc(min(n0 + wave * 8 + rl, N - 1));
-  typename LB::RowC rcb2 = lb.rowc(
-      min(n0 + (bc2 < 12 ? bc2 * 8 + rl : 0), N - 1));
+  // B: wave w stages chunk w (+ chunk 8+w when BSTG=2); chunks past
+  // BCH are zero-area pads that keep the per-thread glds count uniform
+#pragma unroll
+  for (int i = 0; i < BSTG; ++i) {
+    const int bc = wave + i * 8;
+    rcb[i] = lb.rowc(min(n0 + (bc < BCH ? bc * 8 + rl : 0), N - 1));
+  }
 
   auto stage = [&](int buf, int kt) {
     char* base = smem + buf * N8_BUF;
@@ -479,17 +484,11 @@ __global__ __launch_bounds__(512, 1) void gemm_ntn8_k(
     }
     const auto kcb0 = lb.kc(k8base + (k8 ^ rl));
     const auto kcb1 = lb.kc(k8base + (k8 ^ rl ^ 1));
-    {
-      const int chunk = 32 + wave;                 // B chunk `wave`
-      const void* sb = lb.addr_rk(rcb, (wave & 1) ? kcb1 : kcb0);
-      __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) void*)sb,
-          (__attribute__((address_space(3))) void*)(base + chunk * 1024),
-          16, 0, 0);
-    }
-    {
-      const int chunk = 32 + bc2;                  // B chunk 8+wave
-      const void* sb = lb.addr_rk(rcb2, (bc2 & 1) ? kcb1 : kcb0);
+#pragma unroll
+    for (int i = 0; i < BSTG; ++i) {
+      const int bc = wave + i * 8;
+      const int chunk = 32 + bc;
+      const void* sb = lb.addr_rk(rcb[i], (bc & 1) ? kcb1 : kcb0);
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) void*)sb,
           (__attribute__((address_space(3))) void*)(base + chunk * 1024),
@@ -497,18 +496,19 @@ __global__ __launch_bounds__(512, 1) void gemm_ntn8_k(
     }
   };
 
-  f32x4 acc[2][6] = {};
+  f32x4 acc[2][BNT / 16] = {};
 
   if (nk > 0) stage(0, 0);
   if (nk > 1) stage(1, 1);
   int buf = 0;
   for (int t = 0; t < nk; ++t) {
     __builtin_amdgcn_s_barrier();
+    constexpr int NL = 4 + BSTG;
     if (t + 2 < nk) {
       stage((buf + 2) % 3, t + 2);
-      asm volatile("s_waitcnt vmcnt(12)" ::: "memory");
+      asm volatile("s_waitcnt vmcnt(%0)" :: "i"(2 * NL) : "memory");
     } else if (t + 1 < nk) {
-      asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+      asm volatile("s_waitcnt vmcnt(%0)" :: "i"(NL) : "memory");
     } else {
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     }
@@ -519,7 +519,7 @@ __global__ __launch_bounds__(512, 1) void gemm_ntn8_k(
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
       const int ko = ks * 4 + (lane >> 4);
-      short8 af[2], bf[6];
+      short8 af[2], bf[BNT / 16];
 #pragma unroll
       for (int i = 0; i < 2; ++i) {
         const int row = wave * 32 + i * 16 + fr;
@@ -527,7 +527,7 @@ __global__ __launch_bounds__(512, 1) void gemm_ntn8_k(
         af[i] = *(const short8*)(As + row * 128 + slot * 16);
       }
 #pragma unroll
-      for (int i = 0; i < 6; ++i) {
+      for (int i = 0; i < BNT / 16; ++i) {
         const int row = i * 16 + fr;
         const int slot = ko ^ (row & 7) ^ ((row >> 3) & 1);
         bf[i] = *(const short8*)(Bs + row * 128 + slot * 16);
@@ -536,7 +536,7 @@ __global__ __launch_bounds__(512, 1) void gemm_ntn8_k(
 #pragma unroll
       for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
-        for (int ni = 0; ni < 6; ++ni)
+        for (int ni = 0; ni < BNT / 16; ++ni)
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
       __builtin_amdgcn_s_setprio(0);
@@ -546,7 +546,7 @@ __global__ __launch_bounds__(512, 1) void gemm_ntn8_k(
 
   const int colw = n0 + (lane & 15);
 #pragma unroll
-  for (int ni = 0; ni < 6; ++ni) {
+  for (int ni = 0; ni < BNT / 16; ++ni) {
     const int col = colw + ni * 16;
     if (col >= N) continue;
     const float bv = bias ? bias[col] : 0.f;
@@ -574,14 +574,32 @@ void launch_gemm_ntn8(const LA& la, const LB& lb, OUT_T* C,
                       const float* bias, int M, int N, int K, int act,
                       hipStream_t stream,
                       const unsigned short* dact = nullptr) {
-  const int tn = (N + 95) / 96;
-  dim3 grid(((M + 255) / 256) * tn), block(512);
-#define GO8(A)                                                               \
-  hipLaunchKernelGGL((gemm_ntn8_k<LA, LB, OUT_T, A>), grid, block,           \
-                     3 * N8_BUF, stream, la, lb, C, bias, dact, M, N, K,     \
-                     tn)
-  ACT_SWITCH(act, GO8);
-#undef GO8
+  dim3 block(512);
+  if (N <= 32) {
+    dim3 grid(((M + 255) / 256) * ((N + 31) / 32));
+#define GO8A(A)                                                              \
+    hipLaunchKernelGGL((gemm_ntn8_k<LA, LB, OUT_T, A, 32>), grid, block,     \
+                       3 * N8_BUF, stream, la, lb, C, bias, dact, M, N,      \
+                       K, (N + 31) / 32)
+    ACT_SWITCH(act, GO8A);
+#undef GO8A
+  } else if (N <= 64) {
+    dim3 grid(((M + 255) / 256) * ((N + 63) / 64));
+#define GO8B(A)                                                              \
+    hipLaunchKernelGGL((gemm_ntn8_k<LA, LB, OUT_T, A, 64>), grid, block,     \
+                       3 * N8_BUF, stream, la, lb, C, bias, dact, M, N,      \
+                       K, (N + 63) / 64)
+    ACT_SWITCH(act, GO8B);
+#undef GO8B
+  } else {
+    dim3 grid(((M + 255) / 256) * ((N + 95) / 96));
+#define GO8C(A)                                                              \
+    hipLaunchKernelGGL((gemm_ntn8_k<LA, LB, OUT_T, A, 96>), grid, block,     \
+                       3 * N8_BUF, stream, la, lb, C, bias, dact, M, N,      \
+                       K, (N + 95) / 96)
+    ACT_SWITCH(act, GO8C);
+#undef GO8C
+  }
   HIP_CHECK_LAST();
 }
 
@@ -1799,8 +1817,11 @@ void conv_fwd_gemm(const unsigned short* x, const unsigned short* w,
     return;
   }
   static const bool ntn8 = getenv("VELES_NO_NTN8") == nullptr;
-  if (ntn8 && (d.C & 7) == 0 && d.K > 64 && d.K <= 96 && (K & 7) == 0 &&
-      K >= 512 && M >= 4096) {
+  // N<=64 shapes take the deep core only at moderate M (VGG's 3.2M-row
+  // conv2 measured -2.4% on it; CIFAR's sub-1M shapes +3%)
+  if (ntn8 && (d.C & 7) == 0 &&
+      (d.K > 64 ? d.K <= 96 : M < (1 << 20)) && d.K <= 96 &&
+      (K & 7) == 0 && K >= 512 && M >= 4096) {
     Im2colNT la{x, zero, d};
     DenseNT lb{w, zero, K, K};
     if (out_f32)
@@ -1868,8 +1889,8 @@ void conv_dgrad_gemm(const unsigned short* eo, const unsigned short* wdg,
   if (d.SY == 1 && d.SX == 1) {
     DgradNT1 la{eo, zero, d};
     static const bool ntn8 = getenv("VELES_NO_NTN8") == nullptr;
-    if (ntn8 && N > 64 && N <= 96 && (N & 7) == 0 && (K & 7) == 0 &&
-        K >= 512 && M >= 4096) {
+    if (ntn8 && (N > 64 ? N <= 96 : M < (1 << 20)) && N <= 96 &&
+        (N & 7) == 0 && (K & 7) == 0 && K >= 512 && M >= 4096) {
       launch_gemm_ntn8(la, lb, ei, (const float*)nullptr, M, N, K, act,
                        s, dact);
       return;
