@@ -256,3 +256,37 @@ def test_launcher_rejects_cpu_multirank():
     lch.dist = _Stub()
     with pytest.raises(RuntimeError, match="single-process"):
         lch.boot()
+
+
+def test_cli_snapshot_resume(tmp_path):
+    """ADVICE r1 regression: `-s snapshot` resume through the CLI (the
+    launcher must be handed the unpickled workflow — boot() crashed on
+    launcher.workflow None before the fix)."""
+    snapdir = tmp_path / "snaps"
+    cfg = tmp_path / "cfg.py"
+    cfg.write_text(
+        "root.common.dirs.snapshots = %r\n"
+        "root.wine.decision.max_epochs = 2\n"
+        "root.wine.loader.n_train = 30\n"
+        "root.wine.loader.n_valid = 10\n"
+        "root.wine.snapshotter.enabled = True\n"
+        "root.wine.snapshotter.interval = 1\n"
+        "root.wine.snapshotter.time_interval = 0\n" % str(snapdir))
+    out = subprocess.run(
+        [sys.executable, "-m", "veles",
+         "veles/znicz_amd/models/wine.py", str(cfg), "-b", "numpy",
+         "--random-seed", "7"],
+        cwd=REPO, capture_output=True, text=True, timeout=600)
+    assert out.returncode == 0, out.stderr[-2000:]
+    snaps = sorted(snapdir.glob("*.pickle*"))
+    assert snaps, "training wrote no snapshot"
+    # resume from the snapshot and run 2 more epochs
+    cfg2 = tmp_path / "cfg2.py"
+    cfg2.write_text(cfg.read_text() +
+                    "root.wine.decision.max_epochs = 4\n")
+    out2 = subprocess.run(
+        [sys.executable, "-m", "veles",
+         "veles/znicz_amd/models/wine.py", str(cfg2), "-b", "numpy",
+         "-s", str(snaps[-1])],
+        cwd=REPO, capture_output=True, text=True, timeout=600)
+    assert out2.returncode == 0, out2.stderr[-2000:]
