@@ -77,6 +77,10 @@ def test_gemm_nt_fc_matrix(M, N, K):
     (8, 8, 8, 32, 64, 5, 1, 2),        # cifar conv3: BNT=64 fwd
     (4, 23, 23, 4, 96, 11, 4, 0),      # alexnet conv1: C=4 packed DMA
     (8, 15, 15, 4, 32, 3, 1, 1),       # C=4 WITH padding: all-slow path
+    (24, 19, 19, 96, 96, 5, 1, 2),     # deep ntn8 BNT=96 fwd+dgrad,
+                                       # ragged M (24*361 % 256 != 0)
+    (20, 21, 21, 64, 72, 3, 1, 1),     # deep ntn8 BNT=64 dgrad (M<2^20,
+                                       # K=648>=512), ragged tiles
 ])
 def test_conv_fwd_dgrad_vs_torch(N, H, W, C, K, KH, SY, PT):
     KW, SX, PL = KH, SY, PT
