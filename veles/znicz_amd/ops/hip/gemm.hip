@@ -240,8 +240,12 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_k(
         const int row = rbase + r;
         if (row >= M) continue;
         if (sp > 1) {
-          unsafeAtomicAdd((float*)&C[(long)row * N + col],
-                          acc[mi][ni][r]);
+          // each K-slice writes its own disjoint fp32 slab (plain
+          // stores); the finisher sums the slabs — measured class:
+          // slab reduce beats fp32 L2 atomics, and full overwrite
+          // removes the zero pass
+          ((float*)C)[(long)sid * M * N + (long)row * N + col] =
+              acc[mi][ni][r];
           continue;
         }
         // dact != null: this is a dgrad whose consumer's activation
@@ -759,15 +763,16 @@ __global__ void bias_act_cast_k(const float* __restrict__ ws,
                                 OUT_T* __restrict__ out,
                                 const float* __restrict__ bias,
                                 const unsigned short* __restrict__ dact,
-                                long n, int N) {
+                                long n, int N, int nslab) {
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   long stride = (long)gridDim.x * blockDim.x;
   for (long p = i; p < n; p += stride) {
-    float v;
+    float v = ws[p];
+    for (int sl = 1; sl < nslab; ++sl) v += ws[p + (long)sl * n];
     if (dact)
-      v = ws[p] * act_dy<ACT>(bf2f(dact[p]));
+      v = v * act_dy<ACT>(bf2f(dact[p]));
     else
-      v = act_fwd<ACT>(ws[p] + (bias ? bias[p % N] : 0.f));
+      v = act_fwd<ACT>(v + (bias ? bias[p % N] : 0.f));
     if constexpr (sizeof(OUT_T) == 2)
       out[p] = (OUT_T)f2bf(v);
     else
@@ -1640,8 +1645,7 @@ void gemm_nt_dense(const unsigned short* A, const unsigned short* B,
     static std::mutex ws_mu;
     static float* ws = nullptr;
     static size_t cap = 0;
-    size_t need = (size_t)cdiv(M, BM) * BM * cdiv(N, BN) * BN *
-                  sizeof(float);
+    size_t need = (size_t)splits * M * N * sizeof(float);
     {
       std::lock_guard<std::mutex> lk(ws_mu);
       if (need > cap) {
@@ -1654,8 +1658,6 @@ void gemm_nt_dense(const unsigned short* A, const unsigned short* B,
         }
       }
     }
-    void zero_f32_launch(float*, long, hipStream_t);
-    zero_f32_launch(ws, (long)M * N, s);
     dim3 grid(tiles, splits), block(256);
     hipLaunchKernelGGL((gemm_nt_k<DenseNT, DenseNT, float, ACT_LINEAR>),
                        grid, block, NT_LDS_TOTAL, s, la, lb, ws,
@@ -1670,11 +1672,11 @@ void gemm_nt_dense(const unsigned short* A, const unsigned short* B,
       if (c_is_f32)                                                          \
         hipLaunchKernelGGL((bias_act_cast_k<float, A_>), dim3(b),            \
                            dim3(256), 0, s, ws, (float*)C, bias, dact, n,    \
-                           N);                                               \
+                           N, splits);                                       \
       else                                                                   \
         hipLaunchKernelGGL((bias_act_cast_k<unsigned short, A_>), dim3(b),   \
                            dim3(256), 0, s, ws, (unsigned short*)C, bias,    \
-                           dact, n, N);                                      \
+                           dact, n, N, splits);                              \
     } while (0)
     ACT_SWITCH(act, FIN);
 #undef FIN
