@@ -1587,7 +1587,11 @@ bool launch_gemm_tn2(const LA& la, const LB& lb, float* C, int M, int N,
   if (bmt == BM) bias_out = nullptr;
   const int tiles = cdiv(M, bmt) * tn;
   const int nk = cdiv(K, TBK);
-  const int splits = max(1, min(nk, g_tn2_split_budget / tiles));
+  // each split must carry >= 4 K-tiles: tiny-K wgrads (K = batch,
+  // nk ~ 8) otherwise split into atomic-heavy slivers (fc8 wgrad
+  // measured 23 TF/s with 2-tile splits)
+  const int splits =
+      max(1, min(min(nk, g_tn2_split_budget / tiles), max(1, nk / 4)));
   if (splits > 1) {
     if (beta == 0.f) {
       void zero_f32_launch(float*, long, hipStream_t);
