@@ -1,0 +1,89 @@
+"""Serving path: snapshot -> InferenceEngine -> predictions (and the
+HTTP JSON front end) on the numpy backend; the GPU graph-replay lane
+lives in test_serving_gpu.py."""
+
+import json
+import threading
+import urllib.request
+
+import numpy
+import pytest
+
+from veles import prng
+from veles.backends import NumpyDevice
+from veles.config import root
+from veles.dummy import DummyLauncher
+
+
+def _train_and_snapshot(tmp_path):
+    prng.seed_all(99)
+    import veles.znicz_amd.models.wine  # noqa: F401
+    root.wine.decision.update({"max_epochs": 6, "fail_iterations": 50})
+    root.wine.loader.update({"n_train": 90, "n_valid": 30,
+                             "minibatch_size": 30})
+    from veles.znicz_amd.models.wine import WineWorkflow
+    wf = WineWorkflow(DummyLauncher())
+    wf.initialize(device=NumpyDevice())
+    wf.run()
+    from veles.snapshotter import SnapshotterToFile
+    snap = SnapshotterToFile(wf, prefix="srv", directory=str(tmp_path),
+                             compression="gz")
+    wf.del_ref(snap)
+    return snap.export(), wf
+
+
+def test_inference_engine_matches_workflow(tmp_path):
+    path, wf = _train_and_snapshot(tmp_path)
+    from veles.serving import InferenceEngine
+    eng = InferenceEngine(path, device=NumpyDevice())
+    rng = numpy.random.RandomState(3)
+    x = rng.normal(0, 1, (7,) + tuple(
+        eng._in.shape[1:])).astype(numpy.float32)
+    probs = eng.predict(x)
+    assert probs.shape[0] == 7
+    # cross-check against the trained workflow's own forward chain
+    wf.loader.minibatch_data.map_invalidate()
+    wf.loader.minibatch_data.mem[:7] = x
+    wf.loader.minibatch_data.mem[7:] = 0
+    for f in wf.forwards:
+        f.run()
+    wf.forwards[-1].output.map_read()
+    want = wf.forwards[-1].output.mem[:7]
+    numpy.testing.assert_allclose(probs.reshape(7, -1),
+                                  want.reshape(7, -1), rtol=1e-5,
+                                  atol=1e-6)
+    # chunking: n > batch_size goes through multiple passes
+    xn = rng.normal(0, 1, (75,) + tuple(
+        eng._in.shape[1:])).astype(numpy.float32)
+    out = eng.predict(xn)
+    assert out.shape[0] == 75
+    labels = eng.predict_labels(xn)
+    assert labels.shape == (75,) and labels.max() < out.shape[-1]
+
+
+def test_serving_http(tmp_path):
+    path, _ = _train_and_snapshot(tmp_path)
+    from veles.serving import InferenceEngine, serve
+    eng = InferenceEngine(path, device=NumpyDevice())
+    import socket
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    t = threading.Thread(target=serve, args=(eng, port), daemon=True)
+    t.start()
+    import time
+    x = numpy.zeros((2,) + tuple(eng._in.shape[1:]),
+                    numpy.float32).tolist()
+    req = urllib.request.Request(
+        "http://127.0.0.1:%d/" % port,
+        data=json.dumps({"inputs": x, "labels_only": True}).encode(),
+        headers={"Content-Type": "application/json"})
+    for attempt in range(20):
+        try:
+            with urllib.request.urlopen(req, timeout=10) as r:
+                resp = json.loads(r.read())
+            break
+        except (ConnectionError, OSError):
+            time.sleep(0.2)
+    assert "labels" in resp and len(resp["labels"]) == 2

@@ -1,0 +1,170 @@
+"""Forward-only inference over trained snapshots (serving path).
+
+The reference ecosystem shipped separate C++ engines (libVeles/libZnicz)
+that loaded Veles snapshots for inference; SURVEY §2b scoped those out
+of the training engine but kept the snapshot format loadable.  This
+module is the MI355X-native serving analog: load a pickled Workflow,
+keep only its forward chain, feed batches straight into the first
+unit's input buffer (no Loader), and on a GPU replay the whole forward
+pass as ONE captured hipGraph per batch — decode-style serving is
+launch-bound, exactly the regime graph replay removes.
+
+    from veles.serving import InferenceEngine
+    eng = InferenceEngine("snap.4.pickle.gz")      # auto device
+    probs = eng.predict(batch_ndarray)             # [n, n_classes]
+    labels = probs.argmax(1)
+
+    python -m veles.serving snap.pickle.gz --port 8080   # HTTP JSON
+"""
+
+import json
+import threading
+
+import numpy
+
+
+class InferenceEngine(object):
+    """Snapshot → forward-only executor with fixed-batch graph replay."""
+
+    def __init__(self, snapshot_path, device=None, batch_size=None):
+        from veles.snapshotter import SnapshotterBase
+        wf = SnapshotterBase.import_(snapshot_path)
+        if not getattr(wf, "forwards", None):
+            raise ValueError("snapshot has no forward chain (%r)" %
+                             snapshot_path)
+        self.wf = wf
+        self.forwards = list(wf.forwards)
+        if device is None:
+            from veles.backends import Device
+            device = Device()
+        self.device = device
+        # initialize ONLY what the forward pass needs: the loader's
+        # minibatch buffer (the first forward's input) plus the chain
+        wf.loader.initialize(device=device)
+        for f in self.forwards:
+            f.forward_mode = True      # inference: dropout = identity
+            f.initialize(device=device)
+        self.batch_size = batch_size or int(wf.loader.max_minibatch_size)
+        self._in = wf.loader.minibatch_data
+        self._graph = None
+        self._lock = threading.Lock()
+
+    # -- core ---------------------------------------------------------------
+    def _forward(self):
+        for f in self.forwards:
+            f.run()
+
+    def _run_device(self, x):
+        import torch
+        n = x.shape[0]
+        dev = self._in.devmem
+        host = torch.from_numpy(
+            numpy.ascontiguousarray(x, numpy.float32))
+        dev[:n].copy_(host.to(dev.dtype), non_blocking=True)
+        if n < dev.shape[0]:
+            dev[n:] = 0
+        if self._graph is None:
+            # warm up once eagerly, then capture the forward chain
+            self._forward()
+            try:
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    self._forward()
+                g.replay()             # capture records, doesn't run
+                self._graph = g
+            except Exception:
+                self._graph = False    # eager fallback, permanently
+                self._forward()
+        elif self._graph is False:
+            self._forward()
+        else:
+            self._graph.replay()
+        out = self.forwards[-1].output
+        out.touch_device()
+        out.map_read()
+        return out.mem[:n].copy()
+
+    def _run_numpy(self, x):
+        self._in.map_invalidate()
+        n = x.shape[0]
+        self._in.mem[:n] = x
+        if n < self._in.mem.shape[0]:
+            self._in.mem[n:] = 0
+        self._forward()
+        out = self.forwards[-1].output
+        out.map_read()
+        return out.mem[:n].copy()
+
+    def predict(self, x):
+        """x: [n, ...sample shape] float array → [n, ...output] fp32.
+        n may exceed the engine batch; inputs are chunked."""
+        x = numpy.asarray(x, numpy.float32)
+        if x.shape[1:] != tuple(self._in.shape[1:]):
+            raise ValueError("sample shape %s != model input %s" %
+                             (x.shape[1:], tuple(self._in.shape[1:])))
+        outs = []
+        with self._lock:
+            on_gpu = self._in.devmem is not None and self._in.on_gpu
+            for i in range(0, x.shape[0], self.batch_size):
+                chunk = x[i:i + self.batch_size]
+                outs.append(self._run_device(chunk) if on_gpu
+                            else self._run_numpy(chunk))
+        return numpy.concatenate(outs) if len(outs) > 1 else outs[0]
+
+    def predict_labels(self, x):
+        return self.predict(x).reshape(x.shape[0], -1).argmax(axis=1)
+
+
+# -- minimal HTTP JSON front end (same dependency-light style as
+# web_status.py; fastapi/uvicorn exist in the image but add nothing
+# here) ----------------------------------------------------------------------
+
+def serve(engine, port=8080):
+    from http.server import BaseHTTPRequestHandler, HTTPServer
+
+    class Handler(BaseHTTPRequestHandler):
+        def do_POST(self):
+            try:
+                n = int(self.headers.get("Content-Length", 0))
+                req = json.loads(self.rfile.read(n))
+                x = numpy.asarray(req["inputs"], numpy.float32)
+                want_labels = req.get("labels_only", False)
+                if want_labels:
+                    resp = {"labels":
+                            engine.predict_labels(x).tolist()}
+                else:
+                    resp = {"outputs": engine.predict(x).tolist()}
+                body = json.dumps(resp).encode()
+                self.send_response(200)
+            except Exception as e:
+                body = json.dumps({"error": repr(e)}).encode()
+                self.send_response(400)
+            self.send_header("Content-Type", "application/json")
+            self.send_header("Content-Length", str(len(body)))
+            self.end_headers()
+            self.wfile.write(body)
+
+        def log_message(self, *a):
+            pass
+
+    srv = HTTPServer(("0.0.0.0", port), Handler)
+    srv.serve_forever()
+
+
+def main(argv=None):
+    import argparse
+    p = argparse.ArgumentParser(
+        prog="veles.serving",
+        description="serve a trained Veles snapshot over HTTP JSON")
+    p.add_argument("snapshot")
+    p.add_argument("--port", type=int, default=8080)
+    p.add_argument("--batch", type=int, default=None)
+    args = p.parse_args(argv)
+    eng = InferenceEngine(args.snapshot, batch_size=args.batch)
+    print("serving %s on :%d (batch %d)" % (args.snapshot, args.port,
+                                            eng.batch_size))
+    serve(eng, args.port)
+
+
+if __name__ == "__main__":
+    main()
