@@ -58,9 +58,17 @@ class InferenceEngine(object):
         import torch
         n = x.shape[0]
         dev = self._in.devmem
-        host = torch.from_numpy(
-            numpy.ascontiguousarray(x, numpy.float32))
-        dev[:n].copy_(host.to(dev.dtype), non_blocking=True)
+        # one pinned buffer in the DEVICE dtype: torch's cross-dtype
+        # host copy converts multithreaded (~600 GB/s measured), then a
+        # single H2D in the small dtype — the fp32-staging variant
+        # cost ~40 ms/batch in host copies for AlexNet-sized inputs
+        if getattr(self, "_pin", None) is None or \
+                self._pin.dtype != dev.dtype:
+            self._pin = torch.empty(dev.shape, dtype=dev.dtype,
+                                    pin_memory=True)
+        self._pin[:n].copy_(torch.from_numpy(
+            numpy.ascontiguousarray(x, numpy.float32)))
+        dev[:n].copy_(self._pin[:n], non_blocking=True)
         if n < dev.shape[0]:
             dev[n:] = 0
         if self._graph is None:
@@ -81,8 +89,14 @@ class InferenceEngine(object):
             self._graph.replay()
         out = self.forwards[-1].output
         out.touch_device()
-        out.map_read()
-        return out.mem[:n].copy()
+        # pinned D2H + host fp32 view: Array.map_read's pageable
+        # `.to("cpu")` measured ~39 ms/call on this ROCm build
+        if getattr(self, "_out_pin", None) is None:
+            self._out_pin = torch.empty(out.devmem.shape,
+                                        dtype=out.devmem.dtype,
+                                        pin_memory=True)
+        self._out_pin.copy_(out.devmem)
+        return self._out_pin[:n].float().numpy()
 
     def _run_numpy(self, x):
         self._in.map_invalidate()
