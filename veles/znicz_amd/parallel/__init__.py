@@ -141,7 +141,8 @@ class GradientReducer(object):
         if bucket_bytes is None:
             # tunable against the 7-link xGMI per-link bound without a
             # code change (the driver's 8-GPU scaling runs)
-            bucket_bytes = int(os.environ.get("VELES_BUCKET_MB", 25))                 << 20
+            bucket_bytes = \
+                int(os.environ.get("VELES_BUCKET_MB", 25)) << 20
         self.ctx = ctx
         self.device = device
         self.bucket_bytes = bucket_bytes
