@@ -1800,6 +1800,101 @@ static unsigned short* materialize_im2col(const unsigned short* x,
   return ws;
 }
 
+// ---------------------------------------------------------------------------
+// strided dgrad via residue classes (one dense GEMM per (a, b) class +
+// an interleave scatter) — see loaders.h DgradNTR.  The generic
+// strided gather paid SY·SX× MFMA work (conv1-s4 dgrad measured ~21 ms
+// standalone; AE-family stride-2 deconvs 4×).
+// ---------------------------------------------------------------------------
+__global__ void dgrad_scatter_k(const unsigned short* __restrict__ src,
+                                unsigned short* __restrict__ ei,
+                                ConvDesc d, DgradClass c, int nunits,
+                                DivM mCu, DivM mWcu) {
+  const unsigned u = blockIdx.x * blockDim.x + threadIdx.x;
+  if (u >= (unsigned)nunits) return;
+  const unsigned row = divq(u, mCu);          // class-local pixel
+  const unsigned cu = u - row * (unsigned)(d.C >> 2);
+  unsigned t2 = divq(row, mWcu);
+  unsigned iwr = row - t2 * c.Wc;
+  unsigned n = divq(t2, c.mHc);
+  unsigned ihr = t2 - n * c.Hc;
+  const long dst = (((long)n * d.H + c.ih0 + ihr * d.SY) * d.W +
+                    c.iw0 + iwr * d.SX) * d.C + cu * 4;
+  *(unsigned long long*)(ei + dst) =
+      src ? *(const unsigned long long*)(src + (long)u * 4) : 0ull;
+}
+
+static unsigned short* dgrad_ws_buf(size_t bytes) {
+  static std::mutex mu;
+  static unsigned short* ws = nullptr;
+  static size_t cap = 0;
+  std::lock_guard<std::mutex> lk(mu);
+  if (bytes > cap) {
+    if (ws) (void)hipFree(ws);
+    cap = bytes;
+    if (hipMalloc(&ws, cap) != hipSuccess) {
+      ws = nullptr;
+      cap = 0;
+      throw std::runtime_error("dgrad residue workspace alloc");
+    }
+  }
+  return ws;
+}
+
+static void conv_dgrad_residue(const unsigned short* eo,
+                               const unsigned short* wdg,
+                               unsigned short* ei,
+                               const unsigned short* zero,
+                               const ConvDesc& d, hipStream_t s) {
+  const int N = d.C;
+  unsigned short* wsb =
+      dgrad_ws_buf((size_t)d.N * d.H * d.W * N * 2);
+  long off = 0;
+  struct ClassRec { DgradClass c; long off; long mc; bool has; };
+  ClassRec recs[64];
+  int nrec = 0;
+  for (int a = 0; a < d.SY; ++a)
+    for (int b = 0; b < d.SX; ++b) {
+      DgradClass c;
+      c.a = a;
+      c.b = b;
+      c.ih0 = ((a - d.PT) % d.SY + d.SY) % d.SY;
+      c.iw0 = ((b - d.PL) % d.SX + d.SX) % d.SX;
+      if (c.ih0 >= d.H || c.iw0 >= d.W) continue;
+      c.Hc = (d.H - c.ih0 + d.SY - 1) / d.SY;
+      c.Wc = (d.W - c.iw0 + d.SX - 1) / d.SX;
+      c.KHc = a < d.KH ? (d.KH - a + d.SY - 1) / d.SY : 0;
+      c.KWc = b < d.KW ? (d.KW - b + d.SX - 1) / d.SX : 0;
+      c.ohb0 = (c.ih0 + d.PT - a) / d.SY;
+      c.owb0 = (c.iw0 + d.PL - b) / d.SX;
+      c.mWc = make_divm((unsigned)c.Wc);
+      c.mHc = make_divm((unsigned)c.Hc);
+      c.mKWc = make_divm((unsigned)(c.KWc > 0 ? c.KWc : 1));
+      const long Mc = (long)d.N * c.Hc * c.Wc;
+      const bool has = c.KHc > 0 && c.KWc > 0;
+      if (has) {
+        const int Kc = c.KHc * c.KWc * d.K;   // d.K % 8 == 0 upstream
+        DgradNTR la{eo, zero, d, c};
+        DenseDgradR lb{wdg, zero, d, c};
+        launch_gemm_nt(la, lb, wsb + off, (const float*)nullptr,
+                       (int)Mc, N, Kc, ACT_LINEAR, s);
+      }
+      recs[nrec++] = ClassRec{c, off, Mc, has};
+      off += Mc * N;
+    }
+  for (int i = 0; i < nrec; ++i) {
+    const ClassRec& r = recs[i];
+    const int nunits = (int)(r.mc * (N >> 2));
+    if (nunits <= 0) continue;
+    dim3 grid((nunits + 255) / 256), block(256);
+    hipLaunchKernelGGL(dgrad_scatter_k, grid, block, 0, s,
+                       r.has ? wsb + r.off : nullptr, ei, d, r.c,
+                       nunits, make_divm((unsigned)(N >> 2)),
+                       r.c.mWc);
+    HIP_CHECK_LAST();
+  }
+}
+
 void conv_fwd_gemm8(const unsigned short*, const unsigned short*, void*,
                     bool, const float*, const unsigned short*,
                     const ConvDesc&, int, hipStream_t);
@@ -1909,6 +2004,11 @@ void conv_dgrad_gemm(const unsigned short* eo, const unsigned short* wdg,
     }
     launch_gemm_nt(la, lb, ei, (const float*)nullptr, M, N, K, act, s,
                    dact);
+    return;
+  }
+  if ((d.K & 7) == 0 && (d.C & 3) == 0 && d.SY < 8 && d.SX < 8 &&
+      !dact) {
+    conv_dgrad_residue(eo, wdg, ei, zero, d, s);
     return;
   }
   DgradNT la{eo, zero, d};

@@ -343,3 +343,99 @@ static __device__ __forceinline__ int xcd_tile(int b, int ntiles) {
   return (b & 7) * per + (b >> 3);
 }
 
+
+// ---------------------------------------------------------------------------
+// Residue-class strided dgrad (SY or SX > 1): the generic DgradNT pays
+// SY·SX× MFMA work on taps whose stride check fails (zero-page reads).
+// Decomposing input pixels by (ih+PT) % SY, (iw+PL) % SX makes every
+// tap of a class DENSE: kh = a + kh'·SY, kw = b + kw'·SX, and the
+// oh/ow arithmetic collapses to adds.  One GEMM per class into a
+// contiguous buffer; an interleave kernel scatters classes into ei.
+// ---------------------------------------------------------------------------
+struct DgradClass {
+  int ih0, iw0;        // first input row/col of the class
+  int Hc, Wc;          // class spatial extent
+  int KHc, KWc;        // taps per class (kh = a + kh'*SY, ...)
+  int a, b;            // residues
+  int ohb0, owb0;      // (ih0+PT-a)/SY, (iw0+PL-b)/SX
+  DivM mWc, mHc, mKWc; // class-local magic divisors
+};
+
+struct DgradNTR {
+  static constexpr bool MAY_SLOW = false;
+  static constexpr bool N96_OK = false;
+  const unsigned short* eo;
+  const unsigned short* zero;
+  ConvDesc d;
+  DgradClass c;
+  struct RowC { long off0; int ohm, owm; };
+  struct KC { int co, kwp, khp; };
+  __device__ RowC rowc(int row) const {
+    unsigned t2 = divq((unsigned)row, c.mWc);
+    unsigned iwr = (unsigned)row - t2 * c.Wc;
+    unsigned n = divq(t2, c.mHc);
+    unsigned ihr = t2 - n * c.Hc;
+    const int ohm = c.ohb0 + (int)ihr;
+    const int owm = c.owb0 + (int)iwr;
+    return RowC{(((long)n * d.OH + ohm) * d.OW + owm) * d.K, ohm, owm};
+  }
+  __device__ KC kc(int k8) const {
+    unsigned k = (unsigned)k8 * 8;
+    unsigned t = divq(k, d.mK);
+    unsigned co = k - t * d.K;
+    unsigned khp = divq(t, c.mKWc);
+    unsigned kwp = t - khp * c.KWc;
+    return KC{(int)co, (int)kwp, (int)khp};
+  }
+  __device__ const void* addr_rk(const RowC& rc, const KC& kk) const {
+    const int oh = rc.ohm - kk.khp;
+    const int ow = rc.owm - kk.kwp;
+    // khp >= KHc <=> k >= Kc: the GEMM K is padded to a BK multiple
+    // and the tail must read zeros (decoded khp would alias to VALID
+    // oh values at high rows otherwise)
+    if (kk.khp >= c.KHc || (unsigned)oh >= (unsigned)d.OH ||
+        (unsigned)ow >= (unsigned)d.OW)
+      return zero;
+    return eo + rc.off0 - ((long)kk.khp * d.OW + kk.kwp) * d.K + kk.co;
+  }
+  __device__ const void* addr(int row, int k8) const {
+    return addr_rk(rowc(row), kc(k8));
+  }
+  __device__ unsigned short elem(int row, int k) const {  // K%8 never
+    return 0;
+  }
+};
+
+// W-dgrad image [C][KH][KW][K] restricted to a class's taps: a 16-B
+// chunk is 8 consecutive co at fixed (kh, kw) — contiguous (K % 8 == 0)
+struct DenseDgradR {
+  static constexpr bool MAY_SLOW = false;
+  static constexpr bool N96_OK = false;
+  const unsigned short* w;
+  const unsigned short* zero;
+  ConvDesc d;
+  DgradClass c;
+  struct RowC { long rbase; };
+  struct KC { int off; };
+  __device__ RowC rowc(int row) const {
+    return RowC{(long)row * d.KH * d.KW * d.K};
+  }
+  __device__ KC kc(int k8) const {
+    unsigned k = (unsigned)k8 * 8;
+    unsigned t = divq(k, d.mK);
+    unsigned co = k - t * d.K;
+    unsigned khp = divq(t, c.mKWc);
+    unsigned kwp = t - khp * c.KWc;
+    if ((int)khp >= c.KHc) return KC{-1};       // padded K tail
+    const int kh = c.a + (int)khp * d.SY;
+    const int kw = c.b + (int)kwp * d.SX;
+    return KC{(int)(((long)kh * d.KW + kw) * d.K + co)};
+  }
+  __device__ const void* addr_rk(const RowC& rc, const KC& kk) const {
+    return kk.off < 0 ? (const void*)zero : (const void*)(w + rc.rbase + kk.off);
+  }
+  __device__ const void* addr(int row, int k8) const {
+    return addr_rk(rowc(row), kc(k8));
+  }
+  __device__ unsigned short elem(int row, int k) const { return 0; }
+};
