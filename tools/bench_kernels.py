@@ -92,14 +92,9 @@ def bench_convs():
             x.view(N, -1), w, y.view(N * OH * OW, K), bias, desc, 3))
         report(nm + " fwd", us, flops)
         wdg = wd.view(C, KH * KW * K)
-        if SY == 1:
-            # stride>1 first-layer dgrad is never computed in training
-            # (need_err_input False) and the generic strided gather
-            # wastes SY*SX x MFMA work — see ROADMAP residue item
-            us = timeit(lambda: ops.conv_dgrad(
-                eo.view(N * OH * OW, K), wdg, ei.view(N * H * W, C),
-                desc))
-            report(nm + " dgrad", us, flops)
+        us = timeit(lambda: ops.conv_dgrad(
+            eo.view(N * OH * OW, K), wdg, ei.view(N * H * W, C), desc))
+        report(nm + " dgrad", us, flops)
         us = timeit(lambda: ops.conv_wgrad(
             eo.view(N * OH * OW, K), x.view(N, -1), gw, desc, 0.0))
         report(nm + " wgrad", us, flops)
