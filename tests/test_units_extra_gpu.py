@@ -338,11 +338,20 @@ def test_pooling_modes_gpu():
     u.output.map_read()
     u.input_offset.map_read()
     out = u.output.mem
-    offs = u.input_offset.mem.reshape(-1)
-    flat_in = data.reshape(-1)
-    picked = flat_in[offs]
-    # chosen values match the recorded offsets (bf16 rounding tolerance)
-    numpy.testing.assert_allclose(out.reshape(-1), picked, atol=0.02)
+    # device offsets are u8 RELATIVE window codes r = ky*KX + kx
+    offs = u.input_offset.mem.astype(numpy.int64)
+    Np, OHp, OWp, C = out.shape
+    picked = numpy.empty_like(out)
+    for n in range(Np):
+        for oy in range(OHp):
+            for ox in range(OWp):
+                for c in range(C):
+                    r = offs[n, oy, ox, c]
+                    ky, kx = divmod(int(r), 2)
+                    picked[n, oy, ox, c] = data[n, oy * 2 + ky,
+                                                ox * 2 + kx, c]
+    # chosen values match the recorded codes (bf16 rounding tolerance)
+    numpy.testing.assert_allclose(out, picked, atol=0.02)
     # positive-biased sampling: mean of picks > mean of inputs
     assert picked.mean() > data.mean() + 0.1
 

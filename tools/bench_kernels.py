@@ -112,7 +112,7 @@ def bench_pools():
         OW = OH
         x = torch.randn(N, H, W, C, device=DEV).bfloat16()
         y = torch.empty(N, OH, OW, C, device=DEV, dtype=torch.bfloat16)
-        offs = torch.empty(N * OH * OW * C, device=DEV, dtype=torch.int32)
+        offs = torch.empty(N * OH * OW * C, device=DEV, dtype=torch.uint8)
         ey = torch.randn(N, OH, OW, C, device=DEV).bfloat16()
         ex = torch.empty_like(x)
         desc = [N, H, W, C, k, k, s, s, OH, OW, 0]

@@ -27,6 +27,9 @@ class Depooling(Forward):
         super(Depooling, self).__init__(workflow, **kwargs)
         self.input_offset = None        # linked from the MaxPooling
         self.output_shape_source = kwargs.get("output_shape_source", None)
+        #: the paired pooling unit (GPU path needs its geometry: device
+        #: offsets are u8 relative window codes, not absolute indices)
+        self.pool_unit = kwargs.get("pool_unit", None)
 
     @property
     def has_weights(self):
@@ -59,10 +62,14 @@ class Depooling(Forward):
 
     def hip_run(self):
         from veles.znicz_amd import ops
-        ops.scatter_by_offset(
+        if self.pool_unit is None:
+            raise RuntimeError(
+                "Depooling on the HIP path needs pool_unit (the paired "
+                "pooling) — device offsets are relative window codes")
+        ops.depool_scatter(
             ops.as2d_bf16(self.input).view(-1),
             self.input_offset.devmem.view(-1),
-            self.output.devmem.view(-1))
+            self.output.devmem.view(-1), self.pool_unit.pool_desc())
         self.output.touch_device()
 
 
@@ -76,6 +83,7 @@ class GDDepooling(GradientDescentBase):
         kwargs["include_bias"] = False
         super(GDDepooling, self).__init__(workflow, **kwargs)
         self.input_offset = None        # linked from the same MaxPooling
+        self.pool_unit = kwargs.get("pool_unit", None)
 
     LINK_FROM_FORWARD = ("input_offset",)
 
@@ -109,8 +117,11 @@ class GDDepooling(GradientDescentBase):
         from veles.znicz_amd import ops
         if not self.need_err_input:
             return
-        ops.gather_by_offset(
+        if self.pool_unit is None:
+            raise RuntimeError(
+                "GDDepooling on the HIP path needs pool_unit")
+        ops.depool_gather(
             ops.as2d_bf16(self.err_output).view(-1),
             self.input_offset.devmem.view(-1),
-            self.err_input.devmem.view(-1))
+            self.err_input.devmem.view(-1), self.pool_unit.pool_desc())
         self.err_input.touch_device()

@@ -66,6 +66,7 @@ class MnistAEWorkflow(NNWorkflow):
         self.depool.link_attrs(self.pool, ("input", "output"),
                                "input_offset")
         self.depool.output_shape_source = self.pool.input
+        self.depool.pool_unit = self.pool
         self.forwards.append(self.depool)
 
         self.deconv = Deconv(self, n_kernels=k, kx=kx, ky=kx,
@@ -109,6 +110,7 @@ class MnistAEWorkflow(NNWorkflow):
         self.gd_depool.link_from(self.gd_deconv)
         self.gd_depool.link_attrs(self.depool, "input", "output",
                                   "input_offset")
+        self.gd_depool.pool_unit = self.pool
         self.gd_depool.link_attrs(self.gd_deconv,
                                   ("err_output", "err_input"))
         self.gd_depool.link_attrs(self.loader,

@@ -359,6 +359,17 @@ def gather_by_offset(src, offs, dst):
     require().gather_by_offset(src, offs, dst)
 
 
+def depool_scatter(src, offs_u8, dst, pool_desc):
+    """Unpooling forward over u8 relative winner codes: dst (large,
+    zeroed by the kernel) gets src at each stored window position."""
+    require().depool_scatter(src, offs_u8, dst, pool_desc)
+
+
+def depool_gather(src, offs_u8, dst, pool_desc):
+    """Unpooling backward: dst[o] = src[winner(o)] (0 when none)."""
+    require().depool_gather(src, offs_u8, dst, pool_desc)
+
+
 def scatter_by_offset(src, offs, dst):
     """dst zeroed then dst[offs[i]] = src[i] (unique offsets)."""
     require().scatter_by_offset(src, offs, dst)

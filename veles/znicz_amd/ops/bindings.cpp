@@ -66,10 +66,15 @@ void eval_mse_launch(const unsigned short*, const unsigned short*,
                      unsigned short*, float*, float*, int, int, int, int,
                      hipStream_t);
 // from pool_lrn.hip
-void pool_fwd_launch(const unsigned short*, unsigned short*, int*,
-                     const PoolDesc&, int, hipStream_t);
-void pool_bwd_launch(const unsigned short*, unsigned short*, const int*,
-                     const PoolDesc&, int, hipStream_t);
+void pool_fwd_launch(const unsigned short*, unsigned short*,
+                     unsigned char*, const PoolDesc&, int, hipStream_t);
+void depool_scatter_launch(const unsigned short*, const unsigned char*,
+                           unsigned short*, const PoolDesc&, hipStream_t);
+void depool_gather_launch(const unsigned short*, const unsigned char*,
+                          unsigned short*, const PoolDesc&, hipStream_t);
+void pool_bwd_launch(const unsigned short*, unsigned short*,
+                     const unsigned char*, const PoolDesc&, int,
+                     hipStream_t);
 void lrn_fwd_launch(const unsigned short*, unsigned short*, float*, long,
                     int, float, float, float, int, hipStream_t);
 void lrn_bwd_launch(const unsigned short*, const unsigned short*,
@@ -304,8 +309,11 @@ void pool_fwd(torch::Tensor x, torch::Tensor y, torch::Tensor offs,
   PoolDesc d{(int)v[0], (int)v[1], (int)v[2], (int)v[3], (int)v[4],
              (int)v[5], (int)v[6], (int)v[7], (int)v[8], (int)v[9],
              (unsigned long long)v[10]};
-  pool_fwd_launch(ubf(x), ubf_mut(y), offs.data_ptr<int>(), d, (int)mode,
-                  cur_stream());
+  TORCH_CHECK(offs.dtype() == torch::kUInt8,
+              "pool offsets are u8 relative window codes");
+  TORCH_CHECK(d.KY * d.KX <= 254, "pool window too large for u8 codes");
+  pool_fwd_launch(ubf(x), ubf_mut(y), offs.data_ptr<uint8_t>(), d,
+                  (int)mode, cur_stream());
 }
 
 void pool_bwd(torch::Tensor ey, torch::Tensor ex, torch::Tensor offs,
@@ -314,8 +322,33 @@ void pool_bwd(torch::Tensor ey, torch::Tensor ex, torch::Tensor offs,
   PoolDesc d{(int)v[0], (int)v[1], (int)v[2], (int)v[3], (int)v[4],
              (int)v[5], (int)v[6], (int)v[7], (int)v[8], (int)v[9],
              (unsigned long long)v[10]};
-  pool_bwd_launch(ubf(ey), ubf_mut(ex), offs.data_ptr<int>(), d, (int)mode,
-                  cur_stream());
+  TORCH_CHECK(offs.dtype() == torch::kUInt8,
+              "pool offsets are u8 relative window codes");
+  pool_bwd_launch(ubf(ey), ubf_mut(ex), offs.data_ptr<uint8_t>(), d,
+                  (int)mode, cur_stream());
+}
+
+static PoolDesc make_pool_desc(const std::vector<int64_t>& v) {
+  TORCH_CHECK(v.size() == 11, "pool desc needs 11 ints");
+  return PoolDesc{(int)v[0], (int)v[1], (int)v[2], (int)v[3], (int)v[4],
+                  (int)v[5], (int)v[6], (int)v[7], (int)v[8], (int)v[9],
+                  (unsigned long long)v[10]};
+}
+
+void depool_scatter(torch::Tensor src, torch::Tensor offs,
+                    torch::Tensor dst, std::vector<int64_t> v) {
+  PoolDesc d = make_pool_desc(v);
+  TORCH_CHECK(offs.dtype() == torch::kUInt8, "u8 offsets");
+  depool_scatter_launch(ubf(src), offs.data_ptr<uint8_t>(), ubf_mut(dst),
+                        d, cur_stream());
+}
+
+void depool_gather(torch::Tensor src, torch::Tensor offs,
+                   torch::Tensor dst, std::vector<int64_t> v) {
+  PoolDesc d = make_pool_desc(v);
+  TORCH_CHECK(offs.dtype() == torch::kUInt8, "u8 offsets");
+  depool_gather_launch(ubf(src), offs.data_ptr<uint8_t>(), ubf_mut(dst),
+                       d, cur_stream());
 }
 
 void lrn_fwd(torch::Tensor x, torch::Tensor y,
@@ -457,6 +490,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("eval_mse", &eval_mse);
   m.def("pool_fwd", &pool_fwd);
   m.def("pool_bwd", &pool_bwd);
+  m.def("depool_scatter", &depool_scatter);
+  m.def("depool_gather", &depool_gather);
   m.def("lrn_fwd", &lrn_fwd);
   m.def("lrn_bwd", &lrn_bwd);
   m.def("dropout_fwd", &dropout_fwd);

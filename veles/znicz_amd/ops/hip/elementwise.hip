@@ -706,3 +706,89 @@ void histogram_launch(const void* x, bool is_bf16, long n, float lo,
                        out);
   HIP_CHECK_LAST();
 }
+
+// ---------------------------------------------------------------------------
+// depooling over u8 RELATIVE winner codes (pool offsets are no longer
+// absolute indices): scatter = unpooling forward, gather = its
+// backward; o enumerates the POOL-OUTPUT domain [N,OH,OW,C].
+// ---------------------------------------------------------------------------
+__global__ void depool_scatter_k(const unsigned short* __restrict__ src,
+                                 const unsigned char* __restrict__ offs,
+                                 unsigned short* __restrict__ dst,
+                                 PoolDesc d, long n, DivM mC, DivM mOW,
+                                 DivM mOH, DivM mKX) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long o = i; o < n; o += stride) {
+    const int r = offs[o];
+    if (r == 255) continue;
+    unsigned t = divq((unsigned)o, mC);              // o / C
+    unsigned c = (unsigned)o - t * d.C;
+    unsigned t2 = divq(t, mOW);
+    unsigned ox = t - t2 * d.OW;
+    unsigned nb = divq(t2, mOH);
+    unsigned oy = t2 - nb * d.OH;
+    unsigned ky = divq((unsigned)r, mKX);
+    unsigned kx = (unsigned)r - ky * d.KX;
+    const long dsti =
+        (((long)nb * d.H + oy * d.SY + ky) * d.W + ox * d.SX + kx) *
+            d.C + c;
+    dst[dsti] = src[o];
+  }
+}
+
+__global__ void depool_gather_k(const unsigned short* __restrict__ src,
+                                const unsigned char* __restrict__ offs,
+                                unsigned short* __restrict__ dst,
+                                PoolDesc d, long n, DivM mC, DivM mOW,
+                                DivM mOH, DivM mKX) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long o = i; o < n; o += stride) {
+    const int r = offs[o];
+    if (r == 255) {
+      dst[o] = 0;
+      continue;
+    }
+    unsigned t = divq((unsigned)o, mC);
+    unsigned c = (unsigned)o - t * d.C;
+    unsigned t2 = divq(t, mOW);
+    unsigned ox = t - t2 * d.OW;
+    unsigned nb = divq(t2, mOH);
+    unsigned oy = t2 - nb * d.OH;
+    unsigned ky = divq((unsigned)r, mKX);
+    unsigned kx = (unsigned)r - ky * d.KX;
+    const long srci =
+        (((long)nb * d.H + oy * d.SY + ky) * d.W + ox * d.SX + kx) *
+            d.C + c;
+    dst[o] = src[srci];
+  }
+}
+
+void depool_scatter_launch(const unsigned short* src,
+                           const unsigned char* offs,
+                           unsigned short* dst, const PoolDesc& d,
+                           hipStream_t s) {
+  const long n = (long)d.N * d.OH * d.OW * d.C;
+  const long big = (long)d.N * d.H * d.W * d.C;
+  void zero_f32_launch(float*, long, hipStream_t);
+  zero_f32_launch((float*)dst, big / 2, s);    // bf16 numel is even
+  int blocks = (int)min((n + 255) / 256, (long)4096);
+  hipLaunchKernelGGL(depool_scatter_k, dim3(blocks), dim3(256), 0, s,
+                     src, offs, dst, d, n, make_divm((unsigned)d.C),
+                     make_divm((unsigned)d.OW), make_divm((unsigned)d.OH),
+                     make_divm((unsigned)d.KX));
+  HIP_CHECK_LAST();
+}
+
+void depool_gather_launch(const unsigned short* src,
+                          const unsigned char* offs, unsigned short* dst,
+                          const PoolDesc& d, hipStream_t s) {
+  const long n = (long)d.N * d.OH * d.OW * d.C;
+  int blocks = (int)min((n + 255) / 256, (long)4096);
+  hipLaunchKernelGGL(depool_gather_k, dim3(blocks), dim3(256), 0, s,
+                     src, offs, dst, d, n, make_divm((unsigned)d.C),
+                     make_divm((unsigned)d.OW), make_divm((unsigned)d.OH),
+                     make_divm((unsigned)d.KX));
+  HIP_CHECK_LAST();
+}

@@ -9,13 +9,18 @@
 template <int MODE>
 __device__ __forceinline__ void pool_fwd_one(
     const unsigned short* __restrict__ x, unsigned short* __restrict__ y,
-    int* __restrict__ offs, const PoolDesc& d, long idx, int n, int ih0,
-    int iw0, int c);
+    unsigned char* __restrict__ offs, const PoolDesc& d, long idx, int n,
+    int ih0, int iw0, int c);
 
 template <int MODE>
+// offsets are u8 RELATIVE window codes (ky*KX + kx; 255 = none) —
+// int32 absolute indices cost 4× the offset traffic on both passes,
+// and the backward can reconstruct the absolute position from its own
+// (ih, iw) arithmetic.  AVG mode stores the window tap COUNT instead.
 __global__ void pool_fwd_k(const unsigned short* __restrict__ x,
                            unsigned short* __restrict__ y,
-                           int* __restrict__ offs, PoolDesc d, int lg) {
+                           unsigned char* __restrict__ offs, PoolDesc d,
+                           int lg) {
   const int C2 = 1 << lg;
   const int c0 = threadIdx.x & (C2 - 1);
   const int rin = threadIdx.x >> lg;
@@ -78,6 +83,7 @@ __global__ void pool_fwd_k(const unsigned short* __restrict__ x,
             int iw = iw0 + kx;
             if (iw >= d.W) break;
             long off = (((long)n * d.H + ih) * d.W + iw) * d.C + cb;
+            const int rcode = ky * d.KX + kx;
             unsigned short v[8];
             *(short8*)v = *(const short8*)(x + off);
             ++count;
@@ -89,33 +95,33 @@ __global__ void pool_fwd_k(const unsigned short* __restrict__ x,
               } else if (MODE == POOL_MAX) {
                 if (bestoff[j] < 0 || vj > best[j]) {
                   best[j] = vj;
-                  bestoff[j] = (int)off + j;
+                  bestoff[j] = rcode;
                 }
               } else if (MODE == POOL_MAXABS) {
                 float aj = fabsf(vj);
                 if (bestoff[j] < 0 || aj > besta[j]) {
                   best[j] = vj;
                   besta[j] = aj;
-                  bestoff[j] = (int)off + j;
+                  bestoff[j] = rcode;
                 }
               } else {
                 float pv = fmaxf(vj, 0.f);
                 st_a[j] += pv;
                 if (bestoff[j] < 0 && st_a[j] > st_t[j] && pv > 0.f) {
                   best[j] = vj;
-                  bestoff[j] = (int)off + j;
+                  bestoff[j] = rcode;
                 }
               }
             }
           }
         }
         unsigned short o8[8];
-        int i8[8];
+        unsigned char i8[8];
         if (MODE == POOL_AVG) {
 #pragma unroll
           for (int j = 0; j < 8; ++j) {
             o8[j] = f2bf(sum[j] / max(count, 1));
-            i8[j] = count;
+            i8[j] = (unsigned char)count;
           }
         } else {
           if (MODE == POOL_STOCHASTIC) {
@@ -127,6 +133,7 @@ __global__ void pool_fwd_k(const unsigned short* __restrict__ x,
                 int iw = iw0 + kx;
                 if (iw >= d.W) break;
                 long off = (((long)n * d.H + ih) * d.W + iw) * d.C + cb;
+                const int rcode = ky * d.KX + kx;
                 unsigned short v[8];
                 *(short8*)v = *(const short8*)(x + off);
 #pragma unroll
@@ -135,7 +142,7 @@ __global__ void pool_fwd_k(const unsigned short* __restrict__ x,
                   if (st_a[j] <= 0.f &&
                       (bestoff[j] < 0 || vj > best[j])) {
                     best[j] = vj;
-                    bestoff[j] = (int)off + j;
+                    bestoff[j] = rcode;
                   }
                 }
               }
@@ -144,14 +151,13 @@ __global__ void pool_fwd_k(const unsigned short* __restrict__ x,
 #pragma unroll
           for (int j = 0; j < 8; ++j) {
             o8[j] = f2bf(best[j]);
-            i8[j] = bestoff[j];
+            i8[j] = bestoff[j] < 0 ? 255 : (unsigned char)bestoff[j];
           }
         }
         *(short8*)(y + idx0) = *(short8*)o8;
-        if (offs) {
-          *(int4*)(offs + idx0) = *(int4*)i8;
-          *(int4*)(offs + idx0 + 4) = *(int4*)(i8 + 4);
-        }
+        if (offs)
+          *(unsigned long long*)(offs + idx0) =
+              *(unsigned long long*)i8;
       }
     } else {
       for (int c = c0; c < d.C; c += C2) {
@@ -165,8 +171,8 @@ __global__ void pool_fwd_k(const unsigned short* __restrict__ x,
 template <int MODE>
 __device__ __forceinline__ void pool_fwd_one(
     const unsigned short* __restrict__ x, unsigned short* __restrict__ y,
-    int* __restrict__ offs, const PoolDesc& d, long idx, int n, int ih0,
-    int iw0, int c) {
+    unsigned char* __restrict__ offs, const PoolDesc& d, long idx, int n,
+    int ih0, int iw0, int c) {
   float best = 0.f, besta = -1.f, sum = 0.f;
   int bestoff = -1, count = 0;
   float stoch_target = 0.f, stoch_acc = 0.f;
@@ -192,29 +198,30 @@ __device__ __forceinline__ void pool_fwd_one(
       int iw = iw0 + kx;
       if (iw >= d.W) break;
       long off = (((long)n * d.H + ih) * d.W + iw) * d.C + c;
+      const int rcode = ky * d.KX + kx;
       float v = bf2f(x[off]);
       ++count;
       if (MODE == POOL_AVG) {
         sum += v;
       } else if (MODE == POOL_MAX) {
-        if (bestoff < 0 || v > best) { best = v; bestoff = (int)off; }
+        if (bestoff < 0 || v > best) { best = v; bestoff = rcode; }
       } else if (MODE == POOL_MAXABS) {
         float a = fabsf(v);
         if (bestoff < 0 || a > besta) { best = v; besta = a;
-                                        bestoff = (int)off; }
+                                        bestoff = rcode; }
       } else {  // stochastic: first cell where running pos-sum crosses
         float pv = fmaxf(v, 0.f);
         stoch_acc += pv;
         if (bestoff < 0 && (stoch_acc > stoch_target && pv > 0.f)) {
           best = v;
-          bestoff = (int)off;
+          bestoff = rcode;
         }
       }
     }
   }
   if (MODE == POOL_AVG) {
     y[idx] = f2bf(sum / max(count, 1));
-    if (offs) offs[idx] = count;      // store divisor for backward
+    if (offs) offs[idx] = (unsigned char)count;  // divisor for backward
   } else {
     if (MODE == POOL_STOCHASTIC && bestoff < 0) {
       // all-nonpositive window: fall back to max
@@ -224,14 +231,14 @@ __device__ __forceinline__ void pool_fwd_one(
         for (int kx = 0; kx < d.KX; ++kx) {
           int iw = iw0 + kx;
           if (iw >= d.W) break;
-          long off = (((long)n * d.H + ih) * d.W + iw) * d.C + c;
-          float v = bf2f(x[off]);
-          if (bestoff < 0 || v > best) { best = v; bestoff = (int)off; }
+          float v = bf2f(x[(((long)n * d.H + ih) * d.W + iw) * d.C + c]);
+          if (bestoff < 0 || v > best) { best = v;
+                                         bestoff = ky * d.KX + kx; }
         }
       }
     }
     y[idx] = f2bf(best);
-    if (offs) offs[idx] = bestoff;
+    if (offs) offs[idx] = bestoff < 0 ? 255 : (unsigned char)bestoff;
   }
 }
 
@@ -240,8 +247,8 @@ __device__ __forceinline__ void pool_fwd_one(
 template <int MODE>
 __global__ void pool_bwd_k(const unsigned short* __restrict__ ey,
                            unsigned short* __restrict__ ex,
-                           const int* __restrict__ offs, PoolDesc d,
-                           int lg) {
+                           const unsigned char* __restrict__ offs,
+                           PoolDesc d, int lg) {
   const int C2 = 1 << lg;
   const int c0 = threadIdx.x & (C2 - 1);
   const int rin = threadIdx.x >> lg;
@@ -269,20 +276,21 @@ __global__ void pool_bwd_k(const unsigned short* __restrict__ ey,
             long o = (((long)n * d.OH + oy) * d.OW + ox) * d.C + cb;
             unsigned short e8[8];
             *(short8*)e8 = *(const short8*)(ey + o);
+            unsigned char w8[8];
+            *(unsigned long long*)w8 =
+                *(const unsigned long long*)(offs + o);
             if (MODE == POOL_AVG) {
-              int c8[8];
-              *(int4*)c8 = *(const int4*)(offs + o);
-              *(int4*)(c8 + 4) = *(const int4*)(offs + o + 4);
 #pragma unroll
               for (int j = 0; j < 8; ++j)
-                g[j] += bf2f(e8[j]) / (float)max(c8[j], 1);
+                g[j] += bf2f(e8[j]) / (float)max((int)w8[j], 1);
             } else {
-              int w8[8];
-              *(int4*)w8 = *(const int4*)(offs + o);
-              *(int4*)(w8 + 4) = *(const int4*)(offs + o + 4);
+              // does window (oy, ox)'s stored RELATIVE winner code
+              // point at THIS input pixel?
+              const int rcode = (ih - oy * d.SY) * d.KX +
+                                (iw - ox * d.SX);
 #pragma unroll
               for (int j = 0; j < 8; ++j)
-                if (w8[j] == (int)(idx0 + j)) g[j] += bf2f(e8[j]);
+                if ((int)w8[j] == rcode) g[j] += bf2f(e8[j]);
             }
           }
         unsigned short o8[8];
@@ -298,9 +306,11 @@ __global__ void pool_bwd_k(const unsigned short* __restrict__ ey,
           for (int ox = ox0; ox <= ox1; ++ox) {
             long o = (((long)n * d.OH + oy) * d.OW + ox) * d.C + c;
             if (MODE == POOL_AVG) {
-              g += bf2f(ey[o]) / (float)max(offs[o], 1);
+              g += bf2f(ey[o]) / (float)max((int)offs[o], 1);
             } else {
-              if (offs[o] == (int)idx) g += bf2f(ey[o]);
+              const int rcode = (ih - oy * d.SY) * d.KX +
+                                (iw - ox * d.SX);
+              if ((int)offs[o] == rcode) g += bf2f(ey[o]);
             }
           }
         ex[idx] = f2bf(g);
@@ -309,7 +319,8 @@ __global__ void pool_bwd_k(const unsigned short* __restrict__ ey,
   }
 }
 
-void pool_fwd_launch(const unsigned short* x, unsigned short* y, int* offs,
+void pool_fwd_launch(const unsigned short* x, unsigned short* y,
+                     unsigned char* offs,
                      const PoolDesc& d, int mode, hipStream_t s) {
   const int lg = (d.C & 7) == 0 ? chan_lg(d.C >> 3) : chan_lg(d.C);
   const int rpb = 256 >> lg;
@@ -328,8 +339,8 @@ void pool_fwd_launch(const unsigned short* x, unsigned short* y, int* offs,
 }
 
 void pool_bwd_launch(const unsigned short* ey, unsigned short* ex,
-                     const int* offs, const PoolDesc& d, int mode,
-                     hipStream_t s) {
+                     const unsigned char* offs, const PoolDesc& d,
+                     int mode, hipStream_t s) {
   const int lg = (d.C & 7) == 0 ? chan_lg(d.C >> 3) : chan_lg(d.C);
   const int rpb = 256 >> lg;
   long rows = (long)d.N * d.H * d.W;

@@ -65,7 +65,11 @@ class Pooling(Forward):
         if not self.input_offset or self.input_offset.shape != oshape:
             self.input_offset.reset(numpy.zeros(oshape, numpy.int32))
         self.init_array(self.output)
-        self.init_array(self.input_offset, device_dtype="i32")
+        # device offsets are u8 RELATIVE window codes (ky*KX + kx;
+        # AVG stores the tap count) — 4× less offset traffic than
+        # absolute int32; the numpy path keeps absolute indices in
+        # .mem (per-backend internal state, never mixed)
+        self.init_array(self.input_offset, device_dtype="u8")
 
     def numpy_run(self):
         self.input.map_read()
