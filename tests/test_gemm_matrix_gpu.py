@@ -81,6 +81,8 @@ def test_gemm_nt_fc_matrix(M, N, K):
                                        # ragged M (24*361 % 256 != 0)
     (20, 21, 21, 64, 72, 3, 1, 1),     # deep ntn8 BNT=64 dgrad (M<2^20,
                                        # K=648>=512), ragged tiles
+    (6, 17, 17, 16, 48, 5, 2, 2),      # residue-class strided dgrad
+                                       # WITH padding (deconv-AE shape)
 ])
 def test_conv_fwd_dgrad_vs_torch(N, H, W, C, K, KH, SY, PT):
     KW, SX, PL = KH, SY, PT

@@ -494,3 +494,46 @@ def test_nn_rollback_restores_device_weights():
     d_rolled = numpy.abs(y_rolled - y0).mean()
     d_trained = numpy.abs(y_trained - y0).mean()
     assert d_rolled < d_trained * 0.5, (d_rolled, d_trained)
+
+
+def test_depooling_gpu_matches_numpy():
+    """Depooling + GDDepooling over the u8 relative-offset device path
+    must reproduce the numpy absolute-offset path exactly."""
+    import torch
+    from veles.znicz_amd.pooling import MaxPooling
+    from veles.znicz_amd.depooling import Depooling, GDDepooling
+    data = RNG.uniform(-1, 1, (3, 10, 10, 8)).astype(numpy.float32)
+    err = RNG.uniform(-1, 1, (3, 10, 10, 8)).astype(numpy.float32)
+    outs = {}
+    for kind in ("numpy", "rocm"):
+        dev = NumpyDevice() if kind == "numpy" else ROCmDevice(0)
+        wf = DummyWorkflow()
+        pool = MaxPooling(wf, kx=2, ky=2, sliding=(2, 2))
+        pool.input = Array(data.copy())
+        pool.input.device_dtype = "bf16"
+        pool.input.initialize(dev)
+        assert pool.initialize(device=dev) is not False
+        pool.run()
+        dp = Depooling(wf, output_shape_source=pool.input,
+                       pool_unit=pool)
+        dp.input = pool.output
+        dp.input_offset = pool.input_offset
+        assert dp.initialize(device=dev) is not False
+        dp.run()
+        dp.output.map_read()
+        gd = GDDepooling(wf, pool_unit=pool)
+        gd.input = dp.input
+        gd.output = dp.output
+        gd.input_offset = dp.input_offset
+        from veles.memory import Array as _A
+        gd.err_output = _A(err.copy())
+        gd.err_output.device_dtype = "bf16"
+        gd.err_output.initialize(dev)
+        assert gd.initialize(device=dev) is not False
+        gd.run()
+        gd.err_input.map_read()
+        if kind == "rocm":
+            torch.cuda.synchronize()
+        outs[kind] = (dp.output.mem.copy(), gd.err_input.mem.copy())
+    assert_close(outs["rocm"][0], outs["numpy"][0])
+    assert_close(outs["rocm"][1], outs["numpy"][1])
