@@ -10,8 +10,11 @@
 // around the MFMA bursts, and the bijective XCD workgroup swizzle.
 // ≈1.7× the 128²/2-barrier structure on well-shaped operands.
 //
-// Used for the implicit-im2col conv forward and gather dgrad when the
-// shape fills it (M ≥ 1024, N ≥ 192); other shapes keep the 128² core.
+// STATUS (round 2, measured): on every real conv shape (N ≤ 384) the
+// 256-wide tile's column padding loses more than the schedule gains —
+// kept as an env-gated experiment (VELES_NT8); the counted-vmcnt /
+// deep-prefetch techniques it pioneered live in gemm.hip's 128² and
+// ntn8 cores.  See profiles/perf_r2.md "measured-negative" records.
 #include <stdexcept>
 #include "common.h"
 #include "loaders.h"
