@@ -290,3 +290,31 @@ def test_cli_snapshot_resume(tmp_path):
          "-s", str(snaps[-1])],
         cwd=REPO, capture_output=True, text=True, timeout=600)
     assert out2.returncode == 0, out2.stderr[-2000:]
+
+
+def test_routing_table_loader_robust(tmp_path, monkeypatch):
+    """A corrupt gfx950_routing.json must never block the ops import."""
+    import veles.znicz_amd.ops as ops_mod
+
+    class FakeMod:
+        calls = []
+
+        @staticmethod
+        def set_routing(a, b, c, d):
+            FakeMod.calls.append((a, b, c, d))
+
+    bad = tmp_path / "gfx950_routing.json"
+    bad.write_text("{not json")
+    real_join = os.path.join
+
+    def fake_join(*a):
+        if a and a[-1] == "gfx950_routing.json":
+            return str(bad)
+        return real_join(*a)
+
+    monkeypatch.setattr(os.path, "join", fake_join)
+    ops_mod._apply_routing(FakeMod)          # must not raise
+    assert FakeMod.calls == []
+    bad.write_text('{"nt_split_budget": 640}')
+    ops_mod._apply_routing(FakeMod)
+    assert FakeMod.calls == [(640, -1, -1, -1)]

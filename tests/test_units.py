@@ -1,6 +1,7 @@
 """Unit link/gate semantics (parity: veles/tests/test_units.py)."""
 
 import numpy
+import pytest
 
 from veles.dummy import DummyWorkflow
 from veles.units import Unit, TrivialUnit
@@ -125,3 +126,40 @@ def test_repeater_loop_with_decision_gate():
     wf.run()
     assert counter["n"] == 5
     assert wf.is_finished
+
+
+def test_initialize_surfaces_genuine_typeerror():
+    """a unit whose initialize RAISES TypeError (e.g. a bug
+    mentioning 'unexpected keyword argument') must not be silently
+    retried without kwargs — the signature-based dispatch never
+    swallows unit exceptions."""
+    from veles.workflow import Workflow
+    from veles.units import Unit
+
+    class Buggy(Unit):
+        def initialize(self, device=None, **kwargs):
+            raise TypeError(
+                "got an unexpected keyword argument 'frobnicate'")
+
+    wf = Workflow()
+    Buggy(wf)
+    with pytest.raises(TypeError, match="frobnicate"):
+        wf.initialize(device=None)
+
+
+def test_initialize_filters_unknown_kwargs():
+    """A unit whose initialize() takes no 'snapshot' kwarg still
+    initializes when the workflow-level kwargs carry one."""
+    from veles.workflow import Workflow
+    from veles.units import Unit
+
+    seen = {}
+
+    class Narrow(Unit):
+        def initialize(self, device=None):
+            seen["device"] = device
+
+    wf = Workflow()
+    Narrow(wf)
+    wf.initialize(device="d", snapshot="ignored.pickle")
+    assert seen["device"] == "d"
