@@ -318,3 +318,30 @@ def test_routing_table_loader_robust(tmp_path, monkeypatch):
     bad.write_text('{"nt_split_budget": 640}')
     ops_mod._apply_routing(FakeMod)
     assert FakeMod.calls == [(640, -1, -1, -1)]
+
+
+def test_golden_snapshot_format_stable():
+    """Format-stability guard: the committed round-2 wine snapshot
+    (tests/data/golden_wine.0.pickle.gz) must keep loading, resuming
+    and training in every future revision — the pickle object graph
+    with numpy-only Arrays IS the on-disk contract."""
+    from veles.backends import NumpyDevice
+    from veles.loader.base import TRAIN
+    from veles.snapshotter import SnapshotterToFile
+    path = os.path.join(REPO, "tests", "data",
+                        "golden_wine.0.pickle.gz")
+    wf = SnapshotterToFile.import_(path)
+    # the pickled state carries the trained weights and epoch counters
+    assert int(wf.loader.epoch_number) == 2
+    assert wf.decision.epoch_n_err_pt[TRAIN] == pytest.approx(
+        3.3333333, abs=1e-3)
+    for f in wf.forwards:
+        w = getattr(f, "weights", None)
+        if w is not None and w:
+            assert numpy.abs(w.mem).max() > 0
+    # resume: 2 more epochs on the numpy reference device
+    wf.decision.max_epochs = 4
+    wf.decision.complete <<= False
+    wf.initialize(device=NumpyDevice())
+    wf.run()
+    assert int(wf.loader.epoch_number) == 4
