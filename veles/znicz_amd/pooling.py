@@ -5,8 +5,10 @@ Parity target: reference ``znicz/pooling.py`` (``Pooling``,
 window kx,ky with sliding, max variants record winner offsets into
 ``input_offset`` for the backward pass.
 
-MI355X: one templated NHWC kernel (ops/hip/pool_lrn.hip); offsets are
-int32 flat input indices (avg stores the divisor count instead).
+MI355X: one templated NHWC kernel (ops/hip/pool_lrn.hip); device
+offsets are u8 RELATIVE window codes ky*KX + kx (avg stores the tap
+count) — 4× less offset traffic than absolute indices; the numpy
+reference path keeps absolute int32 in ``.mem``.
 """
 
 import numpy
