@@ -53,6 +53,25 @@ def test_cli_config_file(tmp_path):
         (out.stdout[-500:], out.stderr[-500:])
 
 
+def test_cli_genetics_optimize(tmp_path):
+    # SURVEY §3.5 multi-run tuning call stack: --optimize runs the GA
+    # over Tune() config leaves, one full workflow run per individual
+    cfg = tmp_path / "cfg.py"
+    cfg.write_text("from veles.genetics import Tune\n"
+                   "root.wine.decision.max_epochs = 2\n"
+                   "root.wine.loader.n_train = 60\n"
+                   "root.wine.lr = Tune(0.05, 0.01, 0.2)\n")
+    out = subprocess.run(
+        [sys.executable, "-m", "veles",
+         "veles/znicz_amd/models/wine.py", str(cfg), "-b", "numpy",
+         "--optimize", "2", "--population", "3"],
+        cwd=REPO, capture_output=True, text=True, timeout=600)
+    assert out.returncode == 0, out.stderr[-2000:]
+    text = out.stdout + out.stderr
+    assert "best root.wine.lr" in text, text[-1000:]
+    assert "best fitness" in text, text[-1000:]
+
+
 def test_genetics_ga():
     from veles.config import root
     from veles.genetics import Tune, find_tunes, GeneticsOptimizer

@@ -34,6 +34,12 @@ def parse_args(argv):
                    help="self-spawn N data-parallel ranks (one per GPU)")
     p.add_argument("--dry-run", action="store_true",
                    help="build + initialize the workflow, then exit")
+    p.add_argument("--optimize", type=int, default=0, metavar="GENS",
+                   help="GA hyper-tuning: run GENS generations over "
+                        "Tune()-marked config leaves (each individual "
+                        "= one full workflow run)")
+    p.add_argument("--population", type=int, default=8,
+                   help="GA population size (with --optimize)")
     p.add_argument("--random-seed", type=int, default=None)
     p.add_argument("--dump-graph", default=None, metavar="DOT",
                    help="write the workflow DAG to this .dot file")
@@ -127,7 +133,58 @@ class Main(object):
         if not hasattr(mod, "run"):
             raise SystemExit("%s does not define run(load, main)" %
                              self.args.workflow)
+        if self.args.optimize > 0:
+            return self._run_genetics(mod)
         mod.run(self.load, self.main)
+        return 0
+
+    def _run_genetics(self, mod):
+        """GA hyper-tuning: the reference's multi-run optimization flow
+        (``scripts/velescli.py --optimize`` driving veles/genetics/)
+        in-process: each individual is one full workflow run with the
+        Tune() leaves replaced by its genes; fitness is −best
+        validation error count (−train error %% when the loader has no
+        validation set)."""
+        import logging
+        from veles.config import root
+        from veles.genetics import (find_tunes, set_config_leaf,
+                                    GeneticsOptimizer)
+        from veles.loader.base import TRAIN
+        tunes = find_tunes(root)
+        if not tunes:
+            raise SystemExit(
+                "--optimize: no Tune() leaves in the config; mark "
+                "tunable leaves with veles.genetics.Tune(default, "
+                "min, max) in a config file")
+        log = logging.getLogger("genetics")
+        log.info("tuning %d leaves: %s", len(tunes),
+                 ", ".join(p for p, _ in tunes))
+        args = self.args
+
+        def evaluate(cfg):
+            m = Main(args)             # fresh Launcher + workflow
+            mod.run(m.load, m.main)
+            dec = getattr(m.workflow, "decision", None)
+            if dec is not None and dec.min_validation_n_err is not None:
+                fit = -float(dec.min_validation_n_err)
+            elif dec is not None:
+                fit = -float(dec.epoch_n_err_pt[TRAIN])
+            else:
+                fit = 0.0
+            log.info("evaluated %s -> fitness %.6g",
+                     {k: round(v, 6) for k, v in sorted(cfg.items())},
+                     fit)
+            return fit
+
+        ga = GeneticsOptimizer(tunes, evaluate,
+                               population=args.population,
+                               generations=args.optimize)
+        best = ga.run()
+        for (path, _), gene in zip(tunes, best.genes):
+            set_config_leaf(path, gene)   # root keeps the winner
+            log.info("best %s = %.6g", path, gene)
+        log.info("best fitness = %.6g over %d generations",
+                 best.fitness, len(ga.history))
         return 0
 
 

@@ -122,7 +122,10 @@ class GeneticsOptimizer(object):
                 break
             nxt = pop[:self.elite]
             while len(nxt) < self.population_size:
-                a, b = self.rng.choice(len(pop) // 2, 2, replace=False)
+                # parents from the better half; max(2,·) keeps
+                # choice(n, 2, replace=False) legal for population < 4
+                a, b = self.rng.choice(max(2, len(pop) // 2), 2,
+                                       replace=False)
                 child = Chromosome.crossover(pop[a], pop[b], self.rng)
                 child.mutate(self.tunes, self.rng)
                 nxt.append(child)
