@@ -87,6 +87,27 @@ def test_genetics_ga():
     assert len(ga.history) == 6
 
 
+def _ga_pmap_fitness(cfg):
+    return -(cfg["root.tga2.lr"] - 0.05) ** 2
+
+
+def test_genetics_ga_parallel_map():
+    # pmap evaluates a whole generation concurrently; leaves are NOT
+    # applied to the shared root (each worker applies its own cfg)
+    import multiprocessing
+    from veles.config import root
+    from veles.genetics import Tune, find_tunes, GeneticsOptimizer
+    root.tga2.lr = Tune(0.01, 0.001, 0.1)
+    tunes = find_tunes(root.tga2, "root.tga2")
+    with multiprocessing.Pool(2) as pool:
+        ga = GeneticsOptimizer(tunes, _ga_pmap_fitness, population=8,
+                               generations=4, pmap=pool.map)
+        best = ga.run()
+    assert abs(best.genes[0] - 0.05) < 0.03
+    # pmap mode must leave the Tune marker in place (no root mutation)
+    assert isinstance(root.tga2.lr, Tune)
+
+
 def test_ensemble():
     from veles.ensemble import EnsembleTrainer, majority_vote
     models = EnsembleTrainer(lambda i, seed: ("m%d" % i, seed),

@@ -82,7 +82,7 @@ class GeneticsOptimizer(object):
     """
 
     def __init__(self, tunes, evaluate, population=8, generations=5,
-                 elite=2, seed=0xBEEF, apply_leaves=True):
+                 elite=2, seed=0xBEEF, apply_leaves=True, pmap=None):
         self.tunes = list(tunes)
         self.evaluate = evaluate
         self.population_size = population
@@ -90,6 +90,12 @@ class GeneticsOptimizer(object):
         self.elite = elite
         self.rng = numpy.random.RandomState(seed)
         self.apply_leaves = apply_leaves
+        # pmap(func, [cfg, ...]) -> [fitness, ...]: evaluates a whole
+        # generation concurrently (e.g. multiprocessing.Pool.map with
+        # a module-level evaluate).  With pmap the leaves are NOT
+        # applied to the global root here — each worker must apply its
+        # own cfg dict (mutating shared root would race).
+        self.pmap = pmap
         self.best = None
         self.history = []
 
@@ -98,21 +104,33 @@ class GeneticsOptimizer(object):
                  for _, t in self.tunes]
         return Chromosome(genes)
 
+    def _cfg(self, chrom):
+        return {path: g for (path, _), g in zip(self.tunes, chrom.genes)}
+
     def _eval(self, chrom):
-        cfg = {path: g for (path, _), g in zip(self.tunes, chrom.genes)}
+        cfg = self._cfg(chrom)
         if self.apply_leaves:
             for path, val in cfg.items():
                 set_config_leaf(path, val)
         chrom.fitness = float(self.evaluate(cfg))
         return chrom.fitness
 
+    def _eval_generation(self, pop):
+        todo = [c for c in pop if c.fitness is None]
+        if self.pmap is not None and len(todo) > 1:
+            fits = list(self.pmap(self.evaluate,
+                                  [self._cfg(c) for c in todo]))
+            for c, f in zip(todo, fits):
+                c.fitness = float(f)
+        else:
+            for c in todo:
+                self._eval(c)
+
     def run(self):
         pop = [self._random_chromosome()
                for _ in range(self.population_size)]
         for gen in range(self.generations):
-            for c in pop:
-                if c.fitness is None:
-                    self._eval(c)
+            self._eval_generation(pop)
             pop.sort(key=lambda c: c.fitness, reverse=True)
             if self.best is None or pop[0].fitness > self.best.fitness:
                 self.best = Chromosome(pop[0].genes)
