@@ -61,6 +61,58 @@ def test_inference_engine_matches_workflow(tmp_path):
     assert labels.shape == (75,) and labels.max() < out.shape[-1]
 
 
+def test_export_flat_manifest(tmp_path):
+    # flat export for the standalone C++ engine: manifest + raw blob
+    import os
+    path, _ = _train_and_snapshot(tmp_path)
+    from veles.serving import InferenceEngine, export_flat
+    eng = InferenceEngine(path, device=NumpyDevice())
+    outdir = str(tmp_path / "flat")
+    export_flat(eng, outdir)
+    with open(os.path.join(outdir, "manifest.txt")) as f:
+        lines = f.read().splitlines()
+    assert lines[0] == "mzserve 1" and lines[-1] == "end"
+    ops = [l.split() for l in lines if l.startswith("op ")]
+    # wine MLP: fc(tanh) + fc(f32) + softmax
+    assert [o[1] for o in ops] == ["fc", "fc", "softmax"]
+    tanh_fc = ops[0]
+    assert tanh_fc[4] == "1" and tanh_fc[-1] == "bf16"  # ACT_TANH
+    soft_fc = ops[1]
+    assert soft_fc[4] == "0" and soft_fc[-1] == "f32"
+    # tensor table: offsets 16-B aligned, blob covers every tensor
+    tens = [l.split() for l in lines if l.startswith("tensor ")]
+    assert len(tens) == 4          # 2 weights + 2 biases
+    blob = open(os.path.join(outdir, "weights.bin"), "rb").read()
+    for _, name, dtype, numel, off in tens:
+        assert int(off) % 16 == 0
+        size = int(numel) * (2 if dtype == "bf16" else 4)
+        assert int(off) + size <= len(blob)
+    # fc weight [N,K] bf16 must round-trip the fp32 master (RNE)
+    eng.forwards[0].weights.map_read()
+    w = eng.forwards[0].weights.mem
+    _, name, dtype, numel, off = tens[0]
+    assert int(numel) == w.size
+    got = numpy.frombuffer(
+        blob[int(off):int(off) + 2 * w.size], numpy.uint16)
+    import torch
+    want = torch.from_numpy(numpy.ascontiguousarray(w)).to(
+        torch.bfloat16).view(torch.int16).numpy().astype(
+        numpy.uint16).ravel()
+    numpy.testing.assert_array_equal(got, want)
+    # the standalone binary (if built) parses the manifest
+    binpath = os.path.join(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))), "veles", "znicz_amd", "serve",
+        "mzserve")
+    if os.path.exists(binpath):
+        import json as js
+        import subprocess
+        out = subprocess.run([binpath, outdir, "--parse-only"],
+                             capture_output=True, text=True, timeout=60)
+        assert out.returncode == 0, out.stderr
+        info = js.loads(out.stdout)
+        assert info["ops"] == 3 and info["tensors"] == 4
+
+
 def test_serving_http(tmp_path):
     path, _ = _train_and_snapshot(tmp_path)
     from veles.serving import InferenceEngine, serve

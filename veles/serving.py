@@ -129,6 +129,121 @@ class InferenceEngine(object):
         return self.predict(x).reshape(x.shape[0], -1).argmax(axis=1)
 
 
+# -- flat export for the standalone C++ engine (mzserve) --------------------
+# The reference ecosystem's libVeles/libZnicz loaded snapshots in C++;
+# pickles are a Python format, so the native engine consumes a flat
+# manifest instead: a line-oriented text file + one raw weights blob,
+# with every tensor already in the exact device layout the kernels
+# take (FC forward image [N,K] bf16, conv image [K, KY*KW*C] bf16,
+# fp32 biases).  veles/znicz_amd/serve/mzserve.cpp links the same
+# kernel translation units the _mzk extension uses.
+
+def export_flat(engine, outdir):
+    """Export an InferenceEngine's forward chain for mzserve.
+
+    Writes ``manifest.txt`` + ``weights.bin`` into *outdir*.  Works on
+    any device (host copies are the source of truth).  Raises
+    NotImplementedError for units mzserve does not implement
+    (stochastic pooling, RBM/Kohonen forwards, cutter...).
+    """
+    import os
+    import torch
+    from veles.znicz_amd import all2all, conv as conv_mod, pooling, \
+        normalization, dropout as dropout_mod, activation as act_mod
+
+    os.makedirs(outdir, exist_ok=True)
+    blob = bytearray()
+    lines = ["mzserve 1"]
+
+    def tensor(name, mem, dtype):
+        a = numpy.ascontiguousarray(mem, numpy.float32)
+        if dtype == "bf16":
+            raw = torch.from_numpy(a).to(torch.bfloat16) \
+                .view(torch.int16).numpy().tobytes()
+        else:
+            raw = a.tobytes()
+        while len(blob) % 16:       # device pointers = base + offset
+            blob.append(0)
+        off = len(blob)
+        blob.extend(raw)
+        lines.append("tensor %s %s %d %d" % (name, dtype, a.size, off))
+        return name
+
+    inp = engine._in
+    lines.append("batch %d" % engine.batch_size)
+    lines.append("in_shape %s" % " ".join(str(s) for s in inp.shape))
+
+    def host(arr):
+        arr.map_read()
+        return arr.mem
+
+    for i, f in enumerate(engine.forwards):
+        out_numel = int(numpy.prod(f.output.shape)) \
+            if getattr(f, "output", None) else 0
+        if isinstance(f, dropout_mod.DropoutForward):
+            continue                     # identity at inference
+        elif isinstance(f, all2all.All2AllSoftmax):
+            w = host(f.weights)
+            if f.weights_transposed:     # host [K,N] -> forward [N,K]
+                w = w.T
+            wn = tensor("w%d" % i, w, "bf16")
+            bn = tensor("b%d" % i, host(f.bias), "f32") \
+                if f.include_bias else "-"
+            M, N = f.output.shape[0], f.output.shape[1]
+            K = w.shape[1]
+            lines.append("op fc %s %s 0 %d %d %d f32" % (wn, bn, M, N, K))
+            lines.append("op softmax %d %d" % (M, N))
+        elif isinstance(f, all2all.All2All):
+            w = host(f.weights)
+            if f.weights_transposed:
+                w = w.T
+            wn = tensor("w%d" % i, w, "bf16")
+            bn = tensor("b%d" % i, host(f.bias), "f32") \
+                if f.include_bias else "-"
+            M = f.output.shape[0]
+            N = int(numpy.prod(f.output.shape[1:]))
+            K = w.shape[1]
+            lines.append("op fc %s %s %d %d %d %d bf16" %
+                         (wn, bn, f.ACTIVATION, M, N, K))
+        elif isinstance(f, conv_mod.Conv):
+            d = f.conv_desc()
+            if d[3] % 4 != 0:
+                raise NotImplementedError(
+                    "mzserve conv needs C%%4==0 (got C=%d); train with "
+                    "padded channels as the GPU path does" % d[3])
+            w = host(f.weights).reshape(f.weights.shape[0], -1)
+            wn = tensor("w%d" % i, w, "bf16")
+            bn = tensor("b%d" % i, host(f.bias), "f32") \
+                if f.include_bias else "-"
+            lines.append("op conv %s %s %d %s %d" %
+                         (wn, bn, f.ACTIVATION,
+                          " ".join(str(x) for x in d), out_numel))
+        elif isinstance(f, pooling.Pooling):
+            from veles.znicz_amd.pooling import POOL_STOCHASTIC
+            if f.MODE == POOL_STOCHASTIC:
+                raise NotImplementedError("mzserve: stochastic pooling")
+            lines.append("op pool %d %s %d" %
+                         (f.MODE,
+                          " ".join(str(x) for x in f.pool_desc()),
+                          out_numel))
+        elif isinstance(f, normalization.LRNormalizerForward):
+            C = f.input.shape[-1]
+            rows = int(numpy.prod(f.input.shape)) // C
+            lines.append("op lrn %d %d %r %r %r %d %d" %
+                         (C, rows, f.alpha, f.beta, f.k, f.n, out_numel))
+        elif isinstance(f, act_mod.ActivationForward):
+            lines.append("op act %d" % f.ACTIVATION)
+        else:
+            raise NotImplementedError(
+                "mzserve: no export for unit %s" % type(f).__name__)
+    lines.append("end")
+    with open(os.path.join(outdir, "weights.bin"), "wb") as fo:
+        fo.write(bytes(blob))
+    with open(os.path.join(outdir, "manifest.txt"), "w") as fo:
+        fo.write("\n".join(lines) + "\n")
+    return outdir
+
+
 # -- minimal HTTP JSON front end (same dependency-light style as
 # web_status.py; fastapi/uvicorn exist in the image but add nothing
 # here) ----------------------------------------------------------------------
