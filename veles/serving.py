@@ -288,8 +288,16 @@ def main(argv=None):
     p.add_argument("snapshot")
     p.add_argument("--port", type=int, default=8080)
     p.add_argument("--batch", type=int, default=None)
+    p.add_argument("--export", default=None, metavar="DIR",
+                   help="write a flat manifest+weights export for the "
+                        "standalone C++ engine (mzserve) and exit")
     args = p.parse_args(argv)
     eng = InferenceEngine(args.snapshot, batch_size=args.batch)
+    if args.export:
+        export_flat(eng, args.export)
+        print("exported %s -> %s (run: veles/znicz_amd/serve/mzserve %s)"
+              % (args.snapshot, args.export, args.export))
+        return
     print("serving %s on :%d (batch %d)" % (args.snapshot, args.port,
                                             eng.batch_size))
     serve(eng, args.port)
