@@ -44,6 +44,7 @@ def test_mzserve_matches_python_engine(tmp_path):
                                     "padding": (1, 1, 1, 1)},
          "<-": dict(gd)},
         {"type": "avg_pooling", "->": {"kx": 2, "ky": 2}},
+        {"type": "activation_tanh"},   # standalone act unit → "op act"
         {"type": "all2all_tanh", "->": {"output_sample_shape": 48},
          "<-": dict(gd)},
         {"type": "softmax", "->": {"output_sample_shape": 10},
@@ -94,7 +95,7 @@ def test_mzserve_matches_python_engine(tmp_path):
     assert r.returncode == 0, (r.stdout, r.stderr)
     import json
     stats = json.loads(r.stdout.strip().splitlines()[-1])
-    assert stats["batch"] == 64 and stats["ops"] == 8
+    assert stats["batch"] == 64 and stats["ops"] == 9
     got = numpy.fromfile(of, "<f4").reshape(64, -1)
     numpy.testing.assert_allclose(got, want.reshape(64, -1),
                                   rtol=1e-3, atol=1e-4)
