@@ -113,6 +113,69 @@ def test_export_flat_manifest(tmp_path):
         assert info["ops"] == 3 and info["tensors"] == 4
 
 
+def test_export_flat_edge_cases(tmp_path):
+    import os
+    import types
+    from veles.dummy import DummyWorkflow
+    from veles.memory import Array
+    from veles.serving import export_flat
+    from veles.znicz_amd.all2all import All2AllTanh
+
+    # weights_transposed FC: export must emit the [N,K] forward image
+    wf = DummyWorkflow()
+    u = All2AllTanh(wf, output_sample_shape=(5,),
+                    weights_transposed=True)
+    u.input = Array(numpy.zeros((4, 7), numpy.float32))
+    u.initialize(device=NumpyDevice())
+    eng = types.SimpleNamespace(_in=u.input, batch_size=4,
+                                forwards=[u])
+    d1 = str(tmp_path / "t")
+    export_flat(eng, d1)
+    blob = open(os.path.join(d1, "weights.bin"), "rb").read()
+    lines = open(os.path.join(d1, "manifest.txt")).read().splitlines()
+    trow = next(l.split() for l in lines if l.startswith("tensor w"))
+    import torch
+    got = numpy.frombuffer(
+        blob[int(trow[4]):int(trow[4]) + 2 * int(trow[3])],
+        numpy.uint16)
+    u.weights.map_read()
+    want = torch.from_numpy(numpy.ascontiguousarray(
+        u.weights.mem.T)).to(torch.bfloat16).view(
+        torch.int16).numpy().astype(numpy.uint16).ravel()
+    numpy.testing.assert_array_equal(got, want)
+    oprow = next(l.split() for l in lines if l.startswith("op fc"))
+    assert oprow[5:8] == ["4", "5", "7"]     # M N K
+
+    # unsupported unit type -> clear NotImplementedError
+    class Alien(object):
+        output = None
+    eng2 = types.SimpleNamespace(_in=u.input, batch_size=4,
+                                 forwards=[Alien()])
+    with pytest.raises(NotImplementedError, match="Alien"):
+        export_flat(eng2, str(tmp_path / "x"))
+
+    # stochastic pooling -> refused
+    from veles.znicz_amd.pooling import StochasticPooling
+    sp = StochasticPooling(wf, kx=2, ky=2)
+    sp.input = Array(numpy.zeros((2, 4, 4, 4), numpy.float32))
+    sp.initialize(device=NumpyDevice())
+    eng3 = types.SimpleNamespace(_in=sp.input, batch_size=2,
+                                 forwards=[sp])
+    with pytest.raises(NotImplementedError, match="stochastic"):
+        export_flat(eng3, str(tmp_path / "y"))
+
+    # conv with C % 4 != 0 -> refused with guidance
+    from veles.znicz_amd.conv import ConvStrictRELU
+    cv = ConvStrictRELU(wf, n_kernels=8, kx=3, ky=3,
+                        padding=(1, 1, 1, 1))
+    cv.input = Array(numpy.zeros((2, 6, 6, 3), numpy.float32))
+    cv.initialize(device=NumpyDevice())
+    eng4 = types.SimpleNamespace(_in=cv.input, batch_size=2,
+                                 forwards=[cv])
+    with pytest.raises(NotImplementedError, match="C%4"):
+        export_flat(eng4, str(tmp_path / "z"))
+
+
 def test_serving_http(tmp_path):
     path, _ = _train_and_snapshot(tmp_path)
     from veles.serving import InferenceEngine, serve
