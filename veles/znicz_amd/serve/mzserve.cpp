@@ -22,6 +22,7 @@
 // Run:
 //   ./mzserve <export_dir> [--input in.f32.bin] [--out out.f32.bin]
 //             [--batches K] [--warmup W] [--no-graph] [--parse-only]
+//             [--h2d]   (end-to-end: per-batch convert+upload timed)
 // prints one JSON line: {"img_s":..., "ms_per_batch":..., ...}
 
 #include <hip/hip_runtime.h>
@@ -309,7 +310,7 @@ int main(int argc, char** argv) {
   }
   std::string dir = argv[1], in_file, out_file;
   int batches = 20, warmup = 3;
-  bool use_graph = true, parse_only = false;
+  bool use_graph = true, parse_only = false, time_h2d = false;
   for (int i = 2; i < argc; ++i) {
     std::string a = argv[i];
     if (a == "--input" && i + 1 < argc) in_file = argv[++i];
@@ -318,6 +319,7 @@ int main(int argc, char** argv) {
     else if (a == "--warmup" && i + 1 < argc) warmup = atoi(argv[++i]);
     else if (a == "--no-graph") use_graph = false;
     else if (a == "--parse-only") parse_only = true;
+    else if (a == "--h2d") time_h2d = true;  // re-upload per batch
   }
 
   Engine eng;
@@ -344,9 +346,12 @@ int main(int argc, char** argv) {
     std::uniform_real_distribution<float> u(-1.f, 1.f);
     for (float& v : host_in) v = u(rng);
   }
-  std::vector<unsigned short> host_bf(eng.in_numel);
+  // pinned staging in the device dtype (mirrors the Python engine's
+  // convert-then-upload host path)
+  unsigned short* host_bf = nullptr;
+  HIP_OK(hipHostMalloc(&host_bf, eng.in_numel * 2));
   for (long i = 0; i < eng.in_numel; ++i) host_bf[i] = f2bf(host_in[i]);
-  HIP_OK(hipMemcpyAsync(eng.input, host_bf.data(), eng.in_numel * 2,
+  HIP_OK(hipMemcpyAsync(eng.input, host_bf, eng.in_numel * 2,
                         hipMemcpyHostToDevice, eng.stream));
 
   // warmup (also allocates the FC split-K workspace pre-capture)
@@ -373,8 +378,17 @@ int main(int argc, char** argv) {
 
   auto t0 = std::chrono::steady_clock::now();
   for (int i = 0; i < batches; ++i) {
+    if (time_h2d) {
+      // end-to-end mode: fresh fp32→bf16 convert + pinned upload per
+      // batch, like a real request path
+      for (long j = 0; j < eng.in_numel; ++j)
+        host_bf[j] = f2bf(host_in[j]);
+      HIP_OK(hipMemcpyAsync(eng.input, host_bf, eng.in_numel * 2,
+                            hipMemcpyHostToDevice, eng.stream));
+    }
     if (gexec) HIP_OK(hipGraphLaunch(gexec, eng.stream));
     else eng.forward();
+    if (time_h2d) HIP_OK(hipStreamSynchronize(eng.stream));
   }
   HIP_OK(hipStreamSynchronize(eng.stream));
   auto t1 = std::chrono::steady_clock::now();
@@ -399,8 +413,9 @@ int main(int argc, char** argv) {
   std::printf(
       "{\"engine\": \"mzserve\", \"batch\": %d, \"batches\": %d, "
       "\"warmup\": %d, \"ms_per_batch\": %.4f, \"img_s\": %.1f, "
-      "\"graph\": %s, \"ops\": %zu}\n",
+      "\"graph\": %s, \"ops\": %zu, \"h2d\": %s}\n",
       eng.m.batch, batches, warmup, ms, eng.m.batch * 1000.0 / ms,
-      gexec ? "true" : "false", eng.m.ops.size());
+      gexec ? "true" : "false", eng.m.ops.size(),
+      time_h2d ? "true" : "false");
   return 0;
 }
