@@ -85,12 +85,14 @@ def main():
 
     binpath = os.path.join(REPO, "veles", "znicz_amd", "serve",
                            "mzserve")
-    r = subprocess.run(
-        [binpath, flat, "--batches", str(args.batches), "--warmup", "5"],
-        capture_output=True, text=True, timeout=600)
-    sys.stderr.write(r.stderr)
-    print(r.stdout.strip())
-    assert r.returncode == 0
+    for extra in ([], ["--h2d"]):
+        r = subprocess.run(
+            [binpath, flat, "--batches", str(args.batches),
+             "--warmup", "5"] + extra,
+            capture_output=True, text=True, timeout=600)
+        sys.stderr.write(r.stderr)
+        print(r.stdout.strip())
+        assert r.returncode == 0
 
 
 if __name__ == "__main__":
