@@ -58,6 +58,7 @@ void dropout_bwd_launch(unsigned short*, const unsigned char*, long, float,
 void histogram_launch(const void*, bool, long, float, float, int,
                       unsigned int*, hipStream_t);
 void set_routing(int, int, int, int);
+void set_fc_split(int, int, int, int, int);
 // from softmax.hip
 void softmax_launch(float*, int*, int, int, hipStream_t);
 void eval_softmax_launch(const float*, const int*, unsigned short*, int*,
@@ -436,6 +437,11 @@ void set_routing_py(int64_t nt_split, int64_t tn_split,
               (int)ntn_max_n);
 }
 
+void set_fc_split_py(int64_t is_tn, int64_t M, int64_t N, int64_t K,
+                     int64_t splits) {
+  set_fc_split((int)is_tn, (int)M, (int)N, (int)K, (int)splits);
+}
+
 void histogram(torch::Tensor x, torch::Tensor counts, double lo,
                double hi) {
   TORCH_CHECK(x.is_cuda() && counts.is_cuda(), "histogram: cuda only");
@@ -499,4 +505,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("histogram", &histogram, "LDS histogram (counts int32)");
   m.def("set_routing", &set_routing_py,
         "override GEMM routing constants (gfx950_routing.json)");
+  m.def("set_fc_split", &set_fc_split_py,
+        "per-shape dense-GEMM split override: (is_tn, M, N, K, splits);"
+        " splits<=0 removes the row");
 }

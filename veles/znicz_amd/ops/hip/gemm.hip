@@ -1497,6 +1497,35 @@ void set_routing(int nt_split, int tn_split, int tn2_split,
   if (ntn_max_n >= 0) g_ntn_max_n = ntn_max_n;
 }
 
+// Per-shape split table for the DENSE (FC) entry points only — the
+// device_infos.json idea one level deeper: exact (M,N,K) rows swept on
+// hardware (tools/gen_fc_splits.py) override the budget formula, which
+// over-splits small-tile FC shapes (measured: fc8 wgrad 33.1→12.9 µs,
+// fc8 fwd 26.0→19.0 µs at batch 256).  Conv GEMMs never consult this.
+// Populated at import / tool time, read-only on the hot path.
+static std::unordered_map<unsigned long long, int> g_fc_nt_splits;
+static std::unordered_map<unsigned long long, int> g_fc_tn_splits;
+
+static inline unsigned long long fc_shape_key(int M, int N, int K) {
+  return ((unsigned long long)(unsigned)M << 42) |
+         ((unsigned long long)(unsigned)N << 21) |
+         (unsigned long long)(unsigned)K;
+}
+
+void set_fc_split(int is_tn, int M, int N, int K, int splits) {
+  auto& t = is_tn ? g_fc_tn_splits : g_fc_nt_splits;
+  if (splits <= 0)
+    t.erase(fc_shape_key(M, N, K));
+  else
+    t[fc_shape_key(M, N, K)] = splits;
+}
+
+static inline int fc_split_lookup(bool is_tn, int M, int N, int K) {
+  const auto& t = is_tn ? g_fc_tn_splits : g_fc_nt_splits;
+  auto it = t.find(fc_shape_key(M, N, K));
+  return it == t.end() ? 0 : it->second;   // 0 = use the budget formula
+}
+
 
 template <typename LA, typename LB, typename OUT_T>
 void launch_gemm_nt(const LA& la, const LB& lb, OUT_T* C, const float* bias,
@@ -1578,7 +1607,7 @@ void launch_gemm_tn(const LA& la, const LB& lb, float* C, int M, int N,
 template <typename LA, typename LB>
 bool launch_gemm_tn2(const LA& la, const LB& lb, float* C, int M, int N,
                      int K, float beta, hipStream_t stream,
-                     float* bias_out = nullptr) {
+                     float* bias_out = nullptr, int split_override = 0) {
   const int tn = cdiv(N, BN);
   const int bmt = M <= 32 ? 32 : (M <= 64 ? 64 : BM);
   // bias fusion is narrow-tile only: in the 128-row kernel the extra
@@ -1590,8 +1619,9 @@ bool launch_gemm_tn2(const LA& la, const LB& lb, float* C, int M, int N,
   // each split must carry >= 4 K-tiles: tiny-K wgrads (K = batch,
   // nk ~ 8) otherwise split into atomic-heavy slivers (fc8 wgrad
   // measured 23 TF/s with 2-tile splits)
-  const int splits =
+  int splits =
       max(1, min(min(nk, g_tn2_split_budget / tiles), max(1, nk / 4)));
+  if (split_override > 0) splits = min(nk, split_override);
   if (splits > 1) {
     if (beta == 0.f) {
       void zero_f32_launch(float*, long, hipStream_t);
@@ -1636,7 +1666,9 @@ void gemm_nt_dense(const unsigned short* A, const unsigned short* B,
   DenseNT la{A, zero, K, K}, lb{B, zero, K, K};
   const int tiles = cdiv(M, BM) * cdiv(N, BN);
   const int nk = cdiv(K, BK);
-  const int splits = max(1, min(nk, g_nt_split_budget / tiles));
+  int splits = max(1, min(nk, g_nt_split_budget / tiles));
+  if (int ov = fc_split_lookup(false, M, N, K))
+    splits = max(1, min(nk, ov));
   if (splits > 1) {
     // fp32 workspace + atomic partials, then one bias/act/cast pass.
     // ONE process-global workspace: every FC split-K launch in this
@@ -1699,7 +1731,8 @@ bool gemm_tn_dense(const unsigned short* A, const unsigned short* B,
                    hipStream_t s, float* bias_out) {
   if ((M & 7) == 0 && (N & 7) == 0) {
     DenseTNv la{A, M, K, M}, lb{B, N, K, N};
-    return launch_gemm_tn2(la, lb, C, M, N, K, beta, s, bias_out);
+    return launch_gemm_tn2(la, lb, C, M, N, K, beta, s, bias_out,
+                           fc_split_lookup(true, M, N, K));
   }
   DenseTN la{A, M, K, M}, lb{B, N, K, N};
   launch_gemm_tn(la, lb, C, M, N, K, beta, s);
