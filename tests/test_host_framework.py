@@ -359,6 +359,38 @@ def test_routing_table_loader_robust(tmp_path, monkeypatch):
     ops_mod._apply_routing(FakeMod)
     assert FakeMod.calls == [(640, -1, -1, -1)]
 
+    # per-shape FC split rows reach set_fc_split; malformed rows only
+    # drop the table, never the import
+    class FakeMod2:
+        calls = []
+        fc = []
+
+        @staticmethod
+        def set_routing(a, b, c, d):
+            FakeMod2.calls.append((a, b, c, d))
+
+        @staticmethod
+        def set_fc_split(is_tn, m, n, k, s):
+            FakeMod2.fc.append((is_tn, m, n, k, s))
+
+    bad.write_text('{"fc_nt_splits": [[256, 1000, 4096, 16]], '
+                   '"fc_tn_splits": [[1000, 4096, 256, 1]]}')
+    ops_mod._apply_routing(FakeMod2)
+    assert FakeMod2.fc == [(0, 256, 1000, 4096, 16),
+                           (1, 1000, 4096, 256, 1)]
+    bad.write_text('{"fc_nt_splits": [[256, "x"]]}')
+    ops_mod._apply_routing(FakeMod2)         # must not raise
+
+    # the COMMITTED table must parse and contain the swept row kinds
+    import json as js
+    real = js.load(open(real_join(
+        os.path.dirname(os.path.abspath(ops_mod.__file__)),
+        "gfx950_routing.json")))
+    for field in ("fc_nt_splits", "fc_tn_splits"):
+        for row in real.get(field, []):
+            assert len(row) == 4 and all(
+                isinstance(v, int) and v > 0 for v in row), row
+
 
 def test_golden_snapshot_format_stable():
     """Format-stability guard: the committed round-2 wine snapshot

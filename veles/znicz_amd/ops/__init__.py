@@ -47,11 +47,14 @@ def _apply_routing(mod):
                         int(r.get("tn2_split_budget", -1)),
                         int(r.get("ntn_max_n", -1)))
         # per-(M,N,K) dense-GEMM split rows (tools/gen_fc_splits.py):
-        # the budget formula over-splits small-tile FC shapes
-        for is_tn, field in ((0, "fc_nt_splits"), (1, "fc_tn_splits")):
-            for row in r.get(field, []):
-                m_, n_, k_, s_ = (int(v) for v in row)
-                mod.set_fc_split(is_tn, m_, n_, k_, s_)
+        # the budget formula over-splits small-tile FC shapes.
+        # VELES_NO_FC_TABLE=1 skips them (A/B and debugging)
+        if os.environ.get("VELES_NO_FC_TABLE", "0") != "1":
+            for is_tn, field in ((0, "fc_nt_splits"),
+                                 (1, "fc_tn_splits")):
+                for row in r.get(field, []):
+                    m_, n_, k_, s_ = (int(v) for v in row)
+                    mod.set_fc_split(is_tn, m_, n_, k_, s_)
     except Exception as e:                       # never block import
         import logging
         logging.getLogger("ops").warning("routing table ignored: %r", e)
