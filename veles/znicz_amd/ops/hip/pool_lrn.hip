@@ -319,6 +319,93 @@ __global__ void pool_bwd_k(const unsigned short* __restrict__ ey,
   }
 }
 
+// -- backward, specialized for <= WW covering windows per axis (WW = 1
+// covers the k2s2 CIFAR pools, WW = 2 the k3s2 AlexNet pools): the
+// window set is compile-time unrolled with masked contributions and
+// ALL (ey, offs) loads issued together, so their latency overlaps
+// instead of serializing per window — the generic kernel measured
+// 43.7% SQ_WAIT_INST_ANY (issue-bound) on pool1 (profiles/perf_r2.md).
+// vec8 (C % 8 == 0) only; the launcher falls back otherwise.
+template <int MODE, int WW>
+__global__ void pool_bwd_w_k(const unsigned short* __restrict__ ey,
+                             unsigned short* __restrict__ ex,
+                             const unsigned char* __restrict__ offs,
+                             PoolDesc d, int lg) {
+  const int C2 = 1 << lg;
+  const int c0 = threadIdx.x & (C2 - 1);
+  const int rin = threadIdx.x >> lg;
+  const int rpb = blockDim.x >> lg;
+  const unsigned rows = (unsigned)d.N * d.H * d.W;
+  const int noct = d.C >> 3;
+  constexpr int NQ = WW * WW;
+  for (unsigned row = blockIdx.x * rpb + rin; row < rows;
+       row += gridDim.x * rpb) {
+    unsigned t = row / d.W;
+    const int iw = row - t * d.W;
+    const int n = t / d.H;
+    const int ih = t - n * d.H;
+    const int oy0 = (ih - d.KY + 1 <= 0) ? 0 : (ih - d.KY) / d.SY + 1;
+    const int oy1 = min(d.OH - 1, ih / d.SY);
+    const int ox0 = (iw - d.KX + 1 <= 0) ? 0 : (iw - d.KX) / d.SX + 1;
+    const int ox1 = min(d.OW - 1, iw / d.SX);
+    // candidate windows; out-of-range slots clamp to a valid address
+    // (duplicate, L1-hot) and mask their contribution to zero
+    int oys[WW], oxs[WW];
+    bool vy[WW], vx[WW];
+    oys[0] = min(oy0, oy1);
+    vy[0] = oy0 <= oy1;
+    oxs[0] = min(ox0, ox1);
+    vx[0] = ox0 <= ox1;
+    if (WW == 2) {
+      oys[WW - 1] = min(oys[0] + 1, oy1);
+      vy[WW - 1] = vy[0] && oys[0] + 1 <= oy1;
+      oxs[WW - 1] = min(oxs[0] + 1, ox1);
+      vx[WW - 1] = vx[0] && oxs[0] + 1 <= ox1;
+    }
+    long ob[NQ];
+    int rc[NQ];
+    bool vv[NQ];
+#pragma unroll
+    for (int q = 0; q < NQ; ++q) {
+      const int oy = oys[q / WW], ox = oxs[q % WW];
+      ob[q] = (((long)n * d.OH + oy) * d.OW + ox) * d.C;
+      rc[q] = (ih - oy * d.SY) * d.KX + (iw - ox * d.SX);
+      vv[q] = vy[q / WW] && vx[q % WW];
+    }
+    for (int oc = c0; oc < noct; oc += C2) {
+      const int cb = oc * 8;
+      unsigned short e8[NQ][8];
+      unsigned char w8[NQ][8];
+#pragma unroll
+      for (int q = 0; q < NQ; ++q) {
+        *(short8*)e8[q] = *(const short8*)(ey + ob[q] + cb);
+        *(unsigned long long*)w8[q] =
+            *(const unsigned long long*)(offs + ob[q] + cb);
+      }
+      float g[8] = {};
+#pragma unroll
+      for (int q = 0; q < NQ; ++q) {
+        if (MODE == POOL_AVG) {
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            g[j] += vv[q]
+                        ? bf2f(e8[q][j]) / (float)max((int)w8[q][j], 1)
+                        : 0.f;
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            g[j] += (vv[q] && (int)w8[q][j] == rc[q]) ? bf2f(e8[q][j])
+                                                      : 0.f;
+        }
+      }
+      unsigned short o8[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) o8[j] = f2bf(g[j]);
+      *(short8*)(ex + (long)row * d.C + cb) = *(short8*)o8;
+    }
+  }
+}
+
 void pool_fwd_launch(const unsigned short* x, unsigned short* y,
                      unsigned char* offs,
                      const PoolDesc& d, int mode, hipStream_t s) {
@@ -341,9 +428,33 @@ void pool_fwd_launch(const unsigned short* x, unsigned short* y,
 void pool_bwd_launch(const unsigned short* ey, unsigned short* ex,
                      const unsigned char* offs, const PoolDesc& d,
                      int mode, hipStream_t s) {
+  long rows = (long)d.N * d.H * d.W;
+  const int wy = (d.KY - 1) / d.SY + 1;   // covering windows per axis
+  const int wx = (d.KX - 1) / d.SX + 1;
+  if ((d.C & 7) == 0 && wy <= 2 && wx <= 2) {
+    // full-lane mapping: largest pow2 DIVIDING the octet count, so no
+    // lane idles on C = 96-style shapes (12 octets over 16 lanes was
+    // 25% idle); the octet loop runs noct/C2 full iterations instead
+    int lg = chan_lg(d.C >> 3);
+    while (lg > 0 && ((d.C >> 3) & ((1 << lg) - 1)) != 0) --lg;
+    const int rpb = 256 >> lg;
+    dim3 grid((unsigned)min((rows + rpb - 1) / rpb, (long)32768)),
+        block(256);
+    const int ww = max(wy, wx);
+#define GOW(M_, W_)                                                         \
+    hipLaunchKernelGGL((pool_bwd_w_k<M_, W_>), grid, block, 0, s, ey, ex,   \
+                       offs, d, lg)
+    if (mode == POOL_AVG) {
+      if (ww == 1) GOW(POOL_AVG, 1); else GOW(POOL_AVG, 2);
+    } else {
+      if (ww == 1) GOW(POOL_MAX, 1); else GOW(POOL_MAX, 2);
+    }
+#undef GOW
+    HIP_CHECK_LAST();
+    return;
+  }
   const int lg = (d.C & 7) == 0 ? chan_lg(d.C >> 3) : chan_lg(d.C);
   const int rpb = 256 >> lg;
-  long rows = (long)d.N * d.H * d.W;
   dim3 grid((unsigned)min((rows + rpb - 1) / rpb, (long)32768)), block(256);
 #define GO(M_)                                                              \
   hipLaunchKernelGGL(pool_bwd_k<M_>, grid, block, 0, s, ey, ex, offs, d, lg)
