@@ -431,7 +431,13 @@ void pool_bwd_launch(const unsigned short* ey, unsigned short* ex,
   long rows = (long)d.N * d.H * d.W;
   const int wy = (d.KY - 1) / d.SY + 1;   // covering windows per axis
   const int wx = (d.KX - 1) / d.SX + 1;
-  if ((d.C & 7) == 0 && wy <= 2 && wx <= 2) {
+  const int noct = d.C >> 3;
+  // specialized kernel wins where the round-up lane mapping idles
+  // lanes (C = 96: 12 octets over 16 lanes = 25% idle; measured
+  // pool1 bwd 113.5 -> 67.1 us); on pow2 octet counts the generic
+  // kernel's serial window loop is already ~even (pool2 54.2 vs 56.4)
+  if ((d.C & 7) == 0 && wy <= 2 && wx <= 2 &&
+      (noct & (noct - 1)) != 0) {
     // full-lane mapping: largest pow2 DIVIDING the octet count, so no
     // lane idles on C = 96-style shapes (12 octets over 16 lanes was
     // 25% idle); the octet loop runs noct/C2 full iterations instead
