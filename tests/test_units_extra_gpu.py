@@ -537,3 +537,50 @@ def test_depooling_gpu_matches_numpy():
         outs[kind] = (dp.output.mem.copy(), gd.err_input.mem.copy())
     assert_close(outs["rocm"][0], outs["numpy"][0])
     assert_close(outs["rocm"][1], outs["numpy"][1])
+
+
+def test_pool_bwd_specialized_gpu():
+    """The unrolled window-quad pool backward (non-pow2 octet counts,
+    <=2 covering windows per axis) must match numpy: C=24 (3 octets)
+    exercises the specialized lane mapping, k3s2 the WW=2 quad with
+    border tails, k2s2 the WW=1 single-window variant; MAX and AVG."""
+    import torch
+    from veles.znicz_amd.pooling import MaxPooling, AvgPooling
+    from veles.znicz_amd.gd_pooling import GDMaxPooling, GDAvgPooling
+    cases = [
+        (MaxPooling, GDMaxPooling, dict(kx=3, ky=3, sliding=(2, 2))),
+        (AvgPooling, GDAvgPooling, dict(kx=3, ky=3, sliding=(2, 2))),
+        (MaxPooling, GDMaxPooling, dict(kx=2, ky=2, sliding=(2, 2))),
+        (AvgPooling, GDAvgPooling, dict(kx=2, ky=2, sliding=(2, 2))),
+    ]
+    data = RNG.uniform(-1, 1, (2, 9, 9, 24)).astype(numpy.float32)
+    for fwd_cls, gd_cls, kw in cases:
+        outs = {}
+        err = None
+        for kind in ("numpy", "rocm"):
+            dev = NumpyDevice() if kind == "numpy" else ROCmDevice(0)
+            wf = DummyWorkflow()
+            pool = fwd_cls(wf, **kw)
+            pool.input = Array(data.copy())
+            pool.input.device_dtype = "bf16"
+            pool.input.initialize(dev)
+            assert pool.initialize(device=dev) is not False
+            pool.run()
+            oh, ow = pool.out_spatial()
+            if err is None:
+                err = RNG.uniform(-1, 1, (2, oh, ow, 24)).astype(
+                    numpy.float32)
+            gd = gd_cls(wf)
+            gd.forward_unit = pool
+            gd.input = pool.input
+            gd.input_offset = pool.input_offset
+            gd.err_output = Array(err.copy())
+            gd.err_output.device_dtype = "bf16"
+            gd.err_output.initialize(dev)
+            assert gd.initialize(device=dev) is not False
+            gd.run()
+            gd.err_input.map_read()
+            if kind == "rocm":
+                torch.cuda.synchronize()
+            outs[kind] = gd.err_input.mem.copy()
+        assert_close(outs["rocm"], outs["numpy"])
