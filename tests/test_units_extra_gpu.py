@@ -541,11 +541,16 @@ def test_depooling_gpu_matches_numpy():
 
 def test_pool_bwd_specialized_gpu():
     """The unrolled window-quad pool backward (non-pow2 octet counts,
-    <=2 covering windows per axis) must match numpy: C=24 (3 octets)
-    exercises the specialized lane mapping, k3s2 the WW=2 quad with
-    border tails, k2s2 the WW=1 single-window variant; MAX and AVG."""
+    <=2 covering windows per axis) must honor the u8 offset-code
+    contract exactly: scatter each err_output to the coded winner
+    (MAX) / spread by the stored count (AVG).  Reference is built on
+    the host FROM THE GPU's own codes, so fp32-vs-bf16 argmax ties
+    cannot produce false failures.  C=24 (3 octets) exercises the
+    specialized lane mapping, k3s2 the WW=2 quad with border tails,
+    k2s2 the WW=1 variant."""
     import torch
-    from veles.znicz_amd.pooling import MaxPooling, AvgPooling
+    from veles.znicz_amd.pooling import (MaxPooling, AvgPooling,
+                                         POOL_AVG)
     from veles.znicz_amd.gd_pooling import GDMaxPooling, GDAvgPooling
     cases = [
         (MaxPooling, GDMaxPooling, dict(kx=3, ky=3, sliding=(2, 2))),
@@ -554,33 +559,50 @@ def test_pool_bwd_specialized_gpu():
         (AvgPooling, GDAvgPooling, dict(kx=2, ky=2, sliding=(2, 2))),
     ]
     data = RNG.uniform(-1, 1, (2, 9, 9, 24)).astype(numpy.float32)
+    dev = ROCmDevice(0)
     for fwd_cls, gd_cls, kw in cases:
-        outs = {}
-        err = None
-        for kind in ("numpy", "rocm"):
-            dev = NumpyDevice() if kind == "numpy" else ROCmDevice(0)
-            wf = DummyWorkflow()
-            pool = fwd_cls(wf, **kw)
-            pool.input = Array(data.copy())
-            pool.input.device_dtype = "bf16"
-            pool.input.initialize(dev)
-            assert pool.initialize(device=dev) is not False
-            pool.run()
-            oh, ow = pool.out_spatial()
-            if err is None:
-                err = RNG.uniform(-1, 1, (2, oh, ow, 24)).astype(
-                    numpy.float32)
-            gd = gd_cls(wf)
-            gd.forward_unit = pool
-            gd.input = pool.input
-            gd.input_offset = pool.input_offset
-            gd.err_output = Array(err.copy())
-            gd.err_output.device_dtype = "bf16"
-            gd.err_output.initialize(dev)
-            assert gd.initialize(device=dev) is not False
-            gd.run()
-            gd.err_input.map_read()
-            if kind == "rocm":
-                torch.cuda.synchronize()
-            outs[kind] = gd.err_input.mem.copy()
-        assert_close(outs["rocm"], outs["numpy"])
+        wf = DummyWorkflow()
+        pool = fwd_cls(wf, **kw)
+        pool.input = Array(data.copy())
+        pool.input.device_dtype = "bf16"
+        pool.input.initialize(dev)
+        assert pool.initialize(device=dev) is not False
+        pool.run()
+        oh, ow = pool.out_spatial()
+        err = RNG.uniform(-1, 1, (2, oh, ow, 24)).astype(numpy.float32)
+        gd = gd_cls(wf)
+        gd.forward_unit = pool
+        gd.input = pool.input
+        gd.input_offset = pool.input_offset
+        gd.err_output = Array(err.copy())
+        gd.err_output.device_dtype = "bf16"
+        gd.err_output.initialize(dev)
+        assert gd.initialize(device=dev) is not False
+        gd.run()
+        torch.cuda.synchronize()
+        gd.err_input.map_read()
+        got = gd.err_input.mem.copy()
+        # host reference from the GPU's own u8 codes + its bf16 eo
+        codes = pool.input_offset.devmem.cpu().numpy().reshape(
+            2, oh, ow, 24)
+        eo = gd.err_output.devmem.float().cpu().numpy()
+        n, h, w, c = data.shape
+        ky, kx = kw["ky"], kw["kx"]
+        sy, sx = kw["sliding"][1], kw["sliding"][0]
+        want = numpy.zeros((n, h, w, c), numpy.float32)
+        avg = pool.MODE == POOL_AVG
+        for ni in range(n):
+            for oy in range(oh):
+                for ox in range(ow):
+                    for ci in range(c):
+                        code = int(codes[ni, oy, ox, ci])
+                        e = eo[ni, oy, ox, ci]
+                        if avg:
+                            iy1 = min(oy * sy + ky, h)
+                            ix1 = min(ox * sx + kx, w)
+                            want[ni, oy * sy:iy1, ox * sx:ix1, ci] += \
+                                e / max(code, 1)
+                        elif code != 255:
+                            want[ni, oy * sy + code // kx,
+                                 ox * sx + code % kx, ci] += e
+        assert_close(got, want)
