@@ -332,6 +332,34 @@ def test_cli_snapshot_resume(tmp_path):
     assert out2.returncode == 0, out2.stderr[-2000:]
 
 
+def test_snapshot_codecs_roundtrip(tmp_path):
+    # all four reference codecs ("", gz, bz2, xz) must export+import
+    from veles import prng
+    prng.seed_all(7)
+    import veles.znicz_amd.models.wine  # noqa: F401
+    from veles.config import root
+    root.wine.decision.max_epochs = 1
+    root.wine.loader.update({"n_train": 30, "n_valid": 0})
+    from veles.dummy import DummyLauncher
+    from veles.backends import NumpyDevice
+    from veles.snapshotter import SnapshotterBase, SnapshotterToFile
+    from veles.znicz_amd.models.wine import WineWorkflow
+    wf = WineWorkflow(DummyLauncher())
+    wf.initialize(device=NumpyDevice())
+    wf.run()
+    for codec in ("", "gz", "bz2", "xz"):
+        snap = SnapshotterToFile(wf, prefix="c_%s" % (codec or "raw"),
+                                 directory=str(tmp_path),
+                                 compression=codec)
+        wf.del_ref(snap)
+        path = snap.export()
+        assert os.path.exists(path), codec
+        wf2 = SnapshotterBase.import_(path)
+        numpy.testing.assert_array_equal(
+            wf2.forwards[0].weights.mem, wf.forwards[0].weights.mem,
+            err_msg=codec)
+
+
 def test_routing_table_loader_robust(tmp_path, monkeypatch):
     """A corrupt gfx950_routing.json must never block the ops import."""
     import veles.znicz_amd.ops as ops_mod
