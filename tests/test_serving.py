@@ -176,6 +176,43 @@ def test_export_flat_edge_cases(tmp_path):
         export_flat(eng4, str(tmp_path / "z"))
 
 
+def test_mzserve_rejects_malformed_manifests(tmp_path):
+    import os
+    import subprocess
+    binpath = os.path.join(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))), "veles", "znicz_amd", "serve",
+        "mzserve")
+    if not os.path.exists(binpath):
+        pytest.skip("mzserve binary not built")
+
+    def run(d):
+        return subprocess.run([binpath, str(d), "--parse-only"],
+                              capture_output=True, text=True,
+                              timeout=60)
+    # missing manifest
+    r = run(tmp_path / "nope")
+    assert r.returncode != 0 and "manifest" in r.stderr
+    # bad header
+    d1 = tmp_path / "bad1"
+    d1.mkdir()
+    (d1 / "manifest.txt").write_text("something else\n")
+    r = run(d1)
+    assert r.returncode != 0 and "header" in r.stderr
+    # unknown op
+    d2 = tmp_path / "bad2"
+    d2.mkdir()
+    (d2 / "manifest.txt").write_text(
+        "mzserve 1\nbatch 4\nin_shape 4 8\nop warp\nend\n")
+    r = run(d2)
+    assert r.returncode != 0 and "unknown op" in r.stderr
+    # structurally empty (no ops)
+    d3 = tmp_path / "bad3"
+    d3.mkdir()
+    (d3 / "manifest.txt").write_text("mzserve 1\nbatch 4\nend\n")
+    r = run(d3)
+    assert r.returncode != 0 and "incomplete" in r.stderr
+
+
 def test_serving_http(tmp_path):
     path, _ = _train_and_snapshot(tmp_path)
     from veles.serving import InferenceEngine, serve
