@@ -211,6 +211,26 @@ def test_mzserve_rejects_malformed_manifests(tmp_path):
     (d3 / "manifest.txt").write_text("mzserve 1\nbatch 4\nend\n")
     r = run(d3)
     assert r.returncode != 0 and "incomplete" in r.stderr
+    # chain starting with an in-place op (would mutate the resident
+    # input across replays)
+    d4 = tmp_path / "bad4"
+    d4.mkdir()
+    (d4 / "manifest.txt").write_text(
+        "mzserve 1\nbatch 4\nin_shape 4 8\nop act 1\nend\n")
+    r = run(d4)
+    assert r.returncode != 0 and "in-place" in r.stderr
+    # tensor extending past weights.bin (caught before any hip call,
+    # so this lane runs on CPU boxes too)
+    d5 = tmp_path / "bad5"
+    d5.mkdir()
+    (d5 / "manifest.txt").write_text(
+        "mzserve 1\nbatch 4\nin_shape 4 8\n"
+        "tensor w0 bf16 100 0\n"
+        "op fc w0 - 0 4 10 8 bf16\nend\n")
+    (d5 / "weights.bin").write_bytes(b"\0" * 64)   # 100 bf16 > 64 B
+    r = subprocess.run([binpath, str(d5)], capture_output=True,
+                       text=True, timeout=60)
+    assert r.returncode != 0 and "exceeds" in r.stderr
 
 
 def test_serving_http(tmp_path):

@@ -182,6 +182,14 @@ Model parse_manifest(const std::string& dir) {
     std::fprintf(stderr, "mzserve: incomplete manifest\n");
     std::exit(1);
   }
+  // the first op must produce into its own buffer: softmax/act run
+  // in place on the PREVIOUS op's output and would otherwise mutate
+  // the resident input across replays
+  if (m.ops[0].kind == "act" || m.ops[0].kind == "softmax") {
+    std::fprintf(stderr, "mzserve: manifest starts with in-place op "
+                         "%s\n", m.ops[0].kind.c_str());
+    std::exit(1);
+  }
   return m;
 }
 
@@ -210,10 +218,23 @@ struct Engine {
   }
 
   void setup(const std::string& dir) {
-    HIP_OK(hipStreamCreateWithFlags(&stream, hipStreamNonBlocking));
+    // weights load + bounds check first: no hip calls yet, so a
+    // malformed export fails cleanly even without a GPU
     std::ifstream wf(dir + "/weights.bin", std::ios::binary);
     std::vector<char> blob((std::istreambuf_iterator<char>(wf)),
                            std::istreambuf_iterator<char>());
+    for (const auto& kv : m.tensors) {
+      const TensorRef& t = kv.second;
+      long bytes = t.numel * (t.dtype == "f32" ? 4 : 2);
+      if (t.offset < 0 || t.offset + bytes > (long)blob.size()) {
+        std::fprintf(stderr,
+                     "mzserve: tensor %s [%ld B at %ld] exceeds "
+                     "weights.bin (%zu B)\n", kv.first.c_str(), bytes,
+                     t.offset, blob.size());
+        std::exit(1);
+      }
+    }
+    HIP_OK(hipStreamCreateWithFlags(&stream, hipStreamNonBlocking));
     HIP_OK(hipMalloc(&weights, blob.size() ? blob.size() : 16));
     HIP_OK(hipMemcpy(weights, blob.data(), blob.size(),
                      hipMemcpyHostToDevice));
