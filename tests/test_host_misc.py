@@ -1,0 +1,143 @@
+"""Direct tests for the small host-framework modules that are
+otherwise only exercised indirectly (thread pool failure protection,
+distribution hooks, DOT export, path/import/timing utils)."""
+
+import os
+import threading
+import time
+
+import numpy
+import pytest  # noqa: F401
+
+
+def test_thread_pool_runs_and_joins():
+    from veles.thread_pool import ThreadPool
+    pool = ThreadPool(maxthreads=3)
+    hits = []
+    lock = threading.Lock()
+
+    def work(i):
+        with lock:
+            hits.append(i)
+    for i in range(20):
+        pool.callInThread(work, i)
+    pool.join()
+    assert sorted(hits) == list(range(20))
+    pool.shutdown()
+    # restartable after shutdown
+    pool.callInThread(work, 99)
+    pool.join()
+    assert 99 in hits
+    pool.shutdown()
+
+
+def test_thread_pool_failure_callback():
+    """Silent-failure protection: an exploding task reaches the
+    registered failure callback and the pool keeps serving."""
+    from veles.thread_pool import ThreadPool
+    pool = ThreadPool(maxthreads=1)
+    caught = []
+    pool.register_on_failure(caught.append)
+
+    def boom():
+        raise RuntimeError("task failed")
+
+    ok = []
+    pool.callInThread(boom)
+    pool.callInThread(ok.append, 1)
+    pool.join()
+    pool.shutdown()
+    assert len(caught) == 1 and "task failed" in str(caught[0])
+    assert ok == [1]
+
+
+def test_thread_pool_pause_resume():
+    from veles.thread_pool import ThreadPool
+    pool = ThreadPool(maxthreads=1)
+    ran = []
+    pool.pause()
+    pool.callInThread(ran.append, 1)
+    time.sleep(0.1)
+    assert ran == []          # paused: queued but not executed
+    pool.resume()
+    pool.join()
+    assert ran == [1]
+    pool.shutdown()
+
+
+def test_distributable_hooks_are_noops():
+    from veles.distributable import (Distributable,
+                                     TriviallyDistributable)
+    d = TriviallyDistributable()
+    assert d.generate_data_for_master() is None
+    assert d.generate_data_for_slave() is None
+    d.apply_data_from_master({"x": 1})
+    d.apply_data_from_slave({"x": 1})
+    d.drop_slave_state()
+    assert isinstance(d, Distributable)
+
+
+def test_workflow_to_dot(tmp_path):
+    from veles.dummy import DummyLauncher
+    from veles.config import root
+    import veles.znicz_amd.models.wine  # noqa: F401
+    root.wine.decision.max_epochs = 1
+    from veles.znicz_amd.models.wine import WineWorkflow
+    from veles.graph_plotter import workflow_to_dot, save_dot
+    wf = WineWorkflow(DummyLauncher())
+    dot = workflow_to_dot(wf)
+    assert dot.startswith("digraph") and dot.rstrip().endswith("}")
+    # every unit appears as a node; control links appear as edges
+    assert dot.count("->") >= len(list(wf)) - 2
+    for u in wf:
+        assert type(u).__name__ in dot
+    p = save_dot(wf, str(tmp_path / "wf.dot"))
+    assert os.path.exists(p) and open(p).read() == dot
+
+
+def test_import_file_as_module(tmp_path):
+    from veles.import_file import import_file_as_module, \
+        try_to_import_file
+    mod_path = tmp_path / "my_wf.py"
+    mod_path.write_text("VALUE = 41\n\ndef get():\n    return VALUE + 1\n")
+    mod = import_file_as_module(str(mod_path))
+    assert mod.get() == 42
+    bad = tmp_path / "broken.py"
+    bad.write_text("raise ValueError('nope')\n")
+    err = try_to_import_file(str(bad))
+    assert isinstance(err, ValueError)
+
+
+def test_timeit_helpers(capsys):
+    from veles.timeit2 import timeit, timed
+    res, dt = timeit(lambda a, b: a + b, 2, 3)
+    assert res == 5 and dt >= 0
+
+    @timed
+    def f(x):
+        return x * 2
+    assert f(4) == 8
+    assert "took" in capsys.readouterr().out
+
+
+def test_error_types():
+    from veles import error
+    for name in ("BadFormatError", "NotExistsError",
+                 "AlreadyExistsError", "DeviceError",
+                 "MasterSlaveCommunicationError"):
+        cls = getattr(error, name)
+        assert issubclass(cls, error.VelesException)
+        assert issubclass(cls, Exception)
+
+
+def test_paths_expand(monkeypatch):
+    from veles.paths import expand, __root__, __project_root__
+    assert os.path.isdir(__root__) and os.path.isdir(__project_root__)
+    monkeypatch.setenv("VELES_T_DIR", "/tmp/somewhere")
+    assert expand("$VELES_T_DIR/x") == "/tmp/somewhere/x"
+    assert expand("~").startswith("/")
+
+
+def test_numpy_available():
+    # keeps the numpy import (used by other tests' fixtures) honest
+    assert numpy.zeros(3).sum() == 0
