@@ -141,3 +141,47 @@ def test_paths_expand(monkeypatch):
 def test_numpy_available():
     # keeps the numpy import (used by other tests' fixtures) honest
     assert numpy.zeros(3).sum() == 0
+
+
+def test_accelerated_unit_dispatch():
+    """run() dispatches numpy_run on host devices, and missing
+    implementations fail loudly instead of silently no-opping."""
+    from veles.accelerated_units import AcceleratedUnit
+    from veles.backends import NumpyDevice
+    from veles.dummy import DummyWorkflow
+
+    calls = []
+
+    class U(AcceleratedUnit):
+        def numpy_run(self):
+            calls.append("numpy")
+
+        def hip_run(self):
+            calls.append("hip")
+
+    wf = DummyWorkflow()
+    u = U(wf)
+    u.initialize(device=NumpyDevice())
+    assert not u.on_gpu
+    u.run()
+    assert calls == ["numpy"]
+
+    class Bare(AcceleratedUnit):
+        pass
+    b = Bare(wf)
+    b.initialize(device=NumpyDevice())
+    with pytest.raises(NotImplementedError, match="numpy_run"):
+        b.run()
+    # legacy kernel-compilation shims: parity surface behaves sanely
+    assert u.build_program(defines={"X": 1}) is None
+    with pytest.raises(RuntimeError, match="precompiled"):
+        u.execute_kernel(global_size=(1,))
+
+
+def test_logger_setup_idempotent(capsys):
+    import logging
+    from veles.logger import setup_logging
+    setup_logging(logging.INFO)
+    n = len(logging.getLogger().handlers)
+    setup_logging(logging.INFO)     # second call must not stack handlers
+    assert len(logging.getLogger().handlers) == n
