@@ -134,6 +134,11 @@ class Main(object):
             raise SystemExit("%s does not define run(load, main)" %
                              self.args.workflow)
         if self.args.optimize > 0:
+            if int(os.environ.get("WORLD_SIZE", 1)) > 1 or \
+                    self.args.gpus > 1:
+                raise SystemExit(
+                    "--optimize runs single-process (each individual "
+                    "is a full workflow run); drop --gpus")
             return self._run_genetics(mod)
         mod.run(self.load, self.main)
         return 0
