@@ -185,3 +185,26 @@ def test_logger_setup_idempotent(capsys):
     n = len(logging.getLogger().handlers)
     setup_logging(logging.INFO)     # second call must not stack handlers
     assert len(logging.getLogger().handlers) == n
+
+
+def test_per_unit_runtime_accounting():
+    """per-unit wall-time accounting inside Unit.run()
+    wrappers + the workflow's print_stats table."""
+    from veles import prng
+    prng.seed_all(4)
+    from veles.config import root
+    import veles.znicz_amd.models.wine  # noqa: F401
+    root.wine.decision.max_epochs = 1
+    from veles.backends import NumpyDevice
+    from veles.dummy import DummyLauncher
+    from veles.znicz_amd.models.wine import WineWorkflow
+    wf = WineWorkflow(DummyLauncher())
+    wf.initialize(device=NumpyDevice())
+    wf.run()
+    ran = [u for u in wf if u.run_executed_counter > 0]
+    assert len(ran) >= 5
+    assert all(u.run_time >= 0 for u in ran)
+    assert any(u.run_time > 0 for u in ran)
+    rows = wf.print_stats(top=5)
+    assert len(rows) == 5
+    assert rows == sorted(rows, reverse=True)   # slowest first
