@@ -72,3 +72,35 @@ def test_short_minibatch_padding():
     assert ld.minibatch_size == 4
     assert (ld.minibatch_labels.mem[4:] == -1).all()
     assert (ld.minibatch_data.mem[4:] == 0).all()
+
+
+def test_sharding_non_divisible_world():
+    """world=3 over lengths not divisible by 3: the contract is EQUAL
+    per-rank stripes (ln // world per class — the remainder samples are
+    dropped that epoch) so epoch boundaries stay aligned across ranks;
+    what IS served must be disjoint and duplicate-free."""
+    lds = [make_loader(mb=9, n_train=40, n_valid=20, world=3, rank=r)
+           for r in (0, 1, 2)]
+    served = [[] for _ in range(3)]
+    for r, ld in enumerate(lds):
+        # equal truncated stripes on every rank
+        assert ld.class_lengths == [0, 20 // 3, 40 // 3]
+        guard = 0
+        while True:
+            ld.run()
+            served[r].extend(ld.minibatch_indices.mem[
+                :ld.minibatch_size].tolist())
+            guard += 1
+            assert guard < 40
+            if bool(ld.epoch_ended):
+                break
+        # all ranks finish their epoch after the same number of
+        # steps: ceil(6/9) VALID + ceil(13/9) TRAIN minibatches
+        assert guard == 3
+    flat = sum(served, [])
+    for a in range(3):
+        for b in range(a + 1, 3):
+            assert not (set(served[a]) & set(served[b])), (a, b)
+    # each rank served exactly its stripe, no duplicates; the
+    # remainder (60 - 3*(6+13) = 3 samples) sits out this epoch
+    assert len(flat) == len(set(flat)) == 3 * (6 + 13)
