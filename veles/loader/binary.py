@@ -53,8 +53,10 @@ def write_idx(path, arr):
     """Inverse of read_idx (used by tests and dataset tooling)."""
     codes = {numpy.dtype(numpy.uint8): 0x08,
              numpy.dtype(numpy.int8): 0x09,
+             numpy.dtype(numpy.int16): 0x0B,
              numpy.dtype(numpy.int32): 0x0C,
-             numpy.dtype(numpy.float32): 0x0D}
+             numpy.dtype(numpy.float32): 0x0D,
+             numpy.dtype(numpy.float64): 0x0E}
     code = codes[arr.dtype]
     opener = gzip.open if path.endswith(".gz") else open
     with opener(path, "wb") as f:
