@@ -134,3 +134,34 @@ def test_roundup_properties(n, align):
     from veles.memory import roundup
     r = roundup(n, align)
     assert r >= n and r % align == 0 and r - n < align
+
+
+# -- mzserve f2bf: RNE float->bf16 must match torch's host convert ----------
+
+def _f2bf(f):
+    """Transliteration of serve/mzserve.cpp f2bf (round-to-nearest-even)."""
+    import struct
+    u = struct.unpack("<I", struct.pack("<f", f))[0]
+    if (u & 0x7F800000) == 0x7F800000:     # inf/nan: truncate
+        return (u >> 16) & 0xFFFF
+    u = (u + 0x7FFF + ((u >> 16) & 1)) & 0xFFFFFFFF
+    return (u >> 16) & 0xFFFF
+
+
+@settings(max_examples=500, deadline=None)
+@given(st.floats(width=32, allow_nan=False))
+def test_f2bf_matches_torch(f):
+    import torch
+    want = torch.tensor([f], dtype=torch.float32).to(
+        torch.bfloat16).view(torch.int16).item() & 0xFFFF
+    assert _f2bf(f) == want, f
+
+
+def test_f2bf_specials():
+    import torch
+    for f in (0.0, -0.0, 1.0, -1.0, 3.0e38, -3.0e38, 1e-38, 6e-39):
+        want = torch.tensor([f], dtype=torch.float32).to(
+            torch.bfloat16).view(torch.int16).item() & 0xFFFF
+        assert _f2bf(f) == want, f
+    assert _f2bf(float("inf")) == 0x7F80
+    assert _f2bf(float("-inf")) == 0xFF80
