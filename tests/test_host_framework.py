@@ -445,3 +445,17 @@ def test_golden_snapshot_format_stable():
     wf.initialize(device=NumpyDevice())
     wf.run()
     assert int(wf.loader.epoch_number) == 4
+
+
+def test_velescli_shim(tmp_path):
+    # the reference's scripts/velescli.py entry point works like -m veles
+    cfg = tmp_path / "cfg.py"
+    cfg.write_text("root.wine.decision.max_epochs = 1\n"
+                   "root.wine.loader.n_train = 30\n"
+                   "root.wine.loader.n_valid = 0\n")
+    out = subprocess.run(
+        [sys.executable, "veles/scripts/velescli.py",
+         "veles/znicz_amd/models/wine.py", str(cfg), "-b", "numpy"],
+        cwd=REPO, capture_output=True, text=True, timeout=600)
+    assert out.returncode == 0, out.stderr[-2000:]
+    assert "epoch 1" in out.stderr or "epoch 1" in out.stdout
