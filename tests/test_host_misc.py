@@ -208,3 +208,11 @@ def test_per_unit_runtime_accounting():
     rows = wf.print_stats(top=5)
     assert len(rows) == 5
     assert rows == sorted(rows, reverse=True)   # slowest first
+
+
+def test_compat_shims():
+    from veles.compat import from_none, which, BytesIO, StringIO, PYPY
+    e = from_none(ValueError("x"))
+    assert e.__suppress_context__ and e.__cause__ is None
+    assert which("python3")
+    assert BytesIO and StringIO and PYPY is False
