@@ -163,3 +163,20 @@ def test_initialize_filters_unknown_kwargs():
     Narrow(wf)
     wf.initialize(device="d", snapshot="ignored.pickle")
     assert seen["device"] == "d"
+
+
+def test_link_attrs_one_way():
+    # two_way=False: reads alias, rebinding through the alias refuses
+    wf = DummyWorkflow()
+    a = TrivialUnit(wf, name="a")
+    b = TrivialUnit(wf, name="b")
+    a.output = numpy.zeros(3)
+    b.link_attrs(a, ("input", "output"), two_way=False)
+    assert b.input is a.output
+    try:
+        b.input = numpy.ones(2)
+    except AttributeError:
+        pass
+    else:
+        raise AssertionError("one-way link accepted a write")
+    assert a.output.shape == (3,)

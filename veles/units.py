@@ -111,11 +111,13 @@ class Unit(Logger, Distributable):
             u.unlink_from(self)
         return self
 
-    def link_attrs(self, other, *args, two_way=False):
+    def link_attrs(self, other, *args, two_way=True):
         """Alias attributes of *other* into self **by reference**.
 
         Each arg is either ``"attr"`` (same name both sides) or a tuple
-        ``("my_attr", "other_attr")``.
+        ``("my_attr", "other_attr")``.  Two-way by default (rebinding
+        the alias writes through to *other*, the reference semantics);
+        ``two_way=False`` makes rebinding through the alias an error.
         """
         table = self.__dict__.setdefault("_linked_attrs_", {})
         for arg in args:
@@ -126,7 +128,7 @@ class Unit(Logger, Distributable):
             # an existing instance attribute would shadow the link
             self.__dict__.pop(mine, None)
             table[mine] = LinkableAttribute(other, theirs,
-                                            two_way=True)
+                                            two_way=two_way)
         return self
 
     def unlink_attrs(self, *names):
