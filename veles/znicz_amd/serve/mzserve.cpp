@@ -326,7 +326,7 @@ int main(int argc, char** argv) {
     std::fprintf(stderr,
                  "usage: mzserve <export_dir> [--input f32.bin] "
                  "[--out f32.bin] [--batches K] [--warmup W] "
-                 "[--no-graph] [--parse-only]\n");
+                 "[--no-graph] [--parse-only] [--h2d]\n");
     return 2;
   }
   std::string dir = argv[1], in_file, out_file;
