@@ -178,3 +178,32 @@ def test_file_image_loader_mse_pairs(tmp_path):
     numpy.testing.assert_array_equal(
         ldr.minibatch_targets.mem[:5],
         ldr.original_targets.mem[idx])
+
+
+def test_file_image_dirs_train_end_to_end(tmp_path):
+    """PNG class directories → StandardWorkflow MLP → training learns:
+    the full files→train path for images (idx/cifar-bin have their own
+    end-to-end lanes above)."""
+    prng.seed_all(23)
+    pattern = _write_png_classes(tmp_path, n_per_class=10)
+    from veles.loader.image import FileImageLoader
+    from veles.znicz_amd.standard_workflow import StandardWorkflow
+    gd = {"learning_rate": 0.05, "gradient_moment": 0.9}
+    wf = StandardWorkflow(
+        DummyLauncher(),
+        layers=[{"type": "all2all_tanh",
+                 "->": {"output_sample_shape": 16}, "<-": dict(gd)},
+                {"type": "softmax", "->": {"output_sample_shape": 2},
+                 "<-": dict(gd)}],
+        loader_factory=lambda w, **kw: FileImageLoader(
+            w, train_paths=[pattern], scale=(8, 8),
+            normalize_range=(-1, 1), **kw),
+        loader_config={"minibatch_size": 10},
+        decision_config={"max_epochs": 6, "fail_iterations": 100},
+        snapshotter_config={"enabled": False})
+    wf.initialize(device=NumpyDevice())
+    wf.run()
+    assert wf.is_finished
+    # two well-separated brightness classes: must be nearly solved
+    assert wf.decision.epoch_n_err_pt[TRAIN] < 20.0, \
+        wf.decision.epoch_n_err_pt
