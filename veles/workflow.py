@@ -112,6 +112,10 @@ class Workflow(Container):
                 return u
         raise KeyError(name)
 
+    #: reference API parity: this engine's run() is always blocking (the
+    #: reference could also run detached under its Twisted reactor)
+    run_is_blocking = True
+
     @property
     def is_running(self):
         return self._is_running
