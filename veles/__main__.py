@@ -8,7 +8,9 @@ mutate the global ``root``), then invokes the module's
 
 MI355X additions: ``--gpus N`` self-spawns one process per GPU via
 ``torch.distributed.run`` (RCCL inside one node) instead of the
-reference's master/slave flags.
+reference's master/slave flags; ``--optimize GENS`` runs the genetics
+GA over ``Tune()`` config leaves (the reference's multi-run tuning
+flow, in-process).
 """
 
 import argparse
